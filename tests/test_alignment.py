@@ -1,0 +1,122 @@
+"""List alignment + recursive alignment tests (behavioral contract of
+reference consensus_utils.py:81-613, majority_sorting.py)."""
+
+import pytest
+
+from kllms_amd.consensus import (
+    lists_alignment,
+    recursive_list_alignments,
+    exists_nested_lists,
+    low_cutoff_bound,
+    sort_by_original_majority,
+)
+from kllms_amd.consensus.similarity import generic_similarity
+
+
+def no_embed(texts):
+    raise AssertionError("embeddings must not be called")
+
+
+def sim_fn(a, b):
+    return generic_similarity(a, b, "levenshtein", no_embed)
+
+
+class TestListsAlignment:
+    def test_identical_lists(self):
+        lists = [["apple", "banana"], ["apple", "banana"], ["apple", "banana"]]
+        aligned, orig_idx = lists_alignment(lists, sim_fn, min_support_ratio=0.51)
+        assert all(len(lst) == 2 for lst in aligned)
+        for lst in aligned:
+            assert set(lst) == {"apple", "banana"}
+        # column order follows majority original order
+        assert aligned[0] == ["apple", "banana"]
+        assert orig_idx[0] == [0, 1]
+
+    def test_permuted_lists(self):
+        lists = [["apple pie", "banana split"], ["banana split", "apple pie"], ["apple pie", "banana split"]]
+        aligned, _ = lists_alignment(lists, sim_fn, min_support_ratio=0.51)
+        # every row contains both items, aligned per column
+        cols = list(zip(*aligned))
+        for col in cols:
+            vals = {v for v in col if v is not None}
+            assert len(vals) == 1
+
+    def test_missing_element_pruned(self):
+        lists = [["aaaa"], ["aaaa"], ["aaaa", "zzzz"]]
+        aligned, _ = lists_alignment(lists, sim_fn, min_support_ratio=0.51)
+        # "zzzz" supported by 1/3 lists -> pruned
+        assert all(len(lst) == 1 for lst in aligned)
+        assert aligned[2] == ["aaaa"]
+
+    def test_empty_lists(self):
+        aligned, orig = lists_alignment([[], []], sim_fn)
+        assert aligned == [[], []]
+
+    def test_known_reference_idx(self):
+        lists = [["b", "a"], ["a", "b"]]
+        aligned, _ = lists_alignment(lists, sim_fn, reference_list_idx=0)
+        assert aligned[0] == ["b", "a"]
+        assert aligned[1] == ["b", "a"]
+
+
+class TestRecursiveAlignment:
+    def test_scalar_passthrough(self):
+        values = ["x", "y", None]
+        aligned, km = recursive_list_alignments(values, "levenshtein", no_embed, None, 0.51)
+        assert aligned == values
+        assert km[""] == ["", "", None]
+
+    def test_dict_of_lists(self):
+        values = [
+            {"items": ["alpha", "beta"]},
+            {"items": ["beta", "alpha"]},
+            {"items": ["alpha", "beta"]},
+        ]
+        aligned, km = recursive_list_alignments(values, "levenshtein", no_embed, None, 0.51)
+        for d in aligned:
+            assert len(d["items"]) == 2
+        cols = list(zip(*[d["items"] for d in aligned]))
+        for col in cols:
+            assert len({v for v in col if v is not None}) == 1
+        # key_mappings records aligned path -> original path per source
+        assert "items.0" in km
+        assert len(km["items.0"]) == 3
+
+    def test_does_not_mutate_input(self):
+        values = [{"a": ["x"]}, {"a": ["x"]}]
+        snapshot = [{"a": ["x"]}, {"a": ["x"]}]
+        recursive_list_alignments(values, "levenshtein", no_embed, None, 0.51)
+        assert values == snapshot
+
+    def test_all_none(self):
+        aligned, km = recursive_list_alignments([None, None], "levenshtein", no_embed, None, 0.51, current_path="p")
+        assert aligned == [None, None]
+        assert km == {"p": ["p", "p"]}
+
+    def test_mixed_types_passthrough(self):
+        values = [{"a": 1}, ["list"], "str"]
+        aligned, km = recursive_list_alignments(values, "levenshtein", no_embed, None, 0.51)
+        assert aligned == values
+
+
+def test_exists_nested_lists():
+    assert exists_nested_lists([[1]])
+    assert exists_nested_lists([{"a": [1]}])
+    assert not exists_nested_lists(["x", 1, {"a": "b"}])
+    assert not exists_nested_lists([])
+
+
+def test_low_cutoff_bound_empty():
+    assert low_cutoff_bound([]) == 0.0
+
+
+def test_sort_by_original_majority_basic():
+    originals = [["a", "b", "c"], ["a", "b", "c"]]
+    # aligned columns scrambled: (c, a, b) by identity
+    aligned = [
+        [originals[0][2], originals[0][0], originals[0][1]],
+        [originals[1][2], originals[1][0], originals[1][1]],
+    ]
+    sorted_lists, idx = sort_by_original_majority(aligned, originals)
+    assert sorted_lists[0] == ["a", "b", "c"]
+    assert idx[0] == [0, 1, 2]
