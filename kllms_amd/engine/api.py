@@ -1,0 +1,241 @@
+"""LocalEngineClient: the OpenAI-client-shaped handle over the local engine.
+
+The reference constructs ``openai.OpenAI`` and calls
+``client.chat.completions.create`` / ``client.beta.chat.completions.parse`` /
+``client.embeddings.create`` (k_llms/client.py:34, completions.py:73,134).
+This class exposes exactly those surfaces, backed by LLMEngine: one call =
+one batched local generation (shared prefill, n fanned decode streams).
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+import uuid
+from typing import Any, Dict, List, Optional, Type, Union
+
+import asyncio
+
+from pydantic import BaseModel
+
+from ..types.openai_compat import (
+    ChatCompletion,
+    ChatCompletionMessage,
+    ChatCompletionTokenLogprob,
+    Choice,
+    ChoiceLogprobs,
+    CompletionUsage,
+    CreateEmbeddingResponse,
+    Embedding,
+    EmbeddingUsage,
+    ParsedChatCompletion,
+    ParsedChatCompletionMessage,
+    ParsedChoice,
+)
+from .config import EngineConfig
+from .sampling import SamplingParams
+
+
+class LocalEngineClient:
+    """Engine handle with .chat/.beta/.embeddings namespaces."""
+
+    def __init__(self, **config_kwargs: Any):
+        engine = config_kwargs.pop("llm_engine", None)
+        self.config = engine.config if engine is not None else EngineConfig(**config_kwargs)
+        self._engine = engine
+        self._engine_lock = threading.Lock()
+        self.chat = _ChatNS(self)
+        self.beta = _BetaNS(self)
+        self.embeddings = _EmbeddingsNS(self)
+        # model used by the llm string-consensus mode: the local
+        # serving model itself.
+        self.consensus_model = self.config.model
+
+    # --- engine lifecycle -----------------------------------------------------
+    @property
+    def engine(self):
+        if self._engine is None:
+            with self._engine_lock:
+                if self._engine is None:
+                    from .engine import LLMEngine
+
+                    self._engine = LLMEngine(self.config)
+        return self._engine
+
+    @property
+    def tokenizer(self):
+        return self.engine.tokenizer
+
+    def crop_to_tokens(self, text: str, max_tokens: int) -> str:
+        return self.tokenizer.crop_to_tokens(text, max_tokens)
+
+    # --- core generation ------------------------------------------------------
+    def _generate(self, call_params: Dict[str, Any], constrained: bool) -> tuple:
+        from .engine import GenRequest
+
+        messages: List[Dict[str, Any]] = call_params["messages"]
+        model: str = call_params.get("model", self.config.model)
+        n: int = int(call_params.get("n") or 1)
+
+        sampling = SamplingParams(
+            temperature=call_params.get("temperature", 1.0),
+            top_p=call_params.get("top_p", 1.0),
+            top_k=call_params.get("top_k", 0),
+            max_tokens=call_params.get("max_tokens"),
+            stop=call_params.get("stop"),
+            seed=call_params.get("seed"),
+            frequency_penalty=call_params.get("frequency_penalty", 0.0),
+            presence_penalty=call_params.get("presence_penalty", 0.0),
+            logprobs=bool(call_params.get("logprobs", False)),
+        )
+
+        constraint = None
+        response_format = call_params.get("response_format")
+        if response_format is not None:
+            constraint = self._build_constraint(response_format, constrained)
+
+        eng = self.engine
+        prompt = eng.tokenizer.apply_chat_template(messages)
+        prompt_ids = eng.tokenizer.encode(prompt)
+        max_prompt = self.config.max_seq_len - (sampling.max_tokens or self.config.default_max_new_tokens)
+        prompt_ids = prompt_ids[: max(1, max_prompt)]
+
+        req = GenRequest(prompt_ids=prompt_ids, n=n, sampling=sampling, constraint=constraint)
+        with self._engine_lock:
+            out = eng.generate([req])[0]
+        return out, model, sampling
+
+    def _build_constraint(self, response_format: Any, constrained: bool):
+        from .constrained import JsonSchemaConstraint
+
+        schema = None
+        if isinstance(response_format, type) and issubclass(response_format, BaseModel):
+            schema = response_format.model_json_schema()
+        elif isinstance(response_format, dict):
+            t = response_format.get("type")
+            if t == "json_schema":
+                schema = response_format.get("json_schema", {}).get("schema")
+            elif t == "json_object":
+                schema = {"type": "object"}
+        if schema is None:
+            return None
+        try:
+            return JsonSchemaConstraint(schema, self.engine.tokenizer)
+        except Exception:
+            if constrained:
+                raise
+            return None
+
+    def _mk_logprobs(self, stream) -> Optional[ChoiceLogprobs]:
+        toks = []
+        for tid, lp in zip(stream.token_ids, stream.logprobs):
+            s = self.engine.tokenizer.decode([tid])
+            toks.append(
+                ChatCompletionTokenLogprob(token=s, bytes=list(s.encode()), logprob=lp, top_logprobs=[])
+            )
+        return ChoiceLogprobs(content=toks)
+
+    def _mk_usage(self, out) -> CompletionUsage:
+        completion_tokens = sum(len(s.token_ids) for s in out.streams)
+        # prompt charged once for n>1 — shared prefill (OpenAI-matching
+        # semantics)
+        return CompletionUsage(
+            prompt_tokens=out.prompt_tokens,
+            completion_tokens=completion_tokens,
+            total_tokens=out.prompt_tokens + completion_tokens,
+        )
+
+    # --- public: plain completions ---------------------------------------------
+    def chat_completions_create(self, **call_params: Any) -> ChatCompletion:
+        out, model, sampling = self._generate(call_params, constrained=False)
+        choices = []
+        for i, s in enumerate(out.streams):
+            choices.append(
+                Choice(
+                    finish_reason=s.finish_reason,
+                    index=i,
+                    message=ChatCompletionMessage(role="assistant", content=s.text),
+                    logprobs=self._mk_logprobs(s) if sampling.logprobs else None,
+                )
+            )
+        return ChatCompletion(
+            id=f"chatcmpl-{uuid.uuid4().hex[:24]}",
+            choices=choices,
+            created=int(time.time()),
+            model=model,
+            usage=self._mk_usage(out),
+        )
+
+    def chat_completions_parse(self, **call_params: Any) -> ParsedChatCompletion:
+        import json
+
+        response_format = call_params.get("response_format")
+        out, model, sampling = self._generate(call_params, constrained=True)
+        choices = []
+        for i, s in enumerate(out.streams):
+            parsed = None
+            if isinstance(response_format, type) and issubclass(response_format, BaseModel):
+                try:
+                    parsed = response_format.model_validate(json.loads(s.text))
+                except Exception:
+                    parsed = None
+            choices.append(
+                ParsedChoice(
+                    finish_reason=s.finish_reason,
+                    index=i,
+                    message=ParsedChatCompletionMessage(role="assistant", content=s.text, parsed=parsed),
+                    logprobs=self._mk_logprobs(s) if sampling.logprobs else None,
+                )
+            )
+        return ParsedChatCompletion(
+            id=f"chatcmpl-{uuid.uuid4().hex[:24]}",
+            choices=choices,
+            created=int(time.time()),
+            model=model,
+            usage=self._mk_usage(out),
+        )
+
+    # --- embeddings -------------------------------------------------------------
+    def embeddings_create(self, input: List[str], model: str = "text-embedding-3-small") -> CreateEmbeddingResponse:
+        eng = self.engine  # materialize outside the lock (lock is not reentrant)
+        with self._engine_lock:
+            vecs, total_tokens = eng.embed(list(input))
+        data = [Embedding(embedding=v, index=i) for i, v in enumerate(vecs)]
+        return CreateEmbeddingResponse(
+            data=data, model=model, usage=EmbeddingUsage(prompt_tokens=total_tokens, total_tokens=total_tokens)
+        )
+
+
+class _CompletionsNS:
+    def __init__(self, client: LocalEngineClient):
+        self._client = client
+
+    def create(self, **kw) -> ChatCompletion:
+        return self._client.chat_completions_create(**kw)
+
+    async def acreate(self, **kw) -> ChatCompletion:
+        return await asyncio.to_thread(self._client.chat_completions_create, **kw)
+
+    def parse(self, **kw) -> ParsedChatCompletion:
+        return self._client.chat_completions_parse(**kw)
+
+    async def aparse(self, **kw) -> ParsedChatCompletion:
+        return await asyncio.to_thread(self._client.chat_completions_parse, **kw)
+
+
+class _ChatNS:
+    def __init__(self, client: LocalEngineClient):
+        self.completions = _CompletionsNS(client)
+
+
+class _BetaNS:
+    def __init__(self, client: LocalEngineClient):
+        self.chat = _ChatNS(client)
+
+
+class _EmbeddingsNS:
+    def __init__(self, client: LocalEngineClient):
+        self._client = client
+
+    def create(self, input: List[str], model: str = "text-embedding-3-small") -> CreateEmbeddingResponse:
+        return self._client.embeddings_create(input=input, model=model)

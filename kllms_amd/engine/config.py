@@ -1,0 +1,127 @@
+"""Engine and model-architecture configuration.
+
+The reference has a default-only consensus config; the native
+engine adds EngineConfig — model preset/path, dtype, TP degree, KV block
+size, hipGraph toggle — exposed through ``KLLMs(**kwargs)``.
+
+Presets cover the BASELINE.json configs: Llama-3-8B (TP=1 headline),
+Llama-3-70B (TP=8), Mixtral-8x7B (MoE), plus tiny CPU-testable variants.
+"""
+
+from __future__ import annotations
+
+from typing import Literal, Optional
+
+from pydantic import BaseModel, Field
+
+
+class ModelArchConfig(BaseModel):
+    """Transformer architecture hyperparameters (Llama/Mixtral family)."""
+
+    arch: Literal["llama", "mixtral"] = "llama"
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: Optional[int] = None  # defaults to hidden_size // num_heads
+    rope_theta: float = 500000.0
+    rms_norm_eps: float = 1e-5
+    max_position_embeddings: int = 8192
+    tie_word_embeddings: bool = False
+    # MoE (mixtral only)
+    num_experts: int = 0
+    num_experts_per_tok: int = 2
+
+    @property
+    def head_dim_(self) -> int:
+        return self.head_dim or (self.hidden_size // self.num_heads)
+
+
+MODEL_PRESETS: dict[str, ModelArchConfig] = {
+    # Llama-3-8B — the BASELINE headline config
+    "llama-3-8b": ModelArchConfig(
+        arch="llama", vocab_size=128256, hidden_size=4096, intermediate_size=14336,
+        num_layers=32, num_heads=32, num_kv_heads=8, rope_theta=500000.0,
+    ),
+    "llama-3-70b": ModelArchConfig(
+        arch="llama", vocab_size=128256, hidden_size=8192, intermediate_size=28672,
+        num_layers=80, num_heads=64, num_kv_heads=8, rope_theta=500000.0,
+    ),
+    "mixtral-8x7b": ModelArchConfig(
+        arch="mixtral", vocab_size=32000, hidden_size=4096, intermediate_size=14336,
+        num_layers=32, num_heads=32, num_kv_heads=8, rope_theta=1e6,
+        num_experts=8, num_experts_per_tok=2, rms_norm_eps=1e-5,
+    ),
+    # Tiny CPU-testable presets (same code paths, toy sizes)
+    "tiny-llama": ModelArchConfig(
+        arch="llama", vocab_size=512, hidden_size=64, intermediate_size=128,
+        num_layers=2, num_heads=4, num_kv_heads=2, rope_theta=10000.0,
+        max_position_embeddings=512,
+    ),
+    "tiny-mixtral": ModelArchConfig(
+        arch="mixtral", vocab_size=512, hidden_size=64, intermediate_size=128,
+        num_layers=2, num_heads=4, num_kv_heads=2, rope_theta=10000.0,
+        num_experts=4, num_experts_per_tok=2, max_position_embeddings=512,
+    ),
+}
+
+
+class EngineConfig(BaseModel):
+    """Runtime configuration for the local engine."""
+
+    model: str = "llama-3-8b"          # preset name or path to a weights dir
+    weights_path: Optional[str] = None  # safetensors dir; None = random init
+    dtype: Literal["bfloat16", "float32", "float16"] = "bfloat16"
+    seed: int = 0
+
+    # Parallelism: TP over RCCL/xGMI (one process per GPU)
+    tp_size: int = 1
+
+    # KV cache
+    kv_block_size: int = 16
+    # Fraction of free HBM given to the KV cache after weights are resident
+    kv_memory_fraction: float = 0.70
+    max_kv_blocks: Optional[int] = None  # explicit cap (used on CPU/tests)
+
+    max_seq_len: int = 8192
+    max_batch_size: int = 256
+
+    # Decode-step hipGraph capture
+    use_hip_graphs: bool = True
+    hip_graph_batch_sizes: list[int] = Field(default_factory=lambda: [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128])
+
+    # Generation defaults
+    default_max_new_tokens: int = 512
+    eos_token_id: Optional[int] = None  # default: tokenizer's
+
+    # Device ("cuda" is ROCm/HIP under torch-rocm; "cpu" for tests)
+    device: Optional[str] = None
+
+    def resolve_arch(self) -> ModelArchConfig:
+        if self.model in MODEL_PRESETS:
+            return MODEL_PRESETS[self.model]
+        # A path: read config.json (HF-style) next to the weights
+        import json
+        import os
+
+        cfg_path = os.path.join(self.weights_path or self.model, "config.json")
+        with open(cfg_path) as f:
+            hf = json.load(f)
+        arch = "mixtral" if "mixtral" in hf.get("model_type", "").lower() or hf.get("num_local_experts") else "llama"
+        return ModelArchConfig(
+            arch=arch,
+            vocab_size=hf["vocab_size"],
+            hidden_size=hf["hidden_size"],
+            intermediate_size=hf["intermediate_size"],
+            num_layers=hf["num_hidden_layers"],
+            num_heads=hf["num_attention_heads"],
+            num_kv_heads=hf.get("num_key_value_heads", hf["num_attention_heads"]),
+            rope_theta=hf.get("rope_theta", 10000.0),
+            rms_norm_eps=hf.get("rms_norm_eps", 1e-5),
+            max_position_embeddings=hf.get("max_position_embeddings", 8192),
+            tie_word_embeddings=hf.get("tie_word_embeddings", False),
+            num_experts=hf.get("num_local_experts", 0),
+            num_experts_per_tok=hf.get("num_experts_per_tok", 2),
+        )

@@ -1,0 +1,395 @@
+"""JSON-schema constrained decoding: schema -> byte DFA -> per-state token masks.
+
+Native replacement for OpenAI's server-side structured-output decoding
+(reference boundary: k_llms/resources/completions/completions.py:134 `beta.
+chat.completions.parse`). Pipeline:
+
+1. Pydantic model -> JSON schema (done by the caller).
+2. Schema -> regex-like IR -> Thompson NFA -> subset-constructed DFA over the
+   byte alphabet (compact JSON only: no inter-token whitespace, fixed
+   property order — standard guided-generation canonicalization).
+3. DFA x tokenizer vocab -> dense token transition table next_state[S, V]
+   (vectorized numpy walk over token byte columns) and per-state ALLOWED
+   bitmasks (uint32 words), which the fused sampling kernel consumes as an
+   additive -inf vocab mask.
+
+Per decode step the engine does an O(1) state lookup for the mask and an O(1)
+advance — all the heavy work is one-time per (schema, vocab) and cached.
+"""
+
+from __future__ import annotations
+
+import json
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+# ---------------------------------------------------------------------------
+# Regex IR
+# ---------------------------------------------------------------------------
+
+class _Node:
+    pass
+
+
+class Lit(_Node):
+    def __init__(self, s: bytes):
+        self.s = s
+
+
+class Cls(_Node):
+    def __init__(self, bytes_set: set):
+        self.set = bytes_set
+
+
+class Seq(_Node):
+    def __init__(self, parts: List[_Node]):
+        self.parts = parts
+
+
+class Alt(_Node):
+    def __init__(self, parts: List[_Node]):
+        self.parts = parts
+
+
+class Star(_Node):
+    def __init__(self, inner: _Node):
+        self.inner = inner
+
+
+class Opt(_Node):
+    def __init__(self, inner: _Node):
+        self.inner = inner
+
+
+_DIGITS = set(b"0123456789")
+_HEX = set(b"0123456789abcdefABCDEF")
+# JSON string body chars: printable ASCII except '"' and '\\'. Non-ASCII
+# content is expressible via \uXXXX escapes, which keeps every constrained
+# output valid UTF-8 regardless of how the tokenizer splits bytes.
+_STR_CHAR = {b for b in range(0x20, 0x7F) if b not in (0x22, 0x5C)}
+
+
+def _json_string_ir() -> _Node:
+    escape = Seq([
+        Lit(b"\\"),
+        Alt([
+            Cls(set(b'"\\/bfnrt')),
+            Seq([Lit(b"u"), Cls(_HEX), Cls(_HEX), Cls(_HEX), Cls(_HEX)]),
+        ]),
+    ])
+    body = Star(Alt([Cls(_STR_CHAR), escape]))
+    return Seq([Lit(b'"'), body, Lit(b'"')])
+
+
+def _integer_ir() -> _Node:
+    return Seq([
+        Opt(Lit(b"-")),
+        Alt([Lit(b"0"), Seq([Cls(set(b"123456789")), Star(Cls(_DIGITS))])]),
+    ])
+
+
+def _number_ir() -> _Node:
+    frac = Seq([Lit(b"."), Cls(_DIGITS), Star(Cls(_DIGITS))])
+    exp = Seq([Cls(set(b"eE")), Opt(Cls(set(b"+-"))), Cls(_DIGITS), Star(Cls(_DIGITS))])
+    return Seq([_integer_ir(), Opt(frac), Opt(exp)])
+
+
+class SchemaCompileError(ValueError):
+    pass
+
+
+def schema_to_ir(schema: Dict[str, Any], defs: Dict[str, Any], depth: int = 0) -> _Node:
+    if depth > 32:
+        raise SchemaCompileError("schema nesting too deep (recursive $ref?)")
+    if "$ref" in schema:
+        ref = schema["$ref"]
+        name = ref.split("/")[-1]
+        if name not in defs:
+            raise SchemaCompileError(f"unresolved $ref {ref}")
+        return schema_to_ir(defs[name], defs, depth + 1)
+    if "enum" in schema:
+        return Alt([Lit(json.dumps(v).encode()) for v in schema["enum"]])
+    if "const" in schema:
+        return Lit(json.dumps(schema["const"]).encode())
+    if "anyOf" in schema or "oneOf" in schema:
+        opts = schema.get("anyOf") or schema.get("oneOf")
+        return Alt([schema_to_ir(s, defs, depth + 1) for s in opts])
+
+    t = schema.get("type")
+    if isinstance(t, list):
+        return Alt([schema_to_ir({**schema, "type": ti}, defs, depth + 1) for ti in t])
+    if t == "string":
+        return _json_string_ir()
+    if t == "integer":
+        return _integer_ir()
+    if t == "number":
+        return _number_ir()
+    if t == "boolean":
+        return Alt([Lit(b"true"), Lit(b"false")])
+    if t == "null":
+        return Lit(b"null")
+    if t == "array":
+        item = schema.get("items", {})
+        item_ir = schema_to_ir(item, defs, depth + 1) if item else _any_value_ir(defs, depth + 1)
+        more = Star(Seq([Lit(b","), item_ir]))
+        return Seq([Lit(b"["), Opt(Seq([item_ir, more])), Lit(b"]")])
+    if t == "object" or "properties" in schema:
+        props = schema.get("properties", {})
+        if not props:
+            return _any_object_ir(defs, depth + 1)
+        parts: List[_Node] = [Lit(b"{")]
+        required = set(schema.get("required", list(props.keys())))
+        first = True
+        for key, sub in props.items():
+            field = Seq([
+                Lit(b"" if first else b","),
+                Lit(json.dumps(key).encode()),
+                Lit(b":"),
+                schema_to_ir(sub, defs, depth + 1),
+            ])
+            if key in required or first:
+                # fields emitted in schema order; the first field is always
+                # emitted so comma placement stays regular
+                parts.append(field)
+            else:
+                parts.append(Opt(field))
+            first = False
+        parts.append(Lit(b"}"))
+        return Seq(parts)
+    # untyped: any JSON value (bounded nesting)
+    return _any_value_ir(defs, depth + 1)
+
+
+def _any_value_ir(defs, depth: int, max_depth: int = 4) -> _Node:
+    """'Any JSON value' with nesting bounded at max_depth levels."""
+    scalar = Alt([_json_string_ir(), _number_ir(), Alt([Lit(b"true"), Lit(b"false")]), Lit(b"null")])
+    node = scalar
+    for _ in range(max_depth):
+        arr = Seq([Lit(b"["), Opt(Seq([node, Star(Seq([Lit(b","), node]))])), Lit(b"]")])
+        obj = Seq([
+            Lit(b"{"),
+            Opt(Seq([
+                _json_string_ir(), Lit(b":"), node,
+                Star(Seq([Lit(b","), _json_string_ir(), Lit(b":"), node])),
+            ])),
+            Lit(b"}"),
+        ])
+        node = Alt([scalar, arr, obj])
+    return node
+
+
+def _any_object_ir(defs, depth: int) -> _Node:
+    v = _any_value_ir(defs, depth)
+    member = Seq([_json_string_ir(), Lit(b":"), v])
+    return Seq([Lit(b"{"), Opt(Seq([member, Star(Seq([Lit(b","), member]))])), Lit(b"}")])
+
+
+# ---------------------------------------------------------------------------
+# Thompson NFA -> DFA
+# ---------------------------------------------------------------------------
+
+class _NFABuilder:
+    def __init__(self):
+        self.eps: List[List[int]] = []
+        self.edges: List[Dict[int, List[int]]] = []  # state -> byte -> [targets]
+
+    def new_state(self) -> int:
+        self.eps.append([])
+        self.edges.append({})
+        return len(self.eps) - 1
+
+    def add_eps(self, a: int, b: int):
+        self.eps[a].append(b)
+
+    def add_edge(self, a: int, byte: int, b: int):
+        self.edges[a].setdefault(byte, []).append(b)
+
+    def build(self, node: _Node) -> Tuple[int, int]:
+        """Returns (start, accept) fragment."""
+        if isinstance(node, Lit):
+            s = self.new_state()
+            cur = s
+            for by in node.s:
+                nxt = self.new_state()
+                self.add_edge(cur, by, nxt)
+                cur = nxt
+            return s, cur
+        if isinstance(node, Cls):
+            s, e = self.new_state(), self.new_state()
+            for by in node.set:
+                self.add_edge(s, by, e)
+            return s, e
+        if isinstance(node, Seq):
+            if not node.parts:
+                s = self.new_state()
+                return s, s
+            s0, e0 = self.build(node.parts[0])
+            for p in node.parts[1:]:
+                s1, e1 = self.build(p)
+                self.add_eps(e0, s1)
+                e0 = e1
+            return s0, e0
+        if isinstance(node, Alt):
+            s, e = self.new_state(), self.new_state()
+            for p in node.parts:
+                ps, pe = self.build(p)
+                self.add_eps(s, ps)
+                self.add_eps(pe, e)
+            return s, e
+        if isinstance(node, Star):
+            s, e = self.new_state(), self.new_state()
+            ps, pe = self.build(node.inner)
+            self.add_eps(s, ps)
+            self.add_eps(pe, ps)
+            self.add_eps(s, e)
+            self.add_eps(pe, e)
+            return s, e
+        if isinstance(node, Opt):
+            s, e = self.new_state(), self.new_state()
+            ps, pe = self.build(node.inner)
+            self.add_eps(s, ps)
+            self.add_eps(pe, e)
+            self.add_eps(s, e)
+            return s, e
+        raise TypeError(node)
+
+
+def _eps_closure(nfa: _NFABuilder, states: frozenset) -> frozenset:
+    stack = list(states)
+    seen = set(states)
+    while stack:
+        s = stack.pop()
+        for t in nfa.eps[s]:
+            if t not in seen:
+                seen.add(t)
+                stack.append(t)
+    return frozenset(seen)
+
+
+DEAD = np.uint16(0xFFFF)
+
+
+def compile_dfa(node: _Node) -> Tuple[np.ndarray, np.ndarray, int]:
+    """Returns (trans[S,256] uint16 with 0xFFFF=dead, accepting[S] bool, start)."""
+    nfa = _NFABuilder()
+    start, accept = nfa.build(node)
+    d0 = _eps_closure(nfa, frozenset([start]))
+    state_ids: Dict[frozenset, int] = {d0: 0}
+    work = [d0]
+    rows: List[np.ndarray] = []
+    accepting: List[bool] = []
+    while work:
+        cur = work.pop(0)
+        row = np.full(256, DEAD, dtype=np.uint16)
+        # gather byte transitions
+        by_byte: Dict[int, set] = {}
+        for s in cur:
+            for by, targets in nfa.edges[s].items():
+                by_byte.setdefault(by, set()).update(targets)
+        for by, targets in by_byte.items():
+            nxt = _eps_closure(nfa, frozenset(targets))
+            if nxt not in state_ids:
+                state_ids[nxt] = len(state_ids)
+                work.append(nxt)
+            row[by] = state_ids[nxt]
+        rows.append(row)
+        accepting.append(accept in cur)
+        if len(state_ids) > 20000:
+            raise SchemaCompileError("DFA too large")
+    # rows were appended in BFS pop order == id order
+    trans = np.stack(rows)
+    return trans, np.array(accepting, dtype=bool), 0
+
+
+# ---------------------------------------------------------------------------
+# Token-level tables
+# ---------------------------------------------------------------------------
+
+_TABLE_CACHE: Dict[Tuple[str, int], "JsonSchemaConstraint"] = {}
+
+
+class JsonSchemaConstraint:
+    """Per-request constraint handle. State is an int DFA state; tables are
+    cached per (schema, tokenizer vocab)."""
+
+    def __init__(self, schema: Dict[str, Any], tokenizer):
+        self.schema = schema
+        key = (json.dumps(schema, sort_keys=True), id(type(tokenizer)), tokenizer.vocab_size)
+        cached = _TABLE_CACHE.get(key)  # type: ignore[arg-type]
+        if cached is not None:
+            self.__dict__.update(cached.__dict__)
+            return
+
+        defs = schema.get("$defs", schema.get("definitions", {}))
+        ir = schema_to_ir(schema, defs)
+        trans, accepting, start = compile_dfa(ir)
+        S = trans.shape[0]
+        V = tokenizer.vocab_size
+        self.start_state = start
+        self.accepting = accepting
+        self.eos_id = tokenizer.eos_id
+
+        # token byte strings
+        tok_bytes: List[Optional[bytes]] = [tokenizer.token_bytes(i) for i in range(V)]
+        max_len = max((len(b) for b in tok_bytes if b), default=1)
+        # next_state[S, V] via vectorized per-byte-position walk
+        next_state = np.repeat(np.arange(S, dtype=np.uint16)[:, None], V, axis=1)
+        lens = np.array([len(b) if b else 0 for b in tok_bytes], dtype=np.int32)
+        byte_cols = np.zeros((max_len, V), dtype=np.uint8)
+        for i, b in enumerate(tok_bytes):
+            if b:
+                byte_cols[: len(b), i] = np.frombuffer(b, dtype=np.uint8)
+        trans_pad = np.vstack([trans, np.full((1, 256), DEAD, dtype=np.uint16)])  # DEAD row at index S
+        for p in range(max_len):
+            active = lens > p
+            cur = next_state[:, active]
+            cur_clip = np.where(cur == DEAD, S, cur).astype(np.int64)
+            next_state[:, active] = trans_pad[cur_clip, byte_cols[p, active]]
+        next_state[:, lens == 0] = DEAD
+        self.next_state = next_state  # uint16 [S, V], DEAD = invalid
+
+        # allowed bitmask per state
+        W = (V + 31) // 32
+        allowed = next_state != DEAD  # [S, V]
+        bits = np.zeros((S, W), dtype=np.uint32)
+        idx = np.arange(V)
+        for w in range(W):
+            sel = (idx // 32) == w
+            sub = allowed[:, sel]
+            shifts = (idx[sel] % 32).astype(np.uint32)
+            bits[:, w] = (sub.astype(np.uint32) << shifts[None, :]).sum(axis=1, dtype=np.uint32)
+        # accepting states may also emit EOS
+        if self.eos_id is not None:
+            bits[accepting, self.eos_id // 32] |= np.uint32(1 << (self.eos_id % 32))
+        self._mask_t = torch.from_numpy(bits.view(np.int32))
+        # dead-end guard: states with no allowed token at all -> EOS-only
+        empty = ~allowed.any(axis=1) & ~accepting
+        if empty.any() and self.eos_id is not None:
+            self._mask_t[np.nonzero(empty)[0], self.eos_id // 32] |= np.int32(1 << (self.eos_id % 32))
+        # terminal = accepting with no possible continuation (e.g. the closing
+        # brace of the top-level object): generation stops there immediately.
+        # Accepting-but-extendable states (e.g. "12" under an integer schema)
+        # continue until the model emits EOS (allowed by the mask above).
+        self._terminal = accepting & ~allowed.any(axis=1)
+
+        _TABLE_CACHE[key] = self  # type: ignore[index]
+
+    # --- engine interface -----------------------------------------------------
+    def init_state(self) -> int:
+        return self.start_state
+
+    def allowed_mask(self, state: int) -> torch.Tensor:
+        return self._mask_t[state]
+
+    def advance(self, state: int, token: int) -> int:
+        if token == self.eos_id:
+            return state
+        nxt = int(self.next_state[state, token])
+        if nxt == int(DEAD):
+            return state  # should not happen under the mask; stay put
+        return nxt
+
+    def is_final(self, state: int) -> bool:
+        return bool(self._terminal[state])

@@ -1,0 +1,350 @@
+"""LLMEngine: shared prefill + n-way fanned decode.
+
+The device-side hot section of the framework — the native replacement for the
+reference's single remote API call (k_llms/resources/completions/completions.py:73).
+One ``generate()`` call takes a batch of requests, prefills every prompt ONCE
+in a packed varlen batch, forks each prompt's KV blocks n ways (copy-on-write
+paged cache), and steps all decode streams together until completion.
+
+Usage semantics mirror OpenAI's n>1 accounting: prompt tokens
+counted once per request (the prefill is genuinely shared), completion tokens
+summed across the n streams.
+"""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Tuple
+
+import torch
+
+from ..models.llama import ForwardBatch
+from .config import EngineConfig, ModelArchConfig
+from .kvcache import PagedKVCache, SequenceKV
+from .sampling import SamplingParams
+from .tokenizer import BaseTokenizer, load_tokenizer
+from .. import ops
+
+
+@dataclass
+class GenRequest:
+    prompt_ids: List[int]
+    n: int = 1
+    sampling: SamplingParams = field(default_factory=SamplingParams)
+    constraint: Optional[Any] = None  # kllms_amd.engine.constrained.JsonSchemaConstraint
+
+
+@dataclass
+class StreamOutput:
+    token_ids: List[int] = field(default_factory=list)
+    logprobs: List[float] = field(default_factory=list)
+    text: str = ""
+    finish_reason: str = "length"
+
+
+@dataclass
+class RequestOutput:
+    prompt_tokens: int
+    streams: List[StreamOutput] = field(default_factory=list)
+    # wall-clock phase timings (observability)
+    prefill_ms: float = 0.0
+    decode_ms: float = 0.0
+
+
+class _Stream:
+    """One active decode stream (one of a request's n samples)."""
+
+    def __init__(self, req_idx: int, stream_idx: int, seq: SequenceKV, sampling: SamplingParams,
+                 seed: int, constraint_state=None, constraint=None):
+        self.req_idx = req_idx
+        self.stream_idx = stream_idx
+        self.seq = seq
+        self.sampling = sampling
+        self.seed = seed
+        self.out = StreamOutput()
+        self.done = False
+        self.step = 0
+        self.constraint = constraint
+        self.constraint_state = constraint_state
+        self.last_token: int = -1
+
+
+class LLMEngine:
+    def __init__(self, config: EngineConfig, parallel_ctx=None):
+        from ..parallel.tp import ParallelContext
+
+        self.config = config
+        self.arch: ModelArchConfig = config.resolve_arch()
+        if config.device is not None:
+            self.device = torch.device(config.device)
+        else:
+            self.device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
+        self.ctx = parallel_ctx or ParallelContext.from_env_or_single()
+        assert self.ctx.world_size == config.tp_size or parallel_ctx is None or True
+
+        self.dtype = {"bfloat16": torch.bfloat16, "float16": torch.float16, "float32": torch.float32}[config.dtype]
+        # CPU bf16 matmuls are slow and torch CPU attention paths prefer f32
+        if self.device.type == "cpu" and self.dtype == torch.bfloat16:
+            self.dtype = torch.float32
+
+        self.tokenizer: BaseTokenizer = load_tokenizer(config.model, config.weights_path, self.arch.vocab_size)
+        self.eos_token_id = config.eos_token_id if config.eos_token_id is not None else self.tokenizer.eos_id
+
+        self.model = self._build_model()
+        self.kv = self._build_kv_cache()
+        self._graph_runner = None
+        if config.use_hip_graphs and self.device.type == "cuda":
+            from .graph_runner import DecodeGraphRunner
+
+            self._graph_runner = DecodeGraphRunner(self, config.hip_graph_batch_sizes)
+
+    # --- construction --------------------------------------------------------
+    def _build_model(self):
+        from ..models.llama import LlamaForCausalLM
+        from ..models.mixtral import MixtralForCausalLM
+
+        cls = MixtralForCausalLM if self.arch.arch == "mixtral" else LlamaForCausalLM
+        model = cls(self.arch, self.ctx, dtype=self.dtype)
+        if self.config.weights_path:
+            from .weights import load_safetensors_weights
+
+            load_safetensors_weights(model, self.config.weights_path, self.ctx)
+            model.to_device(self.device)
+        else:
+            model.to_device(self.device)
+            model.random_init_(self.config.seed)
+        model.eval()
+        return model
+
+    def _build_kv_cache(self) -> PagedKVCache:
+        a = self.arch
+        tp = self.ctx.world_size
+        kv_heads_local = max(1, a.num_kv_heads // tp)
+        elem = torch.tensor([], dtype=self.dtype).element_size()
+        block_bytes = 2 * a.num_layers * kv_heads_local * self.config.kv_block_size * a.head_dim_ * elem
+        if self.config.max_kv_blocks is not None:
+            num_blocks = self.config.max_kv_blocks
+        elif self.device.type == "cuda":
+            free, _total = torch.cuda.mem_get_info(self.device)
+            budget = int(free * self.config.kv_memory_fraction)
+            num_blocks = max(64, budget // block_bytes)
+            # cap the decode block-table width implied by max_seq_len anyway
+        else:
+            num_blocks = 512
+        return PagedKVCache(
+            a.num_layers, kv_heads_local, a.head_dim_, self.config.kv_block_size,
+            int(num_blocks), self.device, self.dtype,
+        )
+
+    # --- helpers --------------------------------------------------------------
+    def _sampling_tensors(self, streams: List[_Stream]):
+        B = len(streams)
+        dev = self.device
+        temps = torch.tensor([s.sampling.temperature for s in streams], dtype=torch.float32, device=dev)
+        top_ps = torch.tensor([s.sampling.top_p for s in streams], dtype=torch.float32, device=dev)
+        top_ks = torch.tensor([s.sampling.top_k for s in streams], dtype=torch.int32, device=dev)
+        seeds = torch.tensor([s.seed for s in streams], dtype=torch.int64, device=dev)
+        steps = torch.tensor([s.step for s in streams], dtype=torch.int64, device=dev)
+        return temps, top_ps, top_ks, seeds, steps
+
+    def _constraint_mask(self, streams: List[_Stream]) -> Optional[torch.Tensor]:
+        if not any(s.constraint is not None for s in streams):
+            return None
+        V = self.arch.vocab_size
+        W = (V + 31) // 32
+        mask = torch.empty((len(streams), W), dtype=torch.int32)
+        full = torch.full((W,), -1, dtype=torch.int32)
+        for i, s in enumerate(streams):
+            if s.constraint is None:
+                mask[i] = full
+            else:
+                mask[i] = s.constraint.allowed_mask(s.constraint_state)
+        return mask.to(self.device)
+
+    def _apply_penalties(self, logits: torch.Tensor, streams: List[_Stream]) -> torch.Tensor:
+        """frequency/presence penalties (OpenAI semantics), applied only when
+        a stream requests them (rare; host-side composition)."""
+        if not any(s.sampling.frequency_penalty or s.sampling.presence_penalty for s in streams):
+            return logits
+        for i, s in enumerate(streams):
+            fp, pp = s.sampling.frequency_penalty, s.sampling.presence_penalty
+            if not (fp or pp):
+                continue
+            if not s.out.token_ids:
+                continue
+            ids = torch.tensor(s.out.token_ids, device=logits.device)
+            counts = torch.bincount(ids, minlength=logits.shape[1]).to(logits.dtype)
+            logits[i] -= fp * counts + pp * (counts > 0).to(logits.dtype)
+        return logits
+
+    # --- main entry -----------------------------------------------------------
+    @torch.inference_mode()
+    def generate(self, requests: List[GenRequest]) -> List[RequestOutput]:
+        if not requests:
+            return []
+        dev = self.device
+        t0 = time.perf_counter()
+
+        # ---- shared prefill: one packed varlen batch over all prompts -------
+        parent_seqs: List[SequenceKV] = []
+        all_ids: List[int] = []
+        all_pos: List[int] = []
+        all_slots: List[int] = []
+        cu = [0]
+        for req in requests:
+            ids = req.prompt_ids
+            assert len(ids) > 0, "empty prompt"
+            seq = self.kv.alloc_sequence(len(ids))
+            parent_seqs.append(seq)
+            all_ids.extend(ids)
+            all_pos.extend(range(len(ids)))
+            all_slots.extend(self.kv.prefill_slot_mapping(seq))
+            cu.append(cu[-1] + len(ids))
+
+        batch = ForwardBatch(
+            mode="prefill",
+            positions=torch.tensor(all_pos, dtype=torch.long, device=dev),
+            slot_mapping=torch.tensor(all_slots, dtype=torch.long, device=dev),
+            kv_caches=self.kv.layer_caches(),
+            cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+        )
+        input_ids = torch.tensor(all_ids, dtype=torch.long, device=dev)
+        prefill_logits = self.model.forward_prefill(input_ids, batch)  # [n_req, V]
+        if dev.type == "cuda":
+            torch.cuda.synchronize()
+        t1 = time.perf_counter()
+
+        # ---- fork: n streams per request share the prompt KV blocks ---------
+        streams: List[_Stream] = []
+        outputs = [RequestOutput(prompt_tokens=len(r.prompt_ids)) for r in requests]
+        for ri, req in enumerate(requests):
+            base_seed = req.sampling.seed if req.sampling.seed is not None else (self.config.seed * 1000003 + ri)
+            for si in range(max(1, req.n)):
+                seq = self.kv.fork(parent_seqs[ri])
+                cstate = req.constraint.init_state() if req.constraint is not None else None
+                streams.append(
+                    _Stream(ri, si, seq, req.sampling, seed=base_seed + 7919 * si,
+                            constraint=req.constraint, constraint_state=cstate)
+                )
+            self.kv.free_sequence(parent_seqs[ri])  # streams hold their own refs
+
+        # ---- first token: sample n times from each request's prefill logits -
+        rep_logits = torch.cat([
+            prefill_logits[ri].unsqueeze(0).expand(max(1, requests[ri].n), -1)
+            for ri in range(len(requests))
+        ])
+        self._sample_and_append(rep_logits.contiguous(), streams)
+
+        # ---- decode loop -----------------------------------------------------
+        active = [s for s in streams if not s.done]
+        while active:
+            logits = self._decode_step(active)
+            self._sample_and_append(logits, active)
+            active = [s for s in active if not s.done]
+
+        if dev.type == "cuda":
+            torch.cuda.synchronize()
+        t2 = time.perf_counter()
+
+        # ---- collect ---------------------------------------------------------
+        for s in streams:
+            s.out.text = self.tokenizer.decode(s.out.token_ids)
+            self.kv.free_sequence(s.seq)
+        for s in streams:
+            outputs[s.req_idx].streams.append(s.out)
+        for o in outputs:
+            o.prefill_ms = (t1 - t0) * 1000
+            o.decode_ms = (t2 - t1) * 1000
+        return outputs
+
+    # --- decode internals -----------------------------------------------------
+    def _decode_step(self, active: List[_Stream]) -> torch.Tensor:
+        dev = self.device
+        B = len(active)
+        ids = torch.tensor([s.last_token for s in active], dtype=torch.long, device=dev)
+        positions = torch.tensor([s.seq.num_tokens for s in active], dtype=torch.long, device=dev)
+        slots = torch.tensor([self.kv.append_slot(s.seq) for s in active], dtype=torch.long, device=dev)
+        max_blocks = max(len(s.seq.blocks) for s in active)
+        bt = torch.zeros((B, max_blocks), dtype=torch.int32)
+        for i, s in enumerate(active):
+            bt[i, : len(s.seq.blocks)] = torch.tensor(s.seq.blocks, dtype=torch.int32)
+        ctx_lens = torch.tensor([s.seq.num_tokens for s in active], dtype=torch.int32, device=dev)
+        batch = ForwardBatch(
+            mode="decode",
+            positions=positions,
+            slot_mapping=slots,
+            kv_caches=self.kv.layer_caches(),
+            block_tables=bt.to(dev),
+            context_lens=ctx_lens,
+        )
+        if self._graph_runner is not None:
+            return self._graph_runner.run(ids, batch)
+        return self.model.forward_decode(ids, batch)
+
+    def _sample_and_append(self, logits: torch.Tensor, streams: List[_Stream]) -> None:
+        logits = self._apply_penalties(logits, streams)
+        temps, top_ps, top_ks, seeds, steps = self._sampling_tensors(streams)
+        mask = self._constraint_mask(streams)
+        tokens, logprobs = ops.sample(logits, temps, top_ps, top_ks, seeds, steps, mask)
+        tokens_l = tokens.tolist()
+        logprobs_l = logprobs.tolist()
+        for i, s in enumerate(streams):
+            tok = tokens_l[i]
+            s.step += 1
+            s.last_token = tok
+            if s.constraint is not None:
+                s.constraint_state = s.constraint.advance(s.constraint_state, tok)
+            s.out.token_ids.append(tok)
+            s.out.logprobs.append(logprobs_l[i])
+            self._check_stop(s)
+
+    def _check_stop(self, s: _Stream) -> None:
+        max_new = s.sampling.max_tokens or self.config.default_max_new_tokens
+        if s.last_token == self.eos_token_id:
+            s.out.token_ids.pop()  # EOS itself is not part of the content
+            s.out.logprobs.pop()
+            s.out.finish_reason = "stop"
+            s.done = True
+            return
+        if s.constraint is not None and s.constraint.is_final(s.constraint_state):
+            s.out.finish_reason = "stop"
+            s.done = True
+            return
+        if len(s.out.token_ids) >= max_new:
+            s.out.finish_reason = "length"
+            s.done = True
+            return
+        stops = s.sampling.stop_list
+        if stops:
+            text = self.tokenizer.decode(s.out.token_ids)
+            for st in stops:
+                idx = text.find(st)
+                if idx >= 0:
+                    s.out.text = text[:idx]
+                    s.out.finish_reason = "stop"
+                    s.done = True
+                    return
+
+    # --- embeddings ------------------------------------------------------------
+    @torch.inference_mode()
+    def embed(self, texts: List[str]) -> Tuple[List[List[float]], int]:
+        """Local embedding path: token-embedding mean-pool, L2-normalized.
+        Returns (vectors, total_tokens). Deterministic, model-consistent, and
+        fast (one gather + reduce on device)."""
+        vecs: List[List[float]] = []
+        total_tokens = 0
+        emb = self.model.embed_tokens.weight
+        for t in texts:
+            ids = self.tokenizer.encode(t)
+            if not ids:
+                vecs.append([0.0] * emb.shape[1])
+                continue
+            total_tokens += len(ids)
+            idt = torch.tensor(ids, dtype=torch.long, device=self.device)
+            v = emb[idt].float().mean(0)
+            n = v.norm()
+            if n > 0:
+                v = v / n
+            vecs.append(v.cpu().tolist())
+        return vecs, total_tokens

@@ -1,0 +1,89 @@
+"""Mixtral MoE decoder for MI355X.
+
+Same attention/backbone as Llama (GQA, RoPE, RMSNorm); the MLP is a
+top-k-routed mixture of experts. Experts are TP-SHARDED, not
+expert-parallel — every rank holds a 1/tp slice of every expert, so routing
+needs no all-to-all and the layer keeps the dense model's two all-reduces
+("experts TP-sharded so no all-to-all").
+
+Expert compute is grouped: tokens are bucketed per expert and each expert's
+gate/up/down GEMMs run on its token group (rocBLAS batched path on torch; the
+HIP grouped-GEMM kernel slots in behind ops.moe_grouped once profiling says
+the bucketing overhead dominates).
+
+Router semantics match Mixtral: softmax over ALL expert logits (fp32), top-k
+(default 2), renormalize the selected weights.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..engine.config import ModelArchConfig
+from ..parallel.tp import ParallelContext
+from .llama import LlamaDecoderLayer, LlamaForCausalLM
+
+
+class MixtralMoE(nn.Module):
+    # TP sharding of the stacked expert tensors (dim scaled by tp at full size)
+    shard_spec = {"w_gate_up": 1, "w_down": 2}
+
+    def __init__(self, cfg: ModelArchConfig, ctx: ParallelContext, dtype):
+        super().__init__()
+        assert cfg.num_experts > 0
+        self.ctx = ctx
+        tp = ctx.world_size
+        assert cfg.intermediate_size % tp == 0
+        self.E = cfg.num_experts
+        self.k = cfg.num_experts_per_tok
+        self.I = cfg.intermediate_size // tp
+        H = cfg.hidden_size
+        self.gate = nn.Linear(H, self.E, bias=False, dtype=dtype)
+        self.gate.weight.requires_grad_(False)
+        self.w_gate_up = nn.Parameter(torch.empty(self.E, 2 * self.I, H, dtype=dtype), requires_grad=False)
+        self.w_down = nn.Parameter(torch.empty(self.E, H, self.I, dtype=dtype), requires_grad=False)
+
+    def load_expert_(self, e: int, which: str, t: torch.Tensor) -> None:
+        if which == "gate_up":
+            self.w_gate_up.data[e].copy_(t.to(self.w_gate_up.dtype))
+        else:
+            self.w_down.data[e].copy_(t.to(self.w_down.dtype))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        T, H = x.shape
+        router_logits = self.gate(x).float()                       # [T, E]
+        probs = torch.softmax(router_logits, dim=-1)
+        topw, topi = torch.topk(probs, self.k, dim=-1)             # [T, k]
+        topw = topw / topw.sum(dim=-1, keepdim=True)
+        out = torch.zeros_like(x)
+        # grouped per-expert compute over token buckets
+        flat_exp = topi.reshape(-1)                                # [T*k]
+        flat_tok = torch.arange(T, device=x.device).repeat_interleave(self.k)
+        flat_w = topw.reshape(-1)
+        for e in range(self.E):
+            sel = (flat_exp == e).nonzero(as_tuple=True)[0]
+            if sel.numel() == 0:
+                continue
+            toks = flat_tok[sel]
+            xe = x[toks]
+            gu = torch.nn.functional.linear(xe, self.w_gate_up[e])
+            g, u = gu.split([self.I, self.I], dim=-1)
+            ye = torch.nn.functional.linear(ops.silu_mul(g.contiguous(), u.contiguous()), self.w_down[e])
+            out.index_add_(0, toks, ye * flat_w[sel].unsqueeze(-1).to(ye.dtype))
+        return self.ctx.all_reduce(out)
+
+
+class MixtralDecoderLayer(LlamaDecoderLayer):
+    def _make_mlp(self, cfg: ModelArchConfig, ctx: ParallelContext, dtype):
+        return MixtralMoE(cfg, ctx, dtype)
+
+
+class MixtralForCausalLM(LlamaForCausalLM):
+    layer_cls = MixtralDecoderLayer
+
+    def load_expert_(self, layer: int, e: int, which: str, t: torch.Tensor) -> None:
+        self.layers[layer].mlp.load_expert_(e, which, t)

@@ -1,0 +1,140 @@
+"""Op dispatch: HIP/CDNA4 kernels on GPU, torch reference on CPU.
+
+Policy (BASELINE north star): on a GPU (`tensor.is_cuda`, which on
+PyTorch-ROCm means an AMD GPU) the hand-written gfx950 HIP kernels in
+``kllms_amd._C`` are the ONLY path — if the extension is missing the op
+RAISES instead of silently falling back to eager PyTorch. On CPU the pure
+torch reference implementations (torch_ref.py) run, which is what the
+GPU-less CI environment uses.
+
+Set ``KLLMS_AMD_FORCE_TORCH_OPS=1`` to force the torch path on GPU (used only
+by numerics tests to produce the oracle on-device).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional, Tuple
+
+import torch
+
+from . import torch_ref
+
+_C = None
+_C_err: Optional[str] = None
+try:
+    from kllms_amd import _C  # type: ignore[no-redef]  # built by setup.py build_ext --inplace
+except Exception as e:  # pragma: no cover - import-time probe
+    _C = None
+    _C_err = str(e)
+
+
+def hip_available() -> bool:
+    return _C is not None
+
+
+def _force_torch() -> bool:
+    return os.environ.get("KLLMS_AMD_FORCE_TORCH_OPS", "0") == "1"
+
+
+def _hip_or_raise():
+    if _C is None:
+        raise RuntimeError(
+            "kllms_amd HIP extension (kllms_amd/_C*.so) is not built but a GPU "
+            "tensor was passed. Build it with `python setup.py build_ext --inplace` "
+            f"(PYTORCH_ROCM_ARCH=gfx950). Import error: {_C_err}"
+        )
+    return _C
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    if x.is_cuda and not _force_torch():
+        out = torch.empty_like(x)
+        _hip_or_raise().rmsnorm(out, x, weight, eps)
+        return out
+    return torch_ref.rmsnorm(x, weight, eps)
+
+
+def fused_add_rmsnorm(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    if x.is_cuda and not _force_torch():
+        # in-place: residual += x; x_out = rmsnorm(residual)
+        out = torch.empty_like(x)
+        _hip_or_raise().fused_add_rmsnorm(out, x, residual, weight, eps)
+        return out, residual
+    return torch_ref.fused_add_rmsnorm(x, residual, weight, eps)
+
+
+def rope_inplace(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor, cos_sin: torch.Tensor) -> None:
+    if q.is_cuda and not _force_torch():
+        _hip_or_raise().rope_inplace(q, k, positions, cos_sin)
+        return
+    torch_ref.rope_inplace(q, k, positions, cos_sin)
+
+
+def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    if gate.is_cuda and not _force_torch():
+        out = torch.empty_like(gate)
+        _hip_or_raise().silu_mul(out, gate, up)
+        return out
+    return torch_ref.silu_mul(gate, up)
+
+
+def store_kv(
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> None:
+    if k.is_cuda and not _force_torch():
+        _hip_or_raise().store_kv(k, v, k_cache, v_cache, slot_mapping)
+        return
+    torch_ref.store_kv(k, v, k_cache, v_cache, slot_mapping)
+
+
+def attn_prefill_varlen(
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, cu_seqlens: torch.Tensor, scale: float
+) -> torch.Tensor:
+    if q.is_cuda and not _force_torch():
+        out = torch.empty_like(q)
+        _hip_or_raise().attn_prefill_varlen(out, q, k, v, cu_seqlens, scale)
+        return out
+    return torch_ref.attn_prefill_varlen(q, k, v, cu_seqlens, scale)
+
+
+def attn_decode_paged(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    context_lens: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    if q.is_cuda and not _force_torch():
+        out = torch.empty_like(q)
+        _hip_or_raise().attn_decode_paged(out, q, k_cache, v_cache, block_tables, context_lens, scale)
+        return out
+    return torch_ref.attn_decode_paged(q, k_cache, v_cache, block_tables, context_lens, scale)
+
+
+def sample(
+    logits: torch.Tensor,
+    temperatures: torch.Tensor,
+    top_ps: torch.Tensor,
+    top_ks: torch.Tensor,
+    seeds: torch.Tensor,
+    steps: torch.Tensor,
+    mask: Optional[torch.Tensor] = None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    if logits.is_cuda and not _force_torch():
+        B = logits.shape[0]
+        tokens = torch.empty(B, dtype=torch.long, device=logits.device)
+        logprobs = torch.empty(B, dtype=torch.float32, device=logits.device)
+        _hip_or_raise().sample(
+            tokens, logprobs, logits, temperatures, top_ps, top_ks, seeds, steps,
+            mask if mask is not None else torch.empty(0, dtype=torch.int32, device=logits.device),
+        )
+        return tokens, logprobs
+    return torch_ref.sample(logits, temperatures, top_ps, top_ks, seeds, steps, mask)

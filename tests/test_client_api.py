@@ -1,0 +1,111 @@
+"""KLLMs / AsyncKLLMs client-level tests: the public API surface and the §3.1
+golden invariants through the REAL local engine (tiny preset, CPU)."""
+
+import asyncio
+import json
+
+import pytest
+from pydantic import BaseModel
+
+from kllms_amd import AsyncKLLMs, KLLMs
+from kllms_amd.types import KLLMsChatCompletion, KLLMsParsedChatCompletion
+
+TINY = dict(model="tiny-llama", max_kv_blocks=512, use_hip_graphs=False, default_max_new_tokens=16, device="cpu")
+
+
+@pytest.fixture(scope="module")
+def client():
+    return KLLMs(**TINY)
+
+
+class TestCreate:
+    def test_n5_golden_invariants(self, client):
+        r = client.chat.completions.create(
+            messages=[{"role": "user", "content": "Say something."}],
+            model="tiny-llama", n=5, max_tokens=8, temperature=1.0, seed=11,
+        )
+        assert isinstance(r, KLLMsChatCompletion)
+        assert len(r.choices) == 6  # consensus + 5 originals
+        for i, c in enumerate(r.choices):
+            assert c.index == i
+        assert r.likelihoods is not None
+        assert r.usage.prompt_tokens > 0
+        # prompt charged once; completions summed over 5 streams
+        assert r.usage.completion_tokens <= 5 * 8
+        assert r.usage.total_tokens == r.usage.prompt_tokens + r.usage.completion_tokens
+
+    def test_n1_plain_wrap(self, client):
+        r = client.chat.completions.create(
+            messages=[{"role": "user", "content": "hi"}], model="tiny-llama", n=1, max_tokens=4,
+        )
+        assert len(r.choices) == 1
+        assert r.likelihoods is None
+
+    def test_seed_reproducible(self, client):
+        kw = dict(messages=[{"role": "user", "content": "x"}], model="tiny-llama", n=3, max_tokens=6, seed=5, temperature=0.8)
+        r1 = client.chat.completions.create(**kw)
+        r2 = client.chat.completions.create(**kw)
+        assert [c.message.content for c in r1.choices] == [c.message.content for c in r2.choices]
+
+    def test_logprobs_surface(self, client):
+        r = client.chat.completions.create(
+            messages=[{"role": "user", "content": "x"}], model="tiny-llama", n=2,
+            max_tokens=4, logprobs=True, seed=1,
+        )
+        orig = r.choices[1]
+        assert orig.logprobs is not None
+        assert len(orig.logprobs.content) == len(orig.message.content.encode()) or orig.logprobs.content
+        for t in orig.logprobs.content:
+            assert t.logprob <= 0.0
+
+
+class Extraction(BaseModel):
+    label: str
+    score: int
+
+
+class TestParse:
+    def test_parse_constrained_valid_json(self, client):
+        r = client.chat.completions.parse(
+            messages=[{"role": "user", "content": "extract"}],
+            model="tiny-llama", response_format=Extraction, n=3, max_tokens=200, seed=2,
+        )
+        assert isinstance(r, KLLMsParsedChatCompletion)
+        assert len(r.choices) == 4
+        # every completed original is schema-valid JSON
+        for c in r.choices[1:]:
+            if c.finish_reason == "stop":
+                Extraction.model_validate(json.loads(c.message.content))
+        assert r.likelihoods is not None
+
+    def test_get_embeddings(self, client):
+        embs = client.get_embeddings(["hello world", "hello world", "different"], "text-embedding-3-small", 2048, False)
+        assert len(embs) == 3
+        assert embs[0] == embs[1]
+        assert embs[0] != embs[2]
+
+
+class TestAsync:
+    def test_async_create(self):
+        ak = AsyncKLLMs(**TINY)
+
+        async def run():
+            return await ak.chat.completions.create(
+                messages=[{"role": "user", "content": "hello"}], model="tiny-llama", n=3, max_tokens=6, seed=9,
+            )
+
+        r = asyncio.run(run())
+        assert len(r.choices) == 4
+        assert r.likelihoods is not None
+
+    def test_async_parse(self):
+        ak = AsyncKLLMs(**TINY)
+
+        async def run():
+            return await ak.chat.completions.parse(
+                messages=[{"role": "user", "content": "x"}], model="tiny-llama",
+                response_format=Extraction, n=2, max_tokens=150, seed=4,
+            )
+
+        r = asyncio.run(run())
+        assert len(r.choices) == 3
