@@ -1,0 +1,174 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: end-to-end n-way consensus serving on MI355X.
+
+Measures the BASELINE.json headline metric — end-to-end consensus requests/sec
+(and per-request latency) at n=5 on Llama-3-8B TP=1 — on synthetic prompts and
+random-init weights (no network). Each timed step serves a fixed batch of
+consensus requests end to end: shared prefill, n fanned decode streams per
+request (HIP kernels, paged KV, hipGraph decode), then alignment + consensus
+consolidation into the final KLLMsChatCompletion.
+
+Multi-GPU (--gpus N, launched by torch.distributed.run with one rank per GPU):
+data-parallel serving — each rank runs an independent TP=1 engine on its GPU
+and serves its own request stream (weak scaling; the reported value is the
+whole-job aggregate across ranks).
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
+Rank 0 prints exactly one JSON result line.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+REFERENCE_BASELINE_REQ_S = 1.2  # consensus n=3 throughput vs OpenAI API
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def make_prompt(rank: int, step: int, i: int, prompt_len_tokens: int) -> str:
+    # deterministic synthetic prompt of roughly prompt_len_tokens byte-tokens
+    seedtxt = f"Request {rank}-{step}-{i}: extract the entities. "
+    body = ("The quick brown fox jumps over the lazy dog near the riverbank. " * 200)
+    return (seedtxt + body)[:prompt_len_tokens]
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--model", type=str, default="llama-3-8b")
+    ap.add_argument("--n", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=4, help="consensus requests per step per GPU")
+    ap.add_argument("--prompt-len", type=int, default=512, help="approx prompt tokens")
+    ap.add_argument("--max-new", type=int, default=64, help="decode tokens per stream")
+    ap.add_argument("--no-graphs", action="store_true")
+    args = ap.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    on_gpu = torch.cuda.is_available()
+    dist = None
+    if world_size > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        backend = "nccl" if on_gpu else "gloo"
+        dist.init_process_group(backend=backend)
+        if on_gpu:
+            torch.cuda.set_device(local_rank)
+
+    device = f"cuda:{local_rank}" if on_gpu else "cpu"
+
+    from kllms_amd import KLLMs
+    from kllms_amd.consensus.consolidation import consolidate_chat_completions
+
+    log(f"[bench] building engine: {args.model} on {device}")
+    t_build = time.perf_counter()
+    client_kwargs = dict(
+        model=args.model,
+        device=device,
+        use_hip_graphs=(on_gpu and not args.no_graphs),
+        seed=rank,
+        default_max_new_tokens=args.max_new,
+        max_seq_len=max(1024, args.prompt_len + args.max_new + 64),
+    )
+    if not on_gpu:
+        client_kwargs["max_kv_blocks"] = 2048
+    k = KLLMs(**client_kwargs)
+    eng_client = k.client
+    _ = eng_client.engine  # materialize weights now
+    log(f"[bench] engine ready in {time.perf_counter() - t_build:.1f}s")
+
+    def embeddings_wrapper(texts):
+        return k.get_embeddings(texts, "text-embedding-3-small", 2048, False)
+
+    def run_step(step_idx: int) -> None:
+        call_params_list = [
+            {
+                "messages": [{"role": "user", "content": make_prompt(rank, step_idx, i, args.prompt_len)}],
+                "model": args.model,
+                "n": args.n,
+                "temperature": 0.8,
+                "max_tokens": args.max_new,
+                "seed": 10_000 * rank + 100 * step_idx + i,
+            }
+            for i in range(args.batch)
+        ]
+        completions = eng_client.chat_completions_create_many(call_params_list)
+        for comp in completions:
+            consolidate_chat_completions(comp, embeddings_wrapper, client=eng_client)
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    for w in range(args.warmup):
+        log(f"[bench] warmup {w + 1}/{args.warmup}")
+        run_step(-1 - w)
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for s in range(args.steps):
+        run_step(s)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if dist.get_backend() == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_ranks = world_size
+    total_requests = args.steps * args.batch * n_ranks
+    value = total_requests / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        result = {
+            "metric": "end_to_end_consensus_requests_per_s",
+            "value": round(value, 4),
+            "unit": "req/s",
+            "n_gpus": n_ranks,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(value / REFERENCE_BASELINE_REQ_S, 3),
+            "dtype": "bf16" if on_gpu else "float32",
+            "data": "synthetic prompts, random-init weights",
+            "config": {
+                "model": args.model,
+                "n": args.n,
+                "global_batch": args.batch * n_ranks,
+                "seq_len": args.prompt_len,
+                "max_new_tokens": args.max_new,
+                "parallelism": f"dp{n_ranks} tp1",
+                "completions_per_s": round(value * args.n, 3),
+                "consensus_latency_s_per_request": round(elapsed / (args.steps * args.batch), 4),
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

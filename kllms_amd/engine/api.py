@@ -146,6 +146,59 @@ class LocalEngineClient:
         )
 
     # --- public: plain completions ---------------------------------------------
+    def chat_completions_create_many(self, call_params_list: List[Dict[str, Any]]) -> List[ChatCompletion]:
+        """Batched serving path: ALL requests go through ONE engine.generate()
+        call — every prompt prefilled in one packed varlen batch, every decode
+        stream stepped together. Used by the bench/serving layer."""
+        from .engine import GenRequest
+
+        eng = self.engine
+        reqs = []
+        samplings = []
+        for call_params in call_params_list:
+            messages = call_params["messages"]
+            n = int(call_params.get("n") or 1)
+            sampling = SamplingParams(
+                temperature=call_params.get("temperature", 1.0),
+                top_p=call_params.get("top_p", 1.0),
+                top_k=call_params.get("top_k", 0),
+                max_tokens=call_params.get("max_tokens"),
+                stop=call_params.get("stop"),
+                seed=call_params.get("seed"),
+                logprobs=bool(call_params.get("logprobs", False)),
+            )
+            constraint = None
+            rf = call_params.get("response_format")
+            if rf is not None:
+                constraint = self._build_constraint(rf, constrained=False)
+            prompt = eng.tokenizer.apply_chat_template(messages)
+            prompt_ids = eng.tokenizer.encode(prompt)
+            reqs.append(GenRequest(prompt_ids=prompt_ids, n=n, sampling=sampling, constraint=constraint))
+            samplings.append(sampling)
+        with self._engine_lock:
+            outs = eng.generate(reqs)
+        results = []
+        for call_params, out, sampling in zip(call_params_list, outs, samplings):
+            choices = [
+                Choice(
+                    finish_reason=s.finish_reason,
+                    index=i,
+                    message=ChatCompletionMessage(role="assistant", content=s.text),
+                    logprobs=self._mk_logprobs(s) if sampling.logprobs else None,
+                )
+                for i, s in enumerate(out.streams)
+            ]
+            results.append(
+                ChatCompletion(
+                    id=f"chatcmpl-{uuid.uuid4().hex[:24]}",
+                    choices=choices,
+                    created=int(time.time()),
+                    model=call_params.get("model", self.config.model),
+                    usage=self._mk_usage(out),
+                )
+            )
+        return results
+
     def chat_completions_create(self, **call_params: Any) -> ChatCompletion:
         out, model, sampling = self._generate(call_params, constrained=False)
         choices = []

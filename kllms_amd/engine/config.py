@@ -54,6 +54,12 @@ MODEL_PRESETS: dict[str, ModelArchConfig] = {
         num_layers=32, num_heads=32, num_kv_heads=8, rope_theta=1e6,
         num_experts=8, num_experts_per_tok=2, rms_norm_eps=1e-5,
     ),
+    # Mid preset for GPU kernel/engine tests: real head_dim (128), small depth
+    "mid-llama": ModelArchConfig(
+        arch="llama", vocab_size=2048, hidden_size=512, intermediate_size=1024,
+        num_layers=2, num_heads=4, num_kv_heads=2, rope_theta=10000.0,
+        max_position_embeddings=2048,
+    ),
     # Tiny CPU-testable presets (same code paths, toy sizes)
     "tiny-llama": ModelArchConfig(
         arch="llama", vocab_size=512, hidden_size=64, intermediate_size=128,

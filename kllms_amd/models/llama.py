@@ -166,7 +166,7 @@ class LlamaForCausalLM(nn.Module):
         rank = self.ctx.rank
         for name, p in self.named_parameters():
             h = int(hashlib.md5(f"{seed}/{name}".encode()).hexdigest()[:15], 16)
-            g = torch.Generator(device="cpu")
+            g = torch.Generator(device=p.device)
             g.manual_seed(h)
             shard_dim = None
             full_shape = list(p.shape)
@@ -183,10 +183,10 @@ class LlamaForCausalLM(nn.Module):
                 shard_dim = spec[leaf]
                 full_shape[shard_dim] *= tp
             if "layernorm" in name or name.endswith("norm.weight"):
-                full = torch.ones(full_shape, dtype=torch.float32)
+                full = torch.ones(full_shape, dtype=torch.float32, device=p.device)
             else:
                 std = 0.02
-                full = torch.randn(full_shape, generator=g, dtype=torch.float32) * std
+                full = torch.randn(full_shape, generator=g, dtype=torch.float32, device=p.device) * std
             if shard_dim is not None and tp > 1:
                 size = p.shape[shard_dim]
                 full = full.narrow(shard_dim, rank * size, size)

@@ -1,0 +1,143 @@
+// PyTorch bindings for the kllms_amd gfx950 kernels (kllms_amd._C).
+// Thin launch shims: shape/dtype checks, grid math, current-HIP-stream launch.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+// kernels (defined in the sibling .hip TUs)
+extern "C" __global__ void rmsnorm_kernel(bf16_t*, const bf16_t*, const bf16_t*, int, float);
+extern "C" __global__ void fused_add_rmsnorm_kernel(bf16_t*, const bf16_t*, bf16_t*, const bf16_t*, int, float);
+extern "C" __global__ void silu_mul_kernel(bf16_t*, const bf16_t*, const bf16_t*, int64_t);
+extern "C" __global__ void rope_kernel(bf16_t*, bf16_t*, const int64_t*, const float*, int, int, int);
+extern "C" __global__ void store_kv_kernel(const bf16_t*, const bf16_t*, bf16_t*, bf16_t*, const int64_t*, int, int, int, int);
+extern "C" __global__ void attn_decode_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, float, int, int, int, int, int);
+extern "C" __global__ void attn_prefill_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, const int*, float, int, int);
+extern "C" __global__ void sample_kernel(int64_t*, float*, const float*, const float*, const float*, const int*, const int64_t*, const int64_t*, const uint32_t*, int);
+extern "C" __global__ void mfma_selftest_kernel(float*, const bf16_t*, const bf16_t*);
+
+#define CHECK_BF16_CONTIG(t) \
+  TORCH_CHECK((t).is_cuda() && (t).is_contiguous() && (t).scalar_type() == at::kBFloat16, #t " must be contiguous bf16 on GPU")
+
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+static bf16_t* bf(torch::Tensor& t) { return reinterpret_cast<bf16_t*>(t.data_ptr()); }
+static const bf16_t* cbf(const torch::Tensor& t) { return reinterpret_cast<const bf16_t*>(t.data_ptr()); }
+
+void rmsnorm(torch::Tensor out, torch::Tensor in, torch::Tensor weight, double eps) {
+  CHECK_BF16_CONTIG(out); CHECK_BF16_CONTIG(in); CHECK_BF16_CONTIG(weight);
+  const int n = in.size(-1);
+  TORCH_CHECK(n % 8 == 0, "hidden size must be a multiple of 8");
+  const int rows = in.numel() / n;
+  hipLaunchKernelGGL(rmsnorm_kernel, dim3(rows), dim3(256), 0, cur_stream(),
+                     bf(out), cbf(in), cbf(weight), n, (float)eps);
+}
+
+void fused_add_rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor residual,
+                       torch::Tensor weight, double eps) {
+  CHECK_BF16_CONTIG(out); CHECK_BF16_CONTIG(x); CHECK_BF16_CONTIG(residual); CHECK_BF16_CONTIG(weight);
+  const int n = x.size(-1);
+  TORCH_CHECK(n % 8 == 0, "hidden size must be a multiple of 8");
+  const int rows = x.numel() / n;
+  hipLaunchKernelGGL(fused_add_rmsnorm_kernel, dim3(rows), dim3(256), 0, cur_stream(),
+                     bf(out), cbf(x), bf(residual), cbf(weight), n, (float)eps);
+}
+
+void silu_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up) {
+  CHECK_BF16_CONTIG(out); CHECK_BF16_CONTIG(gate); CHECK_BF16_CONTIG(up);
+  const int64_t nvec = gate.numel() / 8;
+  TORCH_CHECK(gate.numel() % 8 == 0, "numel must be a multiple of 8");
+  const int grid = (int)std::min<int64_t>((nvec + 255) / 256, 2048);
+  hipLaunchKernelGGL(silu_mul_kernel, dim3(grid), dim3(256), 0, cur_stream(),
+                     bf(out), cbf(gate), cbf(up), nvec);
+}
+
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions, torch::Tensor cos_sin) {
+  CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k);
+  TORCH_CHECK(positions.scalar_type() == at::kLong && cos_sin.scalar_type() == at::kFloat);
+  const int T = q.size(0), H = q.size(1), D = q.size(2), KVH = k.size(1);
+  hipLaunchKernelGGL(rope_kernel, dim3(T, H + KVH), dim3(D / 2), 0, cur_stream(),
+                     bf(q), bf(k), positions.data_ptr<int64_t>(), cos_sin.data_ptr<float>(),
+                     H, KVH, D);
+}
+
+void store_kv(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache, torch::Tensor v_cache,
+              torch::Tensor slot_mapping) {
+  CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v); CHECK_BF16_CONTIG(k_cache); CHECK_BF16_CONTIG(v_cache);
+  const int T = k.size(0), KVH = k.size(1), D = k.size(2);
+  const int BS = k_cache.size(2);
+  const int64_t total = (int64_t)T * KVH * (D / 8);
+  const int grid = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL(store_kv_kernel, dim3(grid), dim3(256), 0, cur_stream(),
+                     cbf(k), cbf(v), bf(k_cache), bf(v_cache),
+                     slot_mapping.data_ptr<int64_t>(), T, KVH, D, BS);
+}
+
+void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
+                       torch::Tensor v_cache, torch::Tensor block_tables,
+                       torch::Tensor context_lens, double scale) {
+  CHECK_BF16_CONTIG(out); CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k_cache); CHECK_BF16_CONTIG(v_cache);
+  const int B = q.size(0), H = q.size(1), D = q.size(2);
+  const int KVH = k_cache.size(1), BS = k_cache.size(2);
+  TORCH_CHECK(D == 128, "attn_decode: head_dim must be 128");
+  TORCH_CHECK(H % KVH == 0 && H / KVH <= 8, "attn_decode: GQA group must be <= 8");
+  const int gqa = H / KVH;
+  const int max_blocks = block_tables.size(1);
+  const size_t lds = (gqa * 128 + gqa * 256 + 16 + 8 + 8) * sizeof(float);
+  hipLaunchKernelGGL(attn_decode_kernel, dim3(B, KVH), dim3(256), lds, cur_stream(),
+                     bf(out), cbf(q), cbf(k_cache), cbf(v_cache),
+                     block_tables.data_ptr<int>(), context_lens.data_ptr<int>(),
+                     (float)scale, H, KVH, D, BS, max_blocks);
+}
+
+void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                  torch::Tensor tile_seq_start, torch::Tensor tile_qpos0,
+                  torch::Tensor tile_seqlen, double scale) {
+  CHECK_BF16_CONTIG(out); CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
+  const int H = q.size(1), D = q.size(2), KVH = k.size(1);
+  TORCH_CHECK(D == 128, "attn_prefill: head_dim must be 128");
+  const int ntiles = tile_seq_start.size(0);
+  hipLaunchKernelGGL(attn_prefill_kernel, dim3(ntiles, H), dim3(64), 0, cur_stream(),
+                     bf(out), cbf(q), cbf(k), cbf(v),
+                     tile_seq_start.data_ptr<int>(), tile_qpos0.data_ptr<int>(),
+                     tile_seqlen.data_ptr<int>(), (float)scale, H, KVH);
+}
+
+void sample(torch::Tensor tokens, torch::Tensor logprobs, torch::Tensor logits,
+            torch::Tensor temperatures, torch::Tensor top_ps, torch::Tensor top_ks,
+            torch::Tensor seeds, torch::Tensor steps, torch::Tensor mask) {
+  TORCH_CHECK(logits.is_cuda() && logits.is_contiguous() && logits.scalar_type() == at::kFloat);
+  const int B = logits.size(0), V = logits.size(1);
+  const uint32_t* mptr = mask.numel() > 0
+      ? reinterpret_cast<const uint32_t*>(mask.data_ptr<int>()) : nullptr;
+  hipLaunchKernelGGL(sample_kernel, dim3(B), dim3(256), 0, cur_stream(),
+                     tokens.data_ptr<int64_t>(), logprobs.data_ptr<float>(),
+                     logits.data_ptr<float>(), temperatures.data_ptr<float>(),
+                     top_ps.data_ptr<float>(), top_ks.data_ptr<int>(),
+                     seeds.data_ptr<int64_t>(), steps.data_ptr<int64_t>(), mptr, V);
+}
+
+torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor B) {
+  // D[32,32] = A[32,16] x B[16,32], verifying the fragment maps the attention
+  // kernel assumes for v_mfma_f32_32x32x16_bf16.
+  CHECK_BF16_CONTIG(A); CHECK_BF16_CONTIG(B);
+  auto D = torch::empty({32, 32}, torch::dtype(torch::kFloat).device(A.device()));
+  hipLaunchKernelGGL(mfma_selftest_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     D.data_ptr<float>(), cbf(A), cbf(B));
+  return D;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm);
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
+  m.def("silu_mul", &silu_mul);
+  m.def("rope_inplace", &rope_inplace);
+  m.def("store_kv", &store_kv);
+  m.def("attn_decode_paged", &attn_decode_paged);
+  m.def("attn_prefill", &attn_prefill);
+  m.def("sample", &sample);
+  m.def("mfma_selftest", &mfma_selftest);
+}

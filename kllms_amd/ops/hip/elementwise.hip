@@ -1,0 +1,170 @@
+// Memory-bound elementwise / normalization kernels for gfx950.
+//
+// All of these are HBM-bandwidth-bound: the design rules applied are
+// G13 (vectorize bf16 as 8-element/16-byte lane loads), G11 (grid-stride with
+// a capped grid), and fusion (residual-add folded into RMSNorm; cos/sin for
+// RoPE precomputed on host — Appendix B: on-device trig turns memory-bound
+// into VALU-bound).
+
+#include "common.h"
+
+// --------------------------------------------------------------------------
+// RMSNorm: one block per row, row length N (multiple of 8), bf16 in/out.
+// y = x / sqrt(mean(x^2) + eps) * w      (fp32 accumulation)
+// --------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256) rmsnorm_kernel(
+    bf16_t* __restrict__ out, const bf16_t* __restrict__ in,
+    const bf16_t* __restrict__ w, int n, float eps) {
+  __shared__ float red[16];
+  const int row = blockIdx.x;
+  const bf16x8_vec* inv = reinterpret_cast<const bf16x8_vec*>(in + (int64_t)row * n);
+  bf16x8_vec* outv = reinterpret_cast<bf16x8_vec*>(out + (int64_t)row * n);
+  const bf16x8_vec* wv = reinterpret_cast<const bf16x8_vec*>(w);
+  const int nvec = n / 8;
+
+  float ss = 0.0f;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    bf16x8_vec v = inv[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf16_to_f32(v[j]);
+      ss += f * f;
+    }
+  }
+  ss = block_reduce_sum<4>(ss, red);
+  const float inv_rms = rsqrtf(ss / n + eps);
+
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    bf16x8_vec v = inv[i];
+    bf16x8_vec wv8 = wv[i];
+    bf16x8_vec o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = f32_to_bf16(bf16_to_f32(v[j]) * inv_rms * bf16_to_f32(wv8[j]));
+    outv[i] = o;
+  }
+}
+
+// --------------------------------------------------------------------------
+// Fused residual-add + RMSNorm: residual += x (written back), y = rmsnorm(residual)
+// --------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256) fused_add_rmsnorm_kernel(
+    bf16_t* __restrict__ out, const bf16_t* __restrict__ x,
+    bf16_t* __restrict__ residual, const bf16_t* __restrict__ w, int n, float eps) {
+  __shared__ float red[16];
+  const int row = blockIdx.x;
+  const bf16x8_vec* xv = reinterpret_cast<const bf16x8_vec*>(x + (int64_t)row * n);
+  bf16x8_vec* rv = reinterpret_cast<bf16x8_vec*>(residual + (int64_t)row * n);
+  bf16x8_vec* outv = reinterpret_cast<bf16x8_vec*>(out + (int64_t)row * n);
+  const bf16x8_vec* wv = reinterpret_cast<const bf16x8_vec*>(w);
+  const int nvec = n / 8;
+
+  float ss = 0.0f;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    bf16x8_vec a = xv[i];
+    bf16x8_vec b = rv[i];
+    bf16x8_vec s;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf16_to_f32(a[j]) + bf16_to_f32(b[j]);
+      s[j] = f32_to_bf16(f);
+      float fs = bf16_to_f32(s[j]);  // accumulate on the bf16-rounded sum
+      ss += fs * fs;
+    }
+    rv[i] = s;
+  }
+  ss = block_reduce_sum<4>(ss, red);
+  const float inv_rms = rsqrtf(ss / n + eps);
+
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    bf16x8_vec s = rv[i];
+    bf16x8_vec wv8 = wv[i];
+    bf16x8_vec o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = f32_to_bf16(bf16_to_f32(s[j]) * inv_rms * bf16_to_f32(wv8[j]));
+    outv[i] = o;
+  }
+}
+
+// --------------------------------------------------------------------------
+// SwiGLU activation: out = silu(gate) * up, bf16, grid-stride, 8-wide.
+// --------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256) silu_mul_kernel(
+    bf16_t* __restrict__ out, const bf16_t* __restrict__ gate,
+    const bf16_t* __restrict__ up, int64_t nvec) {
+  const bf16x8_vec* gv = reinterpret_cast<const bf16x8_vec*>(gate);
+  const bf16x8_vec* uv = reinterpret_cast<const bf16x8_vec*>(up);
+  bf16x8_vec* ov = reinterpret_cast<bf16x8_vec*>(out);
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bf16x8_vec g = gv[i];
+    bf16x8_vec u = uv[i];
+    bf16x8_vec o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf16_to_f32(g[j]);
+      float s = gf / (1.0f + __expf(-gf));
+      o[j] = f32_to_bf16(s * bf16_to_f32(u[j]));
+    }
+    ov[i] = o;
+  }
+}
+
+// --------------------------------------------------------------------------
+// RoPE (Llama rotate-half): in-place on q [T, H, D] and k [T, KVH, D].
+// cos_sin: [max_pos, D] fp32, cos in [0, D/2), sin in [D/2, D).
+// Grid: (T, H + KVH); block = D/2 threads (one lane per rotation pair).
+// --------------------------------------------------------------------------
+extern "C" __global__ void rope_kernel(
+    bf16_t* __restrict__ q, bf16_t* __restrict__ k,
+    const int64_t* __restrict__ positions, const float* __restrict__ cos_sin,
+    int num_q_heads, int num_kv_heads, int head_dim) {
+  const int t = blockIdx.x;
+  const int h = blockIdx.y;
+  const int i = threadIdx.x;           // pair index in [0, D/2)
+  const int half = head_dim / 2;
+  if (i >= half) return;
+  const int64_t pos = positions[t];
+  const float c = cos_sin[pos * head_dim + i];
+  const float s = cos_sin[pos * head_dim + half + i];
+  bf16_t* base;
+  if (h < num_q_heads) {
+    base = q + ((int64_t)t * num_q_heads + h) * head_dim;
+  } else {
+    base = k + ((int64_t)t * num_kv_heads + (h - num_q_heads)) * head_dim;
+  }
+  float x1 = bf16_to_f32(((const short*)base)[i]);
+  float x2 = bf16_to_f32(((const short*)base)[half + i]);
+  ((short*)base)[i] = f32_to_bf16(x1 * c - x2 * s);
+  ((short*)base)[half + i] = f32_to_bf16(x2 * c + x1 * s);
+}
+
+// --------------------------------------------------------------------------
+// Paged-KV scatter: k/v [T, KVH, D] -> caches [NB, KVH, BS, D] at flat slots.
+// Grid-stride over T*KVH*(D/8); 16-byte lane copies.
+// --------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256) store_kv_kernel(
+    const bf16_t* __restrict__ k, const bf16_t* __restrict__ v,
+    bf16_t* __restrict__ k_cache, bf16_t* __restrict__ v_cache,
+    const int64_t* __restrict__ slots, int num_tokens, int kv_heads,
+    int head_dim, int block_size) {
+  const int dvec = head_dim / 8;
+  const int64_t total = (int64_t)num_tokens * kv_heads * dvec;
+  for (int64_t idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int dv = idx % dvec;
+    const int g = (idx / dvec) % kv_heads;
+    const int t = idx / ((int64_t)dvec * kv_heads);
+    const int64_t slot = slots[t];
+    const int64_t blk = slot / block_size;
+    const int off = slot % block_size;
+    const bf16x8_vec* src_k =
+        reinterpret_cast<const bf16x8_vec*>(k + ((int64_t)t * kv_heads + g) * head_dim) + dv;
+    const bf16x8_vec* src_v =
+        reinterpret_cast<const bf16x8_vec*>(v + ((int64_t)t * kv_heads + g) * head_dim) + dv;
+    int64_t dst_off = ((blk * kv_heads + g) * block_size + off) * head_dim;
+    reinterpret_cast<bf16x8_vec*>(k_cache + dst_off)[dv] = *src_k;
+    reinterpret_cast<bf16x8_vec*>(v_cache + dst_off)[dv] = *src_v;
+  }
+}

@@ -1,0 +1,158 @@
+// Fused sampling + logprob kernel for gfx950.
+//
+// One 256-thread block per sequence row over the full vocab (grid-stride):
+//   pass A: masked max (constrained-decode bitmask applied here) + argmax
+//   pass B: unscaled sum-exp (for the OpenAI-style model logprob) and a
+//           256-bin histogram of TEMPERATURE-SCALED probability mass +
+//           count (for top-p / top-k truncation)
+//   pass C: derive the scaled-logit admission threshold from the histogram
+//           suffix (top-p by mass, top-k by count; bin-width resolution)
+//   pass D: Gumbel-max draw over admitted tokens — counter-based RNG
+//           hash(seed, step, token), deterministic and stream-independent.
+//
+// temperature == 0 -> greedy (argmax from pass A).
+// Top-p/top-k truncation is exact up to one histogram bin (width ~0.08 in
+// scaled-logit units over a 20-logit window) — the admitted set is the
+// smallest histogram-aligned superset of the exact nucleus.
+
+#include "common.h"
+
+#define SBINS 256
+#define SRANGE 20.0f   // scaled-logit window below max covered by the histogram
+
+extern "C" __global__ void __launch_bounds__(256) sample_kernel(
+    int64_t* __restrict__ out_tokens,     // [B]
+    float* __restrict__ out_logprobs,     // [B]
+    const float* __restrict__ logits,     // [B, V]
+    const float* __restrict__ temperatures,
+    const float* __restrict__ top_ps,
+    const int* __restrict__ top_ks,
+    const int64_t* __restrict__ seeds,
+    const int64_t* __restrict__ steps,
+    const uint32_t* __restrict__ mask,    // [B, ceil(V/32)] or nullptr
+    int V) {
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  const float* row = logits + (int64_t)b * V;
+  const int W = (V + 31) / 32;
+  const uint32_t* mrow = mask ? mask + (int64_t)b * W : nullptr;
+  const float temp = temperatures[b];
+  const float top_p = top_ps[b];
+  const int top_k = top_ks[b];
+
+  __shared__ float red[16];
+  __shared__ float hist_mass[SBINS];
+  __shared__ unsigned int hist_cnt[SBINS];
+  __shared__ float sh_thresh;
+  __shared__ int sh_argmax;
+  __shared__ unsigned long long sh_best;   // packed (gumbel-key, token)
+
+  for (int i = tid; i < SBINS; i += blockDim.x) {
+    hist_mass[i] = 0.0f;
+    hist_cnt[i] = 0;
+  }
+  if (tid == 0) {
+    sh_argmax = 0;
+    sh_best = 0;
+  }
+  __syncthreads();
+
+  // ---- pass A: masked max + argmax ---------------------------------------
+  float lmax = -INFINITY;
+  int larg = -1;
+  for (int i = tid; i < V; i += blockDim.x) {
+    if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
+    float x = row[i];
+    if (x > lmax) { lmax = x; larg = i; }
+  }
+  float gmax = block_reduce_max<4>(lmax, red);
+  if (lmax == gmax && larg >= 0) sh_argmax = larg;  // ties: any max index
+  __syncthreads();
+  const int argmax_tok = sh_argmax;
+
+  if (temp == 0.0f) {
+    // greedy — still need the unscaled logprob denominator
+    float lsum = 0.0f;
+    for (int i = tid; i < V; i += blockDim.x) {
+      if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
+      lsum += __expf(row[i] - gmax);
+    }
+    float gsum = block_reduce_sum<4>(lsum, red);
+    if (tid == 0) {
+      out_tokens[b] = argmax_tok;
+      out_logprobs[b] = row[argmax_tok] - gmax - __logf(gsum);
+    }
+    return;
+  }
+
+  // ---- pass B: unscaled sumexp + scaled-mass histogram --------------------
+  const float inv_t = 1.0f / temp;
+  const float smax = gmax * inv_t;
+  float lsum_unscaled = 0.0f;
+  float lsum_scaled = 0.0f;
+  for (int i = tid; i < V; i += blockDim.x) {
+    if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
+    const float x = row[i];
+    lsum_unscaled += __expf(x - gmax);
+    const float sx = x * inv_t - smax;            // <= 0
+    const float w = __expf(sx);
+    lsum_scaled += w;
+    int bin = (int)(-sx * (SBINS / SRANGE));
+    bin = min(bin, SBINS - 1);
+    atomicAdd(&hist_mass[bin], w);
+    atomicAdd(&hist_cnt[bin], 1u);
+  }
+  const float gsum_unscaled = block_reduce_sum<4>(lsum_unscaled, red);
+  const float gsum_scaled = block_reduce_sum<4>(lsum_scaled, red);
+
+  // ---- pass C: admission threshold ----------------------------------------
+  if (tid == 0) {
+    float need_mass = top_p < 1.0f ? top_p * gsum_scaled : INFINITY;
+    int need_cnt = (top_k > 0 && top_k < V) ? top_k : 0x7fffffff;
+    float acc_mass = 0.0f;
+    unsigned int acc_cnt = 0;
+    int cut_bin = SBINS - 1;
+    for (int bn = 0; bn < SBINS; ++bn) {
+      acc_mass += hist_mass[bn];
+      acc_cnt += hist_cnt[bn];
+      if (acc_mass >= need_mass || acc_cnt >= (unsigned)need_cnt) {
+        cut_bin = bn;
+        break;
+      }
+    }
+    // admit tokens whose scaled logit is within (cut_bin+1) bins of the max;
+    // back in UNSCALED logit space: x/temp >= smax - (cut+1)*binwidth
+    sh_thresh = (smax - (cut_bin + 1) * (SRANGE / SBINS)) * temp;
+  }
+  __syncthreads();
+  const float x_thresh = sh_thresh;
+
+  // ---- pass D: Gumbel-max over admitted tokens ----------------------------
+  const uint64_t base = splitmix64((uint64_t)seeds[b] * 0x9E3779B97F4A7C15ULL +
+                                   (uint64_t)steps[b] * 0xBF58476D1CE4E5B9ULL);
+  float best_key = -INFINITY;
+  int best_tok = argmax_tok;
+  for (int i = tid; i < V; i += blockDim.x) {
+    if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
+    const float x = row[i];
+    if (x < x_thresh) continue;
+    const float u = u64_to_uniform(splitmix64(base ^ (uint64_t)i));
+    const float gumbel = -__logf(-__logf(u));
+    const float key = x * inv_t + gumbel;
+    if (key > best_key) { best_key = key; best_tok = i; }
+  }
+  // block argmax over (key, token)
+  {
+    // pack: higher key wins; order-preserving float->uint transform
+    uint32_t kbits = __float_as_uint(best_key);
+    kbits = (kbits & 0x80000000u) ? ~kbits : (kbits | 0x80000000u);
+    unsigned long long packed = ((unsigned long long)kbits << 32) | (uint32_t)best_tok;
+    atomicMax(&sh_best, packed);
+  }
+  __syncthreads();
+  if (tid == 0) {
+    const int tok = (int)(sh_best & 0xffffffffu);
+    out_tokens[b] = tok;
+    out_logprobs[b] = row[tok] - gmax - __logf(gsum_unscaled);
+  }
+}
