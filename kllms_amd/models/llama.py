@@ -62,7 +62,7 @@ class LlamaAttention(nn.Module):
         T = hidden.shape[0]
         qkv = self.qkv_proj(hidden)
         q, k, v = qkv.split([self._q, self._kv, self._kv], dim=-1)
-        q = q.view(T, self.num_heads, self.head_dim)
+        q = q.contiguous().view(T, self.num_heads, self.head_dim)
         k = k.contiguous().view(T, self.num_kv_heads, self.head_dim)
         v = v.contiguous().view(T, self.num_kv_heads, self.head_dim)
         ops.rope_inplace(q, k, batch.positions, cos_sin)

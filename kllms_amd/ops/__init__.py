@@ -94,12 +94,30 @@ def store_kv(
     torch_ref.store_kv(k, v, k_cache, v_cache, slot_mapping)
 
 
+def _prefill_tiles(cu_seqlens: torch.Tensor, device) -> tuple:
+    """Per-32-row q-tile metadata for the MFMA prefill kernel: each tile's
+    sequence start token, row0 position within the sequence, and seq length.
+    One host round-trip per prefill batch (once per generate() call)."""
+    cu = cu_seqlens.cpu().tolist()
+    seq_start, qpos0, seqlen = [], [], []
+    for i in range(len(cu) - 1):
+        s, e = cu[i], cu[i + 1]
+        L = e - s
+        for r0 in range(0, L, 32):
+            seq_start.append(s)
+            qpos0.append(r0)
+            seqlen.append(L)
+    t = lambda x: torch.tensor(x, dtype=torch.int32, device=device)
+    return t(seq_start), t(qpos0), t(seqlen)
+
+
 def attn_prefill_varlen(
     q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, cu_seqlens: torch.Tensor, scale: float
 ) -> torch.Tensor:
     if q.is_cuda and not _force_torch():
         out = torch.empty_like(q)
-        _hip_or_raise().attn_prefill_varlen(out, q, k, v, cu_seqlens, scale)
+        ts, tq, tl = _prefill_tiles(cu_seqlens, q.device)
+        _hip_or_raise().attn_prefill(out, q, k, v, ts, tq, tl, scale)
         return out
     return torch_ref.attn_prefill_varlen(q, k, v, cu_seqlens, scale)
 

@@ -55,9 +55,9 @@ __device__ __forceinline__ float block_reduce_sum(float v, float* lds_scratch) {
   if (lane == 0) lds_scratch[wid] = v;
   __syncthreads();
   const int nwaves = (blockDim.x + WAVE - 1) / WAVE;
-  v = (threadIdx.x < nwaves) ? lds_scratch[threadIdx.x] : 0.0f;
+  // every wave reads the per-wave partials so ALL threads get the total
+  v = (lane < nwaves) ? lds_scratch[lane] : 0.0f;
   v = wave_reduce_sum(v);
-  v = __shfl(v, 0);
   __syncthreads();
   return v;
 }
@@ -70,9 +70,8 @@ __device__ __forceinline__ float block_reduce_max(float v, float* lds_scratch) {
   if (lane == 0) lds_scratch[wid] = v;
   __syncthreads();
   const int nwaves = (blockDim.x + WAVE - 1) / WAVE;
-  v = (threadIdx.x < nwaves) ? lds_scratch[threadIdx.x] : -INFINITY;
+  v = (lane < nwaves) ? lds_scratch[lane] : -INFINITY;
   v = wave_reduce_max(v);
-  v = __shfl(v, 0);
   __syncthreads();
   return v;
 }
