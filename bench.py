@@ -107,8 +107,12 @@ def main():
             for i in range(args.batch)
         ]
         completions = eng_client.chat_completions_create_many(call_params_list)
+        t_gen = time.perf_counter()
         for comp in completions:
             consolidate_chat_completions(comp, embeddings_wrapper, client=eng_client)
+        if os.environ.get("KLLMS_BENCH_VERBOSE"):
+            tm = getattr(eng_client.engine, "last_timings", {})
+            log(f"[bench] step {step_idx}: engine={tm}, consensus={1000 * (time.perf_counter() - t_gen):.1f}ms")
 
     def barrier_sync():
         if dist is not None:

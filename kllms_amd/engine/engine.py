@@ -238,14 +238,30 @@ class LLMEngine:
 
         # ---- decode loop -----------------------------------------------------
         active = [s for s in streams if not s.done]
+        n_decode_steps = 0
+        t_forward = 0.0
+        t_sample = 0.0
         while active:
+            ts = time.perf_counter()
             logits = self._decode_step(active)
+            tf = time.perf_counter()
             self._sample_and_append(logits, active)
             active = [s for s in active if not s.done]
+            t_forward += tf - ts
+            t_sample += time.perf_counter() - tf
+            n_decode_steps += 1
 
         if dev.type == "cuda":
             torch.cuda.synchronize()
         t2 = time.perf_counter()
+        self.last_timings = {
+            "prefill_ms": (t1 - t0) * 1000,
+            "decode_ms": (t2 - t1) * 1000,
+            "decode_steps": n_decode_steps,
+            "decode_forward_ms": t_forward * 1000,
+            "decode_sample_ms": t_sample * 1000,
+            "n_streams": len(streams),
+        }
 
         # ---- collect ---------------------------------------------------------
         for s in streams:
