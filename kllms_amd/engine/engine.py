@@ -70,6 +70,44 @@ class _Stream:
         self.last_token: int = -1
 
 
+class _DecodeBatchState:
+    """Device-resident decode batch: token ids, positions, context lengths,
+    block tables and sampling parameters live on the GPU and are updated in
+    place each step; rebuilt only when stream membership changes."""
+
+    def __init__(self, engine: "LLMEngine", streams: List[_Stream]):
+        dev = engine.device
+        bs = engine.kv.block_size
+        self.streams = streams
+        self.ids = torch.tensor([s.last_token for s in streams], dtype=torch.long, device=dev)
+        self.positions = torch.tensor([s.seq.num_tokens for s in streams], dtype=torch.long, device=dev)
+        self.ctx = torch.tensor([s.seq.num_tokens + 1 for s in streams], dtype=torch.int32, device=dev)
+        # block-table width with headroom so per-step growth is rare
+        cur_max = max(len(s.seq.blocks) for s in streams)
+        horizon = max(s.sampling.max_tokens or engine.config.default_max_new_tokens for s in streams)
+        self.width = cur_max + (horizon + bs - 1) // bs + 1
+        bt = torch.zeros((len(streams), self.width), dtype=torch.int32)
+        for i, s in enumerate(streams):
+            bt[i, : len(s.seq.blocks)] = torch.tensor(s.seq.blocks, dtype=torch.int32)
+        self.bt = bt.to(dev)
+        self.temps = torch.tensor([s.sampling.temperature for s in streams], dtype=torch.float32, device=dev)
+        self.top_ps = torch.tensor([s.sampling.top_p for s in streams], dtype=torch.float32, device=dev)
+        self.top_ks = torch.tensor([s.sampling.top_k for s in streams], dtype=torch.int32, device=dev)
+        self.seeds = torch.tensor([s.seed for s in streams], dtype=torch.int64, device=dev)
+        self.steps = torch.tensor([s.step for s in streams], dtype=torch.int64, device=dev)
+
+    def apply_bt_updates(self, updates: List[tuple]) -> None:
+        need = max(j for _, j, _ in updates) + 1
+        if need > self.width:
+            pad = torch.zeros((self.bt.shape[0], need + 8 - self.width), dtype=torch.int32, device=self.bt.device)
+            self.bt = torch.cat([self.bt, pad], dim=1)
+            self.width = self.bt.shape[1]
+        rows = torch.tensor([u[0] for u in updates], dtype=torch.long, device=self.bt.device)
+        cols = torch.tensor([u[1] for u in updates], dtype=torch.long, device=self.bt.device)
+        vals = torch.tensor([u[2] for u in updates], dtype=torch.int32, device=self.bt.device)
+        self.bt[rows, cols] = vals
+
+
 class LLMEngine:
     def __init__(self, config: EngineConfig, parallel_ctx=None):
         from ..parallel.tp import ParallelContext
@@ -236,17 +274,22 @@ class LLMEngine:
         ])
         self._sample_and_append(rep_logits.contiguous(), streams)
 
-        # ---- decode loop -----------------------------------------------------
+        # ---- decode loop (device-resident batch state) -----------------------
         active = [s for s in streams if not s.done]
         n_decode_steps = 0
         t_forward = 0.0
         t_sample = 0.0
+        state: Optional[_DecodeBatchState] = None
         while active:
             ts = time.perf_counter()
-            logits = self._decode_step(active)
+            if state is None:
+                state = _DecodeBatchState(self, active)
+            logits = self._decode_step_state(state)
             tf = time.perf_counter()
-            self._sample_and_append(logits, active)
-            active = [s for s in active if not s.done]
+            any_done = self._sample_state(state, logits)
+            if any_done:
+                active = [s for s in active if not s.done]
+                state = None
             t_forward += tf - ts
             t_sample += time.perf_counter() - tf
             n_decode_steps += 1
@@ -275,6 +318,63 @@ class LLMEngine:
         return outputs
 
     # --- decode internals -----------------------------------------------------
+    def _decode_step_state(self, state: "_DecodeBatchState") -> torch.Tensor:
+        """One forward over the persistent device-resident batch state: the
+        only host work per step is block-allocator bookkeeping (real work only
+        every block_size tokens)."""
+        bs = self.kv.block_size
+        bt_updates = []
+        for i, s in enumerate(state.streams):
+            p = s.seq.num_tokens
+            if p % bs == 0:
+                s.seq.blocks.append(self.kv.allocator.alloc())
+                bt_updates.append((i, len(s.seq.blocks) - 1, s.seq.blocks[-1]))
+            s.seq.num_tokens += 1
+        if bt_updates:
+            state.apply_bt_updates(bt_updates)
+
+        slots = (
+            state.bt.gather(1, (state.positions // bs).unsqueeze(1).to(torch.int64)).squeeze(1).to(torch.int64) * bs
+            + state.positions % bs
+        )
+        batch = ForwardBatch(
+            mode="decode",
+            positions=state.positions,
+            slot_mapping=slots,
+            kv_caches=self.kv.layer_caches(),
+            block_tables=state.bt,
+            context_lens=state.ctx,
+        )
+        if self._graph_runner is not None:
+            return self._graph_runner.run(state.ids, batch)
+        return self.model.forward_decode(state.ids, batch)
+
+    def _sample_state(self, state: "_DecodeBatchState", logits: torch.Tensor) -> bool:
+        streams = state.streams
+        logits = self._apply_penalties(logits, streams)
+        mask = self._constraint_mask(streams)
+        tokens, logprobs = ops.sample(
+            logits, state.temps, state.top_ps, state.top_ks, state.seeds, state.steps, mask
+        )
+        state.ids = tokens
+        state.steps += 1
+        state.positions += 1
+        state.ctx += 1
+        tokens_l = tokens.tolist()
+        logprobs_l = logprobs.tolist()
+        any_done = False
+        for i, s in enumerate(streams):
+            tok = tokens_l[i]
+            s.step += 1
+            s.last_token = tok
+            if s.constraint is not None:
+                s.constraint_state = s.constraint.advance(s.constraint_state, tok)
+            s.out.token_ids.append(tok)
+            s.out.logprobs.append(logprobs_l[i])
+            self._check_stop(s)
+            any_done |= s.done
+        return any_done
+
     def _decode_step(self, active: List[_Stream]) -> torch.Tensor:
         dev = self.device
         B = len(active)

@@ -73,16 +73,13 @@ class PagedKVCache:
         self.block_size = block_size
         self.num_blocks = num_blocks
         self.device = device
-        # One contiguous allocation per layer pair keeps pointers stable for
-        # hipGraph capture.
-        self.k_caches = [
-            torch.zeros(num_blocks, num_kv_heads, block_size, head_dim, device=device, dtype=dtype)
-            for _ in range(num_layers)
-        ]
-        self.v_caches = [
-            torch.zeros(num_blocks, num_kv_heads, block_size, head_dim, device=device, dtype=dtype)
-            for _ in range(num_layers)
-        ]
+        # One contiguous allocation for all layers keeps pointers stable for
+        # hipGraph capture and makes block copies (COW) two device ops.
+        self.k_all = torch.zeros(num_layers, num_blocks, num_kv_heads, block_size, head_dim,
+                                 device=device, dtype=dtype)
+        self.v_all = torch.zeros_like(self.k_all)
+        self.k_caches = [self.k_all[i] for i in range(num_layers)]
+        self.v_caches = [self.v_all[i] for i in range(num_layers)]
         self.allocator = BlockAllocator(num_blocks)
 
     def layer_caches(self) -> List[Tuple[torch.Tensor, torch.Tensor]]:
@@ -94,16 +91,22 @@ class PagedKVCache:
         return SequenceKV(blocks=[self.allocator.alloc() for _ in range(n_blocks)], num_tokens=num_tokens)
 
     def fork(self, parent: SequenceKV) -> SequenceKV:
-        """Share all of the parent's blocks (copy-on-write on append)."""
-        for b in parent.blocks:
-            self.allocator.incref(b)
-        return SequenceKV(blocks=list(parent.blocks), num_tokens=parent.num_tokens)
+        """Share the parent's blocks by refcount. The block the child will
+        write into next (a partially-filled tail block) is copied EAGERLY so
+        the decode hot loop never needs copy-on-write bookkeeping."""
+        child = SequenceKV(blocks=list(parent.blocks), num_tokens=parent.num_tokens)
+        tail = parent.num_tokens % self.block_size
+        for i, b in enumerate(child.blocks):
+            if i == len(child.blocks) - 1 and tail != 0:
+                child.blocks[i] = self._copy_block(b)
+            else:
+                self.allocator.incref(b)
+        return child
 
     def _copy_block(self, src: int) -> int:
         dst = self.allocator.alloc()
-        for layer in range(self.num_layers):
-            self.k_caches[layer][dst].copy_(self.k_caches[layer][src])
-            self.v_caches[layer][dst].copy_(self.v_caches[layer][src])
+        self.k_all[:, dst].copy_(self.k_all[:, src])
+        self.v_all[:, dst].copy_(self.v_all[:, src])
         return dst
 
     def append_slot(self, seq: SequenceKV) -> int:

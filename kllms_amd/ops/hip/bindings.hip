@@ -12,7 +12,8 @@ extern "C" __global__ void fused_add_rmsnorm_kernel(bf16_t*, const bf16_t*, bf16
 extern "C" __global__ void silu_mul_kernel(bf16_t*, const bf16_t*, const bf16_t*, int64_t);
 extern "C" __global__ void rope_kernel(bf16_t*, bf16_t*, const int64_t*, const float*, int, int, int);
 extern "C" __global__ void store_kv_kernel(const bf16_t*, const bf16_t*, bf16_t*, bf16_t*, const int64_t*, int, int, int, int);
-extern "C" __global__ void attn_decode_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, float, int, int, int, int, int);
+extern "C" __global__ void attn_decode_partial_kernel(float*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, float, int, int, int, int, int);
+extern "C" __global__ void attn_decode_reduce_kernel(bf16_t*, const float*, const int*, int, int, int);
 extern "C" __global__ void attn_prefill_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, const int*, float, int, int);
 extern "C" __global__ void sample_kernel(int64_t*, float*, const float*, const float*, const float*, const int*, const int64_t*, const int64_t*, const uint32_t*, int);
 extern "C" __global__ void mfma_selftest_kernel(float*, const bf16_t*, const bf16_t*);
@@ -86,11 +87,20 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache
   TORCH_CHECK(H % KVH == 0 && H / KVH <= 8, "attn_decode: GQA group must be <= 8");
   const int gqa = H / KVH;
   const int max_blocks = block_tables.size(1);
-  const size_t lds = (gqa * 128 + gqa * 256 + 16 + 8 + 8) * sizeof(float);
-  hipLaunchKernelGGL(attn_decode_kernel, dim3(B, KVH), dim3(256), lds, cur_stream(),
-                     bf(out), cbf(q), cbf(k_cache), cbf(v_cache),
+  const int CHUNK_KEYS = 512;  // keep in sync with CHUNK in attn_decode.hip
+  const int TILE_KEYS = 128;   // keep in sync with TKV
+  const int max_chunks = std::max(1, (max_blocks * BS + CHUNK_KEYS - 1) / CHUNK_KEYS);
+  auto partials = torch::empty({(int64_t)B * KVH * max_chunks * gqa * 130},
+                               torch::dtype(torch::kFloat).device(q.device()));
+  const size_t lds = (gqa * 128 + gqa * TILE_KEYS + 16 + 8 + 8) * sizeof(float) +
+                     TILE_KEYS * sizeof(int64_t);
+  hipLaunchKernelGGL(attn_decode_partial_kernel, dim3(B, KVH, max_chunks), dim3(256), lds, cur_stream(),
+                     partials.data_ptr<float>(), cbf(q), cbf(k_cache), cbf(v_cache),
                      block_tables.data_ptr<int>(), context_lens.data_ptr<int>(),
-                     (float)scale, H, KVH, D, BS, max_blocks);
+                     (float)scale, H, KVH, BS, max_blocks, max_chunks);
+  hipLaunchKernelGGL(attn_decode_reduce_kernel, dim3(B, H), dim3(64), 0, cur_stream(),
+                     bf(out), partials.data_ptr<float>(), context_lens.data_ptr<int>(),
+                     H, KVH, max_chunks);
 }
 
 void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k, torch::Tensor v,
@@ -113,7 +123,8 @@ void sample(torch::Tensor tokens, torch::Tensor logprobs, torch::Tensor logits,
   const int B = logits.size(0), V = logits.size(1);
   const uint32_t* mptr = mask.numel() > 0
       ? reinterpret_cast<const uint32_t*>(mask.data_ptr<int>()) : nullptr;
-  hipLaunchKernelGGL(sample_kernel, dim3(B), dim3(256), 0, cur_stream(),
+  const size_t lds = 16 * 256 * (sizeof(float) + sizeof(unsigned int));  // per-wave histograms
+  hipLaunchKernelGGL(sample_kernel, dim3(B), dim3(1024), lds, cur_stream(),
                      tokens.data_ptr<int64_t>(), logprobs.data_ptr<float>(),
                      logits.data_ptr<float>(), temperatures.data_ptr<float>(),
                      top_ps.data_ptr<float>(), top_ks.data_ptr<int>(),

@@ -1,26 +1,28 @@
 // Fused sampling + logprob kernel for gfx950.
 //
-// One 256-thread block per sequence row over the full vocab (grid-stride):
+// One 1024-thread block per sequence row over the full vocab (grid-stride):
 //   pass A: masked max (constrained-decode bitmask applied here) + argmax
-//   pass B: unscaled sum-exp (for the OpenAI-style model logprob) and a
-//           256-bin histogram of TEMPERATURE-SCALED probability mass +
-//           count (for top-p / top-k truncation)
-//   pass C: derive the scaled-logit admission threshold from the histogram
-//           suffix (top-p by mass, top-k by count; bin-width resolution)
+//   pass B: unscaled sum-exp (for the OpenAI-style model logprob); when top-p
+//           or top-k truncation is requested, also a 256-bin PER-WAVE
+//           histogram of temperature-scaled probability mass + count
+//           (per-wave copies kill the LDS same-bin atomic serialization that
+//           made a single shared histogram ~50x slower)
+//   pass C: admission threshold from the histogram suffix (top-p by mass,
+//           top-k by count; exact up to one bin width)
 //   pass D: Gumbel-max draw over admitted tokens — counter-based RNG
-//           hash(seed, step, token), deterministic and stream-independent.
+//           hash(seed, step, token): deterministic, stream-ordered, and
+//           identical regardless of batch composition.
 //
-// temperature == 0 -> greedy (argmax from pass A).
-// Top-p/top-k truncation is exact up to one histogram bin (width ~0.08 in
-// scaled-logit units over a 20-logit window) — the admitted set is the
-// smallest histogram-aligned superset of the exact nucleus.
+// temperature == 0 -> greedy (argmax + logprob only).
 
 #include "common.h"
 
+#define NT 1024
+#define NW (NT / WAVE)     // 16 waves
 #define SBINS 256
-#define SRANGE 20.0f   // scaled-logit window below max covered by the histogram
+#define SRANGE 20.0f       // scaled-logit window below max covered by the histogram
 
-extern "C" __global__ void __launch_bounds__(256) sample_kernel(
+extern "C" __global__ void __launch_bounds__(NT) sample_kernel(
     int64_t* __restrict__ out_tokens,     // [B]
     float* __restrict__ out_logprobs,     // [B]
     const float* __restrict__ logits,     // [B, V]
@@ -33,106 +35,117 @@ extern "C" __global__ void __launch_bounds__(256) sample_kernel(
     int V) {
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid / WAVE;
   const float* row = logits + (int64_t)b * V;
-  const int W = (V + 31) / 32;
-  const uint32_t* mrow = mask ? mask + (int64_t)b * W : nullptr;
+  const uint32_t* mrow = mask ? mask + (int64_t)b * ((V + 31) / 32) : nullptr;
   const float temp = temperatures[b];
   const float top_p = top_ps[b];
   const int top_k = top_ks[b];
+  const bool truncate = (top_p < 1.0f) || (top_k > 0 && top_k < V);
 
   __shared__ float red[16];
-  __shared__ float hist_mass[SBINS];
-  __shared__ unsigned int hist_cnt[SBINS];
   __shared__ float sh_thresh;
   __shared__ int sh_argmax;
-  __shared__ unsigned long long sh_best;   // packed (gumbel-key, token)
+  __shared__ unsigned long long sh_best;
+  extern __shared__ __attribute__((aligned(16))) char hist_smem[];
+  float* hist_mass = reinterpret_cast<float*>(hist_smem);              // [NW][SBINS]
+  unsigned int* hist_cnt = reinterpret_cast<unsigned int*>(hist_mass + NW * SBINS);
 
-  for (int i = tid; i < SBINS; i += blockDim.x) {
-    hist_mass[i] = 0.0f;
-    hist_cnt[i] = 0;
+  if (truncate) {
+    for (int i = tid; i < NW * SBINS; i += NT) {
+      hist_mass[i] = 0.0f;
+      hist_cnt[i] = 0;
+    }
   }
-  if (tid == 0) {
-    sh_argmax = 0;
-    sh_best = 0;
-  }
+  if (tid == 0) { sh_argmax = 0; sh_best = 0; }
   __syncthreads();
 
   // ---- pass A: masked max + argmax ---------------------------------------
   float lmax = -INFINITY;
   int larg = -1;
-  for (int i = tid; i < V; i += blockDim.x) {
+  for (int i = tid; i < V; i += NT) {
     if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
-    float x = row[i];
+    const float x = row[i];
     if (x > lmax) { lmax = x; larg = i; }
   }
-  float gmax = block_reduce_max<4>(lmax, red);
-  if (lmax == gmax && larg >= 0) sh_argmax = larg;  // ties: any max index
+  const float gmax = block_reduce_max<16>(lmax, red);
+  if (lmax == gmax && larg >= 0) sh_argmax = larg;
   __syncthreads();
   const int argmax_tok = sh_argmax;
 
-  if (temp == 0.0f) {
-    // greedy — still need the unscaled logprob denominator
-    float lsum = 0.0f;
-    for (int i = tid; i < V; i += blockDim.x) {
-      if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
-      lsum += __expf(row[i] - gmax);
+  // ---- pass B: unscaled sumexp (+ optional scaled-mass histogram) ---------
+  const float inv_t = (temp > 0.0f) ? 1.0f / temp : 0.0f;
+  const float smax = gmax * inv_t;
+  float lsum_unscaled = 0.0f;
+  float lsum_scaled = 0.0f;
+  float* my_mass = hist_mass + wid * SBINS;
+  unsigned int* my_cnt = hist_cnt + wid * SBINS;
+  for (int i = tid; i < V; i += NT) {
+    if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
+    const float x = row[i];
+    lsum_unscaled += __expf(x - gmax);
+    if (truncate) {
+      const float sx = x * inv_t - smax;
+      const float w = __expf(sx);
+      lsum_scaled += w;
+      int bin = (int)(-sx * (SBINS / SRANGE));
+      bin = min(bin, SBINS - 1);
+      atomicAdd(&my_mass[bin], w);
+      atomicAdd(&my_cnt[bin], 1u);
     }
-    float gsum = block_reduce_sum<4>(lsum, red);
+  }
+  const float gsum_unscaled = block_reduce_sum<16>(lsum_unscaled, red);
+
+  if (temp == 0.0f) {
     if (tid == 0) {
       out_tokens[b] = argmax_tok;
-      out_logprobs[b] = row[argmax_tok] - gmax - __logf(gsum);
+      out_logprobs[b] = row[argmax_tok] - gmax - __logf(gsum_unscaled);
     }
     return;
   }
 
-  // ---- pass B: unscaled sumexp + scaled-mass histogram --------------------
-  const float inv_t = 1.0f / temp;
-  const float smax = gmax * inv_t;
-  float lsum_unscaled = 0.0f;
-  float lsum_scaled = 0.0f;
-  for (int i = tid; i < V; i += blockDim.x) {
-    if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
-    const float x = row[i];
-    lsum_unscaled += __expf(x - gmax);
-    const float sx = x * inv_t - smax;            // <= 0
-    const float w = __expf(sx);
-    lsum_scaled += w;
-    int bin = (int)(-sx * (SBINS / SRANGE));
-    bin = min(bin, SBINS - 1);
-    atomicAdd(&hist_mass[bin], w);
-    atomicAdd(&hist_cnt[bin], 1u);
-  }
-  const float gsum_unscaled = block_reduce_sum<4>(lsum_unscaled, red);
-  const float gsum_scaled = block_reduce_sum<4>(lsum_scaled, red);
-
   // ---- pass C: admission threshold ----------------------------------------
-  if (tid == 0) {
-    float need_mass = top_p < 1.0f ? top_p * gsum_scaled : INFINITY;
-    int need_cnt = (top_k > 0 && top_k < V) ? top_k : 0x7fffffff;
-    float acc_mass = 0.0f;
-    unsigned int acc_cnt = 0;
-    int cut_bin = SBINS - 1;
-    for (int bn = 0; bn < SBINS; ++bn) {
-      acc_mass += hist_mass[bn];
-      acc_cnt += hist_cnt[bn];
-      if (acc_mass >= need_mass || acc_cnt >= (unsigned)need_cnt) {
-        cut_bin = bn;
-        break;
+  float x_thresh = -INFINITY;
+  if (truncate) {
+    const float gsum_scaled = block_reduce_sum<16>(lsum_scaled, red);
+    // fold the per-wave histograms: thread j < SBINS owns bin j
+    if (tid < SBINS) {
+      float m = 0.0f;
+      unsigned int c = 0;
+#pragma unroll 4
+      for (int w = 0; w < NW; ++w) {
+        m += hist_mass[w * SBINS + tid];
+        c += hist_cnt[w * SBINS + tid];
       }
+      hist_mass[tid] = m;
+      hist_cnt[tid] = c;
     }
-    // admit tokens whose scaled logit is within (cut_bin+1) bins of the max;
-    // back in UNSCALED logit space: x/temp >= smax - (cut+1)*binwidth
-    sh_thresh = (smax - (cut_bin + 1) * (SRANGE / SBINS)) * temp;
+    __syncthreads();
+    if (tid == 0) {
+      const float need_mass = top_p < 1.0f ? top_p * gsum_scaled : INFINITY;
+      const int need_cnt = (top_k > 0 && top_k < V) ? top_k : 0x7fffffff;
+      float acc_mass = 0.0f;
+      unsigned int acc_cnt = 0;
+      int cut_bin = SBINS - 1;
+      for (int bn = 0; bn < SBINS; ++bn) {
+        acc_mass += hist_mass[bn];
+        acc_cnt += hist_cnt[bn];
+        if (acc_mass >= need_mass || acc_cnt >= (unsigned)need_cnt) { cut_bin = bn; break; }
+      }
+      // admit tokens whose scaled logit is within (cut_bin+1) bins of the max
+      sh_thresh = (smax - (cut_bin + 1) * (SRANGE / SBINS)) * temp;
+    }
+    __syncthreads();
+    x_thresh = sh_thresh;
   }
-  __syncthreads();
-  const float x_thresh = sh_thresh;
 
   // ---- pass D: Gumbel-max over admitted tokens ----------------------------
   const uint64_t base = splitmix64((uint64_t)seeds[b] * 0x9E3779B97F4A7C15ULL +
                                    (uint64_t)steps[b] * 0xBF58476D1CE4E5B9ULL);
   float best_key = -INFINITY;
   int best_tok = argmax_tok;
-  for (int i = tid; i < V; i += blockDim.x) {
+  for (int i = tid; i < V; i += NT) {
     if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
     const float x = row[i];
     if (x < x_thresh) continue;
@@ -141,12 +154,10 @@ extern "C" __global__ void __launch_bounds__(256) sample_kernel(
     const float key = x * inv_t + gumbel;
     if (key > best_key) { best_key = key; best_tok = i; }
   }
-  // block argmax over (key, token)
   {
-    // pack: higher key wins; order-preserving float->uint transform
     uint32_t kbits = __float_as_uint(best_key);
     kbits = (kbits & 0x80000000u) ? ~kbits : (kbits | 0x80000000u);
-    unsigned long long packed = ((unsigned long long)kbits << 32) | (uint32_t)best_tok;
+    const unsigned long long packed = ((unsigned long long)kbits << 32) | (uint32_t)best_tok;
     atomicMax(&sh_best, packed);
   }
   __syncthreads();

@@ -79,22 +79,39 @@ class TestPagedDecodeNumerics:
 
 
 class TestKVFork:
-    def test_fork_refcounts_and_cow(self, engine):
+    def test_fork_refcounts_and_eager_tail_copy(self, engine):
         kv: PagedKVCache = engine.kv
         free0 = kv.allocator.num_free
         parent = kv.alloc_sequence(kv.block_size + 3)  # 2 blocks, second partial
         child1 = kv.fork(parent)
         child2 = kv.fork(parent)
+        # full blocks shared by refcount; partial tail block copied eagerly
         assert kv.allocator.refcount(parent.blocks[0]) == 3
-        # appending to a forked child copies the shared partial block
-        slot = kv.append_slot(child1)
+        assert child1.blocks[0] == parent.blocks[0]
         assert child1.blocks[1] != parent.blocks[1]
-        assert kv.allocator.refcount(parent.blocks[1]) == 2  # parent + child2
+        assert child2.blocks[1] != parent.blocks[1]
+        assert kv.allocator.refcount(parent.blocks[1]) == 1
+        assert kv.allocator.refcount(child1.blocks[1]) == 1
+        # appends never trigger COW after an eager fork
+        kv.append_slot(child1)
         assert kv.allocator.refcount(child1.blocks[1]) == 1
         kv.free_sequence(parent)
         kv.free_sequence(child1)
         kv.free_sequence(child2)
         assert kv.allocator.num_free == free0
+
+    def test_fork_at_block_boundary_shares_all(self, engine):
+        kv: PagedKVCache = engine.kv
+        parent = kv.alloc_sequence(kv.block_size)  # exactly one full block
+        child = kv.fork(parent)
+        assert child.blocks == parent.blocks
+        assert kv.allocator.refcount(parent.blocks[0]) == 2
+        # child's first append opens a fresh block
+        kv.append_slot(child)
+        assert len(child.blocks) == 2
+        assert kv.allocator.refcount(child.blocks[1]) == 1
+        kv.free_sequence(parent)
+        kv.free_sequence(child)
 
     def test_no_block_leak_after_generate(self, engine):
         free0 = engine.kv.allocator.num_free
