@@ -61,10 +61,12 @@ class LlamaAttention(nn.Module):
     def forward(self, hidden: torch.Tensor, batch: ForwardBatch, cos_sin: torch.Tensor) -> torch.Tensor:
         T = hidden.shape[0]
         qkv = self.qkv_proj(hidden)
+        # strided views into the fused QKV row — the HIP kernels take the
+        # token stride, so no contiguity copies on the hot path
         q, k, v = qkv.split([self._q, self._kv, self._kv], dim=-1)
-        q = q.contiguous().view(T, self.num_heads, self.head_dim)
-        k = k.contiguous().view(T, self.num_kv_heads, self.head_dim)
-        v = v.contiguous().view(T, self.num_kv_heads, self.head_dim)
+        q = q.view(T, self.num_heads, self.head_dim)
+        k = k.view(T, self.num_kv_heads, self.head_dim)
+        v = v.view(T, self.num_kv_heads, self.head_dim)
         ops.rope_inplace(q, k, batch.positions, cos_sin)
 
         k_cache, v_cache = batch.kv_caches[self.layer_idx]
@@ -89,7 +91,7 @@ class LlamaMLP(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         gate_up = self.gate_up_proj(x)
         gate, up = gate_up.split([self._i, self._i], dim=-1)
-        return self.down_proj(ops.silu_mul(gate.contiguous(), up.contiguous()))
+        return self.down_proj(ops.silu_mul(gate, up))
 
 
 class RMSNormModule(nn.Module):

@@ -72,7 +72,7 @@ class MixtralMoE(nn.Module):
             xe = x[toks]
             gu = torch.nn.functional.linear(xe, self.w_gate_up[e])
             g, u = gu.split([self.I, self.I], dim=-1)
-            ye = torch.nn.functional.linear(ops.silu_mul(g.contiguous(), u.contiguous()), self.w_down[e])
+            ye = torch.nn.functional.linear(ops.silu_mul(g, u), self.w_down[e])
             out.index_add_(0, toks, ye * flat_w[sel].unsqueeze(-1).to(ye.dtype))
         return self.ctx.all_reduce(out)
 

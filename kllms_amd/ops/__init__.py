@@ -75,7 +75,7 @@ def rope_inplace(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor, cos_
 
 def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     if gate.is_cuda and not _force_torch():
-        out = torch.empty_like(gate)
+        out = torch.empty(gate.shape, dtype=gate.dtype, device=gate.device)
         _hip_or_raise().silu_mul(out, gate, up)
         return out
     return torch_ref.silu_mul(gate, up)
@@ -115,7 +115,7 @@ def attn_prefill_varlen(
     q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, cu_seqlens: torch.Tensor, scale: float
 ) -> torch.Tensor:
     if q.is_cuda and not _force_torch():
-        out = torch.empty_like(q)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         ts, tq, tl = _prefill_tiles(cu_seqlens, q.device)
         _hip_or_raise().attn_prefill(out, q, k, v, ts, tq, tl, scale)
         return out
@@ -131,7 +131,7 @@ def attn_decode_paged(
     scale: float,
 ) -> torch.Tensor:
     if q.is_cuda and not _force_torch():
-        out = torch.empty_like(q)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _hip_or_raise().attn_decode_paged(out, q, k_cache, v_cache, block_tables, context_lens, scale)
         return out
     return torch_ref.attn_decode_paged(q, k_cache, v_cache, block_tables, context_lens, scale)

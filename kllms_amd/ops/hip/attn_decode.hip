@@ -19,20 +19,20 @@
 #define TKV 128            // keys per LDS tile
 #define NTHREADS 256
 #define KLANES 8           // lanes cooperating on one key's dot product
-#define CHUNK 512          // keys per workgroup (split-K granule)
+#define CHUNK 256          // keys per workgroup (split-K granule)
 
 // partials layout: [B, KVH, max_chunks, gqa, 130]: 128 o values + m + l
 #define PART_STRIDE 130
 
 extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kernel(
     float* __restrict__ partials,
-    const bf16_t* __restrict__ q,        // [B, H, 128]
+    const bf16_t* __restrict__ q,        // [B, H, 128], row stride q_tstride
     const bf16_t* __restrict__ k_cache,  // [NB, KVH, BS, 128]
     const bf16_t* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [B, max_blocks]
     const int* __restrict__ context_lens,  // [B]
     float scale, int num_q_heads, int num_kv_heads,
-    int block_size, int max_blocks, int max_chunks) {
+    int block_size, int max_blocks, int max_chunks, int q_tstride) {
   const int b = blockIdx.x;
   const int g_kv = blockIdx.y;
   const int chunk = blockIdx.z;
@@ -55,7 +55,7 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
 
   for (int i = tid; i < gqa * 128; i += NTHREADS) {
     const int g = i >> 7, d = i & 127;
-    const bf16_t* qp = q + (((int64_t)b * num_q_heads) + g_kv * gqa + g) * 128;
+    const bf16_t* qp = q + (int64_t)b * q_tstride + (g_kv * gqa + g) * 128;
     q_lds[i] = bf16_to_f32(((const short*)qp)[d]) * scale;
   }
   if (tid < 8) { mstate[tid] = -INFINITY; lstate[tid] = 0.0f; }
@@ -137,7 +137,24 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
       float a0 = o_acc[slot][0] * alpha;
       float a1 = o_acc[slot][1] * alpha;
       const float* sg = s_lds + g * TKV;
-      for (int key = 0; key < nkeys; ++key) {
+      // unroll-by-8 with all loads issued before use: the serial version is
+      // a per-key latency chain (~300 cy x nkeys); 8 independent loads in
+      // flight hide it.
+      int key = 0;
+      for (; key + 8 <= nkeys; key += 8) {
+        uint32_t pairs[8];
+#pragma unroll
+        for (int u = 0; u < 8; ++u)
+          pairs[u] = *reinterpret_cast<const uint32_t*>(
+              (const short*)(v_cache + rowoff[key + u]) + d0);
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const float p = sg[key + u];
+          a0 += p * bf16_to_f32((short)(pairs[u] & 0xffff));
+          a1 += p * bf16_to_f32((short)(pairs[u] >> 16));
+        }
+      }
+      for (; key < nkeys; ++key) {
         const float p = sg[key];
         const uint32_t pair = *reinterpret_cast<const uint32_t*>(
             (const short*)(v_cache + rowoff[key]) + d0);

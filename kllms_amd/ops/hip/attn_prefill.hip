@@ -44,7 +44,7 @@ extern "C" __global__ void __launch_bounds__(64) attn_prefill_kernel(
     const int* __restrict__ tile_seq_start,  // per q-tile: first token of seq
     const int* __restrict__ tile_qpos0,     // per q-tile: row0 position in seq
     const int* __restrict__ tile_seqlen,    // per q-tile: sequence length
-    float scale, int num_q_heads, int num_kv_heads) {
+    float scale, int num_q_heads, int num_kv_heads, int q_tstride, int kv_tstride) {
   const int tile = blockIdx.x;
   const int h = blockIdx.y;
   const int g_kv = h / (num_q_heads / num_kv_heads);
@@ -67,7 +67,7 @@ extern "C" __global__ void __launch_bounds__(64) attn_prefill_kernel(
   bf16x8_mfma q_frag[8];
   {
     const int qrow_clamped = min(my_qpos, seqlen - 1);   // tail rows clamped (never written out)
-    const bf16_t* qp = q + (((int64_t)(seq_start + qrow_clamped)) * num_q_heads + h) * DHEAD;
+    const bf16_t* qp = q + (int64_t)(seq_start + qrow_clamped) * q_tstride + h * DHEAD;
 #pragma unroll
     for (int s = 0; s < 8; ++s)
       q_frag[s] = *reinterpret_cast<const bf16x8_mfma*>((const short*)qp + s * 16 + half * 8);
@@ -86,7 +86,7 @@ extern "C" __global__ void __launch_bounds__(64) attn_prefill_kernel(
     {
       const int key = col;
       const bool valid = key < nkeys;
-      const bf16_t* kp = k + (((int64_t)(seq_start + kt0 + min(key, nkeys - 1))) * num_kv_heads + g_kv) * DHEAD;
+      const bf16_t* kp = k + (int64_t)(seq_start + kt0 + min(key, nkeys - 1)) * kv_tstride + g_kv * DHEAD;
 #pragma unroll
       for (int c = 0; c < 8; ++c) {
         const int byte_off = half * 128 + c * 16;
@@ -97,7 +97,7 @@ extern "C" __global__ void __launch_bounds__(64) attn_prefill_kernel(
             k_lds + key * K_ROW_BYTES + (byte_off ^ ((key & 15) << 4))) = val;
       }
       // V^T: lane loads V[key][half*64 .. half*64+63] and scatters b16 writes
-      const bf16_t* vp = v + (((int64_t)(seq_start + kt0 + min(key, nkeys - 1))) * num_kv_heads + g_kv) * DHEAD;
+      const bf16_t* vp = v + (int64_t)(seq_start + kt0 + min(key, nkeys - 1)) * kv_tstride + g_kv * DHEAD;
 #pragma unroll
       for (int c = 0; c < 8; ++c) {
         bf16x8_vec val = valid

@@ -9,17 +9,23 @@
 // kernels (defined in the sibling .hip TUs)
 extern "C" __global__ void rmsnorm_kernel(bf16_t*, const bf16_t*, const bf16_t*, int, float);
 extern "C" __global__ void fused_add_rmsnorm_kernel(bf16_t*, const bf16_t*, bf16_t*, const bf16_t*, int, float);
-extern "C" __global__ void silu_mul_kernel(bf16_t*, const bf16_t*, const bf16_t*, int64_t);
-extern "C" __global__ void rope_kernel(bf16_t*, bf16_t*, const int64_t*, const float*, int, int, int);
-extern "C" __global__ void store_kv_kernel(const bf16_t*, const bf16_t*, bf16_t*, bf16_t*, const int64_t*, int, int, int, int);
-extern "C" __global__ void attn_decode_partial_kernel(float*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, float, int, int, int, int, int);
+extern "C" __global__ void silu_mul_kernel(bf16_t*, const bf16_t*, const bf16_t*, int64_t, int, int);
+extern "C" __global__ void rope_kernel(bf16_t*, bf16_t*, const int64_t*, const float*, int, int, int, int, int);
+extern "C" __global__ void store_kv_kernel(const bf16_t*, const bf16_t*, bf16_t*, bf16_t*, const int64_t*, int, int, int, int, int);
+extern "C" __global__ void attn_decode_partial_kernel(float*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, float, int, int, int, int, int, int);
 extern "C" __global__ void attn_decode_reduce_kernel(bf16_t*, const float*, const int*, int, int, int);
-extern "C" __global__ void attn_prefill_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, const int*, float, int, int);
+extern "C" __global__ void attn_prefill_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, const int*, float, int, int, int, int);
 extern "C" __global__ void sample_kernel(int64_t*, float*, const float*, const float*, const float*, const int*, const int64_t*, const int64_t*, const uint32_t*, int);
 extern "C" __global__ void mfma_selftest_kernel(float*, const bf16_t*, const bf16_t*);
 
 #define CHECK_BF16_CONTIG(t) \
   TORCH_CHECK((t).is_cuda() && (t).is_contiguous() && (t).scalar_type() == at::kBFloat16, #t " must be contiguous bf16 on GPU")
+
+// [T, H, D] activation views: head/dim dims packed, token stride free (so the
+// fused-QKV splits need no .contiguous() copies)
+#define CHECK_BF16_ROWS(t) \
+  TORCH_CHECK((t).is_cuda() && (t).scalar_type() == at::kBFloat16 && (t).dim() == 3 && \
+              (t).stride(2) == 1 && (t).stride(1) == (t).size(2), #t " must be a bf16 [T,H,D] row view")
 
 static inline hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStream().stream();
@@ -48,39 +54,44 @@ void fused_add_rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor residua
 }
 
 void silu_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up) {
-  CHECK_BF16_CONTIG(out); CHECK_BF16_CONTIG(gate); CHECK_BF16_CONTIG(up);
+  CHECK_BF16_CONTIG(out);
+  TORCH_CHECK(gate.is_cuda() && gate.scalar_type() == at::kBFloat16 && gate.dim() == 2 &&
+              gate.stride(1) == 1, "gate must be a bf16 [T, I] row view");
+  TORCH_CHECK(up.sizes() == gate.sizes() && up.stride(0) == gate.stride(0) && up.stride(1) == 1);
+  const int64_t ncols = gate.size(1);
+  TORCH_CHECK(ncols % 8 == 0 && gate.stride(0) % 8 == 0);
   const int64_t nvec = gate.numel() / 8;
-  TORCH_CHECK(gate.numel() % 8 == 0, "numel must be a multiple of 8");
   const int grid = (int)std::min<int64_t>((nvec + 255) / 256, 2048);
   hipLaunchKernelGGL(silu_mul_kernel, dim3(grid), dim3(256), 0, cur_stream(),
-                     bf(out), cbf(gate), cbf(up), nvec);
+                     bf(out), cbf(gate), cbf(up), nvec, (int)(ncols / 8), (int)(gate.stride(0) / 8));
 }
 
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions, torch::Tensor cos_sin) {
-  CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k);
+  CHECK_BF16_ROWS(q); CHECK_BF16_ROWS(k);
   TORCH_CHECK(positions.scalar_type() == at::kLong && cos_sin.scalar_type() == at::kFloat);
   const int T = q.size(0), H = q.size(1), D = q.size(2), KVH = k.size(1);
   hipLaunchKernelGGL(rope_kernel, dim3(T, H + KVH), dim3(D / 2), 0, cur_stream(),
                      bf(q), bf(k), positions.data_ptr<int64_t>(), cos_sin.data_ptr<float>(),
-                     H, KVH, D);
+                     H, KVH, D, (int)q.stride(0), (int)k.stride(0));
 }
 
 void store_kv(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache, torch::Tensor v_cache,
               torch::Tensor slot_mapping) {
-  CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v); CHECK_BF16_CONTIG(k_cache); CHECK_BF16_CONTIG(v_cache);
+  CHECK_BF16_ROWS(k); CHECK_BF16_ROWS(v); CHECK_BF16_CONTIG(k_cache); CHECK_BF16_CONTIG(v_cache);
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share the token stride");
   const int T = k.size(0), KVH = k.size(1), D = k.size(2);
   const int BS = k_cache.size(2);
   const int64_t total = (int64_t)T * KVH * (D / 8);
   const int grid = (int)std::min<int64_t>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(store_kv_kernel, dim3(grid), dim3(256), 0, cur_stream(),
                      cbf(k), cbf(v), bf(k_cache), bf(v_cache),
-                     slot_mapping.data_ptr<int64_t>(), T, KVH, D, BS);
+                     slot_mapping.data_ptr<int64_t>(), T, KVH, D, BS, (int)k.stride(0));
 }
 
 void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                        torch::Tensor v_cache, torch::Tensor block_tables,
                        torch::Tensor context_lens, double scale) {
-  CHECK_BF16_CONTIG(out); CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k_cache); CHECK_BF16_CONTIG(v_cache);
+  CHECK_BF16_CONTIG(out); CHECK_BF16_ROWS(q); CHECK_BF16_CONTIG(k_cache); CHECK_BF16_CONTIG(v_cache);
   const int B = q.size(0), H = q.size(1), D = q.size(2);
   const int KVH = k_cache.size(1), BS = k_cache.size(2);
   TORCH_CHECK(D == 128, "attn_decode: head_dim must be 128");
@@ -97,7 +108,7 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache
   hipLaunchKernelGGL(attn_decode_partial_kernel, dim3(B, KVH, max_chunks), dim3(256), lds, cur_stream(),
                      partials.data_ptr<float>(), cbf(q), cbf(k_cache), cbf(v_cache),
                      block_tables.data_ptr<int>(), context_lens.data_ptr<int>(),
-                     (float)scale, H, KVH, BS, max_blocks, max_chunks);
+                     (float)scale, H, KVH, BS, max_blocks, max_chunks, (int)q.stride(0));
   hipLaunchKernelGGL(attn_decode_reduce_kernel, dim3(B, H), dim3(64), 0, cur_stream(),
                      bf(out), partials.data_ptr<float>(), context_lens.data_ptr<int>(),
                      H, KVH, max_chunks);
@@ -106,14 +117,16 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache
 void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k, torch::Tensor v,
                   torch::Tensor tile_seq_start, torch::Tensor tile_qpos0,
                   torch::Tensor tile_seqlen, double scale) {
-  CHECK_BF16_CONTIG(out); CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
+  CHECK_BF16_CONTIG(out); CHECK_BF16_ROWS(q); CHECK_BF16_ROWS(k); CHECK_BF16_ROWS(v);
   const int H = q.size(1), D = q.size(2), KVH = k.size(1);
   TORCH_CHECK(D == 128, "attn_prefill: head_dim must be 128");
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share the token stride");
   const int ntiles = tile_seq_start.size(0);
   hipLaunchKernelGGL(attn_prefill_kernel, dim3(ntiles, H), dim3(64), 0, cur_stream(),
                      bf(out), cbf(q), cbf(k), cbf(v),
                      tile_seq_start.data_ptr<int>(), tile_qpos0.data_ptr<int>(),
-                     tile_seqlen.data_ptr<int>(), (float)scale, H, KVH);
+                     tile_seqlen.data_ptr<int>(), (float)scale, H, KVH,
+                     (int)q.stride(0), (int)k.stride(0));
 }
 
 void sample(torch::Tensor tokens, torch::Tensor logprobs, torch::Tensor logits,

@@ -92,14 +92,17 @@ extern "C" __global__ void __launch_bounds__(256) fused_add_rmsnorm_kernel(
 // --------------------------------------------------------------------------
 extern "C" __global__ void __launch_bounds__(256) silu_mul_kernel(
     bf16_t* __restrict__ out, const bf16_t* __restrict__ gate,
-    const bf16_t* __restrict__ up, int64_t nvec) {
-  const bf16x8_vec* gv = reinterpret_cast<const bf16x8_vec*>(gate);
-  const bf16x8_vec* uv = reinterpret_cast<const bf16x8_vec*>(up);
+    const bf16_t* __restrict__ up, int64_t nvec, int ncols_vec, int row_stride_vec) {
+  // gate/up are [rows, ncols] row-views with a free row stride (the fused
+  // gate_up GEMM output), out is contiguous [rows, ncols]
   bf16x8_vec* ov = reinterpret_cast<bf16x8_vec*>(out);
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (int64_t)gridDim.x * blockDim.x) {
-    bf16x8_vec g = gv[i];
-    bf16x8_vec u = uv[i];
+    const int64_t row = i / ncols_vec;
+    const int64_t col = i % ncols_vec;
+    const int64_t src = row * row_stride_vec + col;
+    bf16x8_vec g = reinterpret_cast<const bf16x8_vec*>(gate)[src];
+    bf16x8_vec u = reinterpret_cast<const bf16x8_vec*>(up)[src];
     bf16x8_vec o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -119,7 +122,7 @@ extern "C" __global__ void __launch_bounds__(256) silu_mul_kernel(
 extern "C" __global__ void rope_kernel(
     bf16_t* __restrict__ q, bf16_t* __restrict__ k,
     const int64_t* __restrict__ positions, const float* __restrict__ cos_sin,
-    int num_q_heads, int num_kv_heads, int head_dim) {
+    int num_q_heads, int num_kv_heads, int head_dim, int q_tstride, int k_tstride) {
   const int t = blockIdx.x;
   const int h = blockIdx.y;
   const int i = threadIdx.x;           // pair index in [0, D/2)
@@ -130,9 +133,9 @@ extern "C" __global__ void rope_kernel(
   const float s = cos_sin[pos * head_dim + half + i];
   bf16_t* base;
   if (h < num_q_heads) {
-    base = q + ((int64_t)t * num_q_heads + h) * head_dim;
+    base = q + (int64_t)t * q_tstride + h * head_dim;
   } else {
-    base = k + ((int64_t)t * num_kv_heads + (h - num_q_heads)) * head_dim;
+    base = k + (int64_t)t * k_tstride + (h - num_q_heads) * head_dim;
   }
   float x1 = bf16_to_f32(((const short*)base)[i]);
   float x2 = bf16_to_f32(((const short*)base)[half + i]);
@@ -148,7 +151,7 @@ extern "C" __global__ void __launch_bounds__(256) store_kv_kernel(
     const bf16_t* __restrict__ k, const bf16_t* __restrict__ v,
     bf16_t* __restrict__ k_cache, bf16_t* __restrict__ v_cache,
     const int64_t* __restrict__ slots, int num_tokens, int kv_heads,
-    int head_dim, int block_size) {
+    int head_dim, int block_size, int kv_tstride) {
   const int dvec = head_dim / 8;
   const int64_t total = (int64_t)num_tokens * kv_heads * dvec;
   for (int64_t idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
@@ -160,9 +163,9 @@ extern "C" __global__ void __launch_bounds__(256) store_kv_kernel(
     const int64_t blk = slot / block_size;
     const int off = slot % block_size;
     const bf16x8_vec* src_k =
-        reinterpret_cast<const bf16x8_vec*>(k + ((int64_t)t * kv_heads + g) * head_dim) + dv;
+        reinterpret_cast<const bf16x8_vec*>(k + (int64_t)t * kv_tstride + g * head_dim) + dv;
     const bf16x8_vec* src_v =
-        reinterpret_cast<const bf16x8_vec*>(v + ((int64_t)t * kv_heads + g) * head_dim) + dv;
+        reinterpret_cast<const bf16x8_vec*>(v + (int64_t)t * kv_tstride + g * head_dim) + dv;
     int64_t dst_off = ((blk * kv_heads + g) * block_size + off) * head_dim;
     reinterpret_cast<bf16x8_vec*>(k_cache + dst_off)[dv] = *src_k;
     reinterpret_cast<bf16x8_vec*>(v_cache + dst_off)[dv] = *src_v;
