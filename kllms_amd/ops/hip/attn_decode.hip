@@ -16,10 +16,10 @@
 
 #include "common.h"
 
-#define TKV 128            // keys per LDS tile
+#define TKV ATTN_DECODE_TKV
 #define NTHREADS 256
 #define KLANES 8           // lanes cooperating on one key's dot product
-#define CHUNK 256          // keys per workgroup (split-K granule)
+#define CHUNK ATTN_DECODE_CHUNK
 
 // partials layout: [B, KVH, max_chunks, gqa, 130]: 128 o values + m + l
 #define PART_STRIDE 130

@@ -98,8 +98,8 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache
   TORCH_CHECK(H % KVH == 0 && H / KVH <= 8, "attn_decode: GQA group must be <= 8");
   const int gqa = H / KVH;
   const int max_blocks = block_tables.size(1);
-  const int CHUNK_KEYS = 512;  // keep in sync with CHUNK in attn_decode.hip
-  const int TILE_KEYS = 128;   // keep in sync with TKV
+  const int CHUNK_KEYS = ATTN_DECODE_CHUNK;
+  const int TILE_KEYS = ATTN_DECODE_TKV;
   const int max_chunks = std::max(1, (max_blocks * BS + CHUNK_KEYS - 1) / CHUNK_KEYS);
   auto partials = torch::empty({(int64_t)B * KVH * max_chunks * gqa * 130},
                                torch::dtype(torch::kFloat).device(q.device()));
