@@ -46,6 +46,9 @@ class BaseEngineWrapper:
         self.base_url = base_url
         self.timeout = timeout
         self.max_retries = max_retries
+        # "similarity" (default) or "key" — selects the alignment engine
+        # (reference: the commented import swap at consolidation.py:22)
+        self.consensus_aligner = kwargs.pop("consensus_aligner", "similarity")
         self._extra_kwargs = kwargs
 
     def _make_engine_client(self):

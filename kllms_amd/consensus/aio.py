@@ -102,6 +102,7 @@ async def async_consolidate_chat_completions(
     async_get_openai_embeddings_from_text: ASYNC_GET_EMBEDDINGS_FROM_TEXT_TYPE,
     client: Any = None,
     consensus_settings: ConsensusSettings = None,  # type: ignore[assignment]
+    aligner: str = "similarity",
 ) -> KLLMsChatCompletion:
     """Async mirror of consolidate_chat_completions (ref consolidation.py:219-303)."""
     loop = asyncio.get_running_loop()
@@ -112,6 +113,7 @@ async def async_consolidate_chat_completions(
         sync_embed,
         client,
         consensus_settings or ConsensusSettings(),
+        aligner,
     )
 
 
@@ -121,6 +123,7 @@ async def async_consolidate_parsed_chat_completions(
     client: Any = None,
     consensus_settings: ConsensusSettings = None,  # type: ignore[assignment]
     response_format: Optional[Type[BaseModel]] = None,
+    aligner: str = "similarity",
 ) -> KLLMsParsedChatCompletion:
     """Async mirror of consolidate_parsed_chat_completions (ref consolidation.py:402-493)."""
     loop = asyncio.get_running_loop()
@@ -132,4 +135,5 @@ async def async_consolidate_parsed_chat_completions(
         client,
         consensus_settings or ConsensusSettings(),
         response_format,
+        aligner,
     )

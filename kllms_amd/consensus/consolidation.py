@@ -61,10 +61,20 @@ def _consensus_over_contents(
     get_embeddings_from_text: SYNC_GET_EMBEDDINGS_FROM_TEXT_TYPE,
     client: Any,
     consensus_settings: ConsensusSettings,
+    aligner: str = "similarity",
 ) -> tuple[Any, Any]:
-    """Pre-align then run consensus (ref consolidation.py:97-111)."""
+    """Pre-align then run consensus (ref consolidation.py:97-111).
+
+    ``aligner`` selects the alignment engine: "similarity" (default,
+    pairwise-similarity Hungarian alignment) or "key" (the L1b key-based
+    engine — activated in the reference by the commented import swap at
+    consolidation.py:22)."""
     if len(contents) >= 2:
-        aligned_seq, _ = recursive_list_alignments(
+        if aligner == "key":
+            from .key_based_alignment import recursive_align as align_fn
+        else:
+            align_fn = recursive_list_alignments
+        aligned_seq, _ = align_fn(
             contents,
             consensus_settings.string_similarity_method,
             get_embeddings_from_text,
@@ -101,6 +111,7 @@ def consolidate_chat_completions(
     get_openai_embeddings_from_text: SYNC_GET_EMBEDDINGS_FROM_TEXT_TYPE,
     client: Any = None,
     consensus_settings: ConsensusSettings = None,  # type: ignore[assignment]
+    aligner: str = "similarity",
 ) -> KLLMsChatCompletion:
     """Consolidate one completion with n choices, or a list of completions,
     into a KLLMsChatCompletion with consensus (ref consolidation.py:63-216)."""
@@ -118,7 +129,7 @@ def consolidate_chat_completions(
             _safe_parse_content(c.message.content) for c in completion.choices if c.message.content
         ]
         consensus_content, likelihoods = _consensus_over_contents(
-            choice_contents, get_openai_embeddings_from_text, client, consensus_settings
+            choice_contents, get_openai_embeddings_from_text, client, consensus_settings, aligner
         )
 
         content_str = _format_consensus_content(consensus_content)
@@ -151,7 +162,7 @@ def consolidate_chat_completions(
         if comp.choices and comp.choices[0].message.content
     ]
     consensus_content, likelihoods = _consensus_over_contents(
-        completion_contents, get_openai_embeddings_from_text, client, consensus_settings
+        completion_contents, get_openai_embeddings_from_text, client, consensus_settings, aligner
     )
 
     base_completion = completion_list[0]
@@ -185,6 +196,7 @@ def consolidate_parsed_chat_completions(
     client: Any = None,
     consensus_settings: ConsensusSettings = None,  # type: ignore[assignment]
     response_format: Optional[Type[BaseModel]] = None,
+    aligner: str = "similarity",
 ) -> KLLMsParsedChatCompletion:
     """As consolidate_chat_completions plus re-validation of the consensus dict
     into the response_format model -> message.parsed; silent None on failure
@@ -201,7 +213,7 @@ def consolidate_parsed_chat_completions(
         _safe_parse_content(c.message.content) for c in completion.choices if c.message.content
     ]
     consensus_content, likelihoods = _consensus_over_contents(
-        parsed_choice_contents, get_openai_embeddings_from_text, client, consensus_settings
+        parsed_choice_contents, get_openai_embeddings_from_text, client, consensus_settings, aligner
     )
 
     parsed_consensus = None

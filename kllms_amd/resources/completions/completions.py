@@ -105,6 +105,7 @@ class Completions:
             completion,
             embeddings_wrapper,
             client=self._wrapper.client,
+            aligner=getattr(self._wrapper, "consensus_aligner", "similarity"),
         )
 
     def parse(
@@ -142,6 +143,7 @@ class Completions:
             embeddings_wrapper,
             response_format=response_format,
             client=self._wrapper.client,
+            aligner=getattr(self._wrapper, "consensus_aligner", "similarity"),
         )
 
 
@@ -182,6 +184,7 @@ class AsyncCompletions:
             completion,
             embeddings_wrapper,
             client=self._wrapper.client,
+            aligner=getattr(self._wrapper, "consensus_aligner", "similarity"),
         )
 
     async def parse(
@@ -217,4 +220,5 @@ class AsyncCompletions:
             embeddings_wrapper,
             response_format=response_format,
             client=self._wrapper.client,
+            aligner=getattr(self._wrapper, "consensus_aligner", "similarity"),
         )
