@@ -52,8 +52,12 @@ class LlamaAttention(nn.Module):
         self.scale = 1.0 / math.sqrt(self.head_dim)
         q_size = cfg.num_heads * self.head_dim
         kv_size = cfg.num_kv_heads * self.head_dim
-        # Fused QKV, head-sharded across ranks
-        self.qkv_proj = ColumnParallelLinear(cfg.hidden_size, q_size + 2 * kv_size, ctx, dtype=dtype)
+        # Fused QKV, head-sharded across ranks (each rank holds its q, k and
+        # v head shards — partition-aware slicing)
+        self.qkv_proj = ColumnParallelLinear(
+            cfg.hidden_size, q_size + 2 * kv_size, ctx, dtype=dtype,
+            partition_sizes=[q_size, kv_size, kv_size],
+        )
         self.o_proj = RowParallelLinear(q_size, cfg.hidden_size, ctx, dtype=dtype)
         self._q = self.num_heads * self.head_dim
         self._kv = self.num_kv_heads * self.head_dim
@@ -84,7 +88,10 @@ class LlamaAttention(nn.Module):
 class LlamaMLP(nn.Module):
     def __init__(self, cfg: ModelArchConfig, ctx: ParallelContext, dtype):
         super().__init__()
-        self.gate_up_proj = ColumnParallelLinear(cfg.hidden_size, 2 * cfg.intermediate_size, ctx, dtype=dtype)
+        self.gate_up_proj = ColumnParallelLinear(
+            cfg.hidden_size, 2 * cfg.intermediate_size, ctx, dtype=dtype,
+            partition_sizes=[cfg.intermediate_size, cfg.intermediate_size],
+        )
         self.down_proj = RowParallelLinear(cfg.intermediate_size, cfg.hidden_size, ctx, dtype=dtype)
         self._i = cfg.intermediate_size // ctx.world_size
 
@@ -178,6 +185,11 @@ class LlamaForCausalLM(nn.Module):
             if isinstance(mod, ColumnParallelLinear):
                 shard_dim = 0
                 full_shape[0] *= tp
+                full = torch.ones(full_shape, dtype=torch.float32, device=p.device) if (
+                    "layernorm" in name or name.endswith("norm.weight")
+                ) else torch.randn(full_shape, generator=g, dtype=torch.float32, device=p.device) * 0.02
+                p.copy_(mod.shard_full_tensor(full).to(p.dtype))
+                continue
             elif isinstance(mod, RowParallelLinear):
                 shard_dim = 1
                 full_shape[1] *= tp

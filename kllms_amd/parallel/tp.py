@@ -58,15 +58,36 @@ class ColumnParallelLinear(nn.Module):
     layer (attention QKV -> O, MLP gate/up -> down).
     """
 
-    def __init__(self, in_features: int, out_features: int, ctx: ParallelContext, dtype=None):
+    def __init__(self, in_features: int, out_features: int, ctx: ParallelContext, dtype=None,
+                 partition_sizes=None):
         super().__init__()
         assert out_features % ctx.world_size == 0, (out_features, ctx.world_size)
         self.ctx = ctx
         self.in_features = in_features
         self.out_features_per_rank = out_features // ctx.world_size
+        # For FUSED projections (qkv, gate_up): the full output dim is a
+        # concatenation of logical parts and each rank must hold the
+        # concatenation of its SHARD OF EACH PART — a contiguous slice of the
+        # fused dim would give rank 0 all of q and none of k/v.
+        self.partition_sizes = list(partition_sizes) if partition_sizes else [out_features]
+        assert sum(self.partition_sizes) == out_features
+        assert all(p % ctx.world_size == 0 for p in self.partition_sizes)
         self.weight = nn.Parameter(
             torch.empty(self.out_features_per_rank, in_features, dtype=dtype), requires_grad=False
         )
+
+    def shard_full_tensor(self, full: torch.Tensor) -> torch.Tensor:
+        """Slice this rank's shard out of the FULL fused weight (dim 0)."""
+        tp, rank = self.ctx.world_size, self.ctx.rank
+        if tp == 1:
+            return full
+        pieces = []
+        off = 0
+        for part in self.partition_sizes:
+            sz = part // tp
+            pieces.append(full.narrow(0, off + rank * sz, sz))
+            off += part
+        return torch.cat(pieces, dim=0)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return torch.nn.functional.linear(x, self.weight)
