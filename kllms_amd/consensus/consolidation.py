@@ -70,6 +70,12 @@ def _consensus_over_contents(
     engine — activated in the reference by the commented import swap at
     consolidation.py:22)."""
     if len(contents) >= 2:
+        if consensus_settings.string_similarity_method == "embeddings" and get_embeddings_from_text is not None:
+            # M5: one batched on-device embedding pass + one cosine GEMM
+            # replaces every per-pair embedding call in the alignment hot loop
+            from .accel import precompute_similarity_cache
+
+            precompute_similarity_cache(contents, get_embeddings_from_text)
         if aligner == "key":
             from .key_based_alignment import recursive_align as align_fn
         else:
