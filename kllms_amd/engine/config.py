@@ -60,6 +60,11 @@ MODEL_PRESETS: dict[str, ModelArchConfig] = {
         num_layers=2, num_heads=4, num_kv_heads=2, rope_theta=10000.0,
         max_position_embeddings=2048,
     ),
+    "mid-mixtral": ModelArchConfig(
+        arch="mixtral", vocab_size=2048, hidden_size=512, intermediate_size=1024,
+        num_layers=2, num_heads=4, num_kv_heads=2, rope_theta=10000.0,
+        num_experts=4, num_experts_per_tok=2, max_position_embeddings=2048,
+    ),
     # Tiny CPU-testable presets (same code paths, toy sizes)
     "tiny-llama": ModelArchConfig(
         arch="llama", vocab_size=512, hidden_size=64, intermediate_size=128,

@@ -446,21 +446,31 @@ class LLMEngine:
     @torch.inference_mode()
     def embed(self, texts: List[str]) -> Tuple[List[List[float]], int]:
         """Local embedding path: token-embedding mean-pool, L2-normalized.
-        Returns (vectors, total_tokens). Deterministic, model-consistent, and
-        fast (one gather + reduce on device)."""
-        vecs: List[List[float]] = []
-        total_tokens = 0
+        The whole batch runs as ONE flat gather + segment-mean on device.
+        Returns (vectors, total_tokens)."""
         emb = self.model.embed_tokens.weight
-        for t in texts:
-            ids = self.tokenizer.encode(t)
+        H = emb.shape[1]
+        ids_per_text = [self.tokenizer.encode(t) for t in texts]
+        total_tokens = sum(len(ids) for ids in ids_per_text)
+        flat = [i for ids in ids_per_text for i in ids]
+        if not flat:
+            return [[0.0] * H for _ in texts], 0
+        flat_t = torch.tensor(flat, dtype=torch.long, device=self.device)
+        seg = torch.tensor(
+            [s for s, ids in enumerate(ids_per_text) for _ in ids],
+            dtype=torch.long, device=self.device,
+        )
+        gathered = emb[flat_t].float()
+        sums = torch.zeros(len(texts), H, device=self.device)
+        sums.index_add_(0, seg, gathered)
+        counts = torch.tensor(
+            [max(1, len(ids)) for ids in ids_per_text], dtype=torch.float32, device=self.device
+        ).unsqueeze(1)
+        means = sums / counts
+        norms = means.norm(dim=1, keepdim=True).clamp_min(1e-12)
+        unit = torch.where(norms > 1e-11, means / norms, means)
+        out = unit.cpu().tolist()
+        for i, ids in enumerate(ids_per_text):
             if not ids:
-                vecs.append([0.0] * emb.shape[1])
-                continue
-            total_tokens += len(ids)
-            idt = torch.tensor(ids, dtype=torch.long, device=self.device)
-            v = emb[idt].float().mean(0)
-            n = v.norm()
-            if n > 0:
-                v = v / n
-            vecs.append(v.cpu().tolist())
-        return vecs, total_tokens
+                out[i] = [0.0] * H
+        return out, total_tokens

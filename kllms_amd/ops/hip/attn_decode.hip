@@ -104,29 +104,35 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
     }
     __syncthreads();
 
-    // ---- online softmax ---------------------------------------------------
-    for (int g = 0; g < gqa; ++g) {
-      float lm = -INFINITY;
-      for (int i = tid; i < nkeys; i += NTHREADS) lm = fmaxf(lm, s_lds[g * TKV + i]);
-      float tile_max = block_reduce_max<4>(lm, red);
-      if (tid == 0) {
+    // ---- online softmax: one WAVE per head, no cross-wave reductions ------
+    // (the per-head block-wide reduce version cost ~4 barriers x gqa per
+    // tile; a wave owns a head's whole score row, so everything reduces
+    // with in-wave shuffles and ONE barrier publishes the state)
+    {
+      const int wave = tid >> 6;
+      const int wlane = tid & 63;
+      for (int g = wave; g < gqa; g += NTHREADS / 64) {
+        float lm = -INFINITY;
+        for (int i = wlane; i < nkeys; i += 64) lm = fmaxf(lm, s_lds[g * TKV + i]);
+        const float tile_max = wave_reduce_max(lm);
         const float m_old = mstate[g];
         const float m_new = fmaxf(m_old, tile_max);
-        alpha_lds[g] = (m_old == -INFINITY) ? 0.0f : __expf(m_old - m_new);
-        mstate[g] = m_new;
+        const float alpha = (m_old == -INFINITY) ? 0.0f : __expf(m_old - m_new);
+        float ls = 0.f;
+        for (int i = wlane; i < nkeys; i += 64) {
+          const float p = __expf(s_lds[g * TKV + i] - m_new);
+          s_lds[g * TKV + i] = p;
+          ls += p;
+        }
+        const float tile_sum = wave_reduce_sum(ls);
+        if (wlane == 0) {
+          alpha_lds[g] = alpha;
+          mstate[g] = m_new;
+          lstate[g] = lstate[g] * alpha + tile_sum;
+        }
       }
-      __syncthreads();
-      const float m_new = mstate[g];
-      float ls = 0.f;
-      for (int i = tid; i < nkeys; i += NTHREADS) {
-        const float p = __expf(s_lds[g * TKV + i] - m_new);
-        s_lds[g * TKV + i] = p;
-        ls += p;
-      }
-      const float tile_sum = block_reduce_sum<4>(ls, red);
-      if (tid == 0) lstate[g] = lstate[g] * alpha_lds[g] + tile_sum;
-      __syncthreads();
     }
+    __syncthreads();
 
     // ---- V accumulation ---------------------------------------------------
     const int d0 = (tid & 63) * 2;

@@ -256,6 +256,21 @@ class TestEngineGPU:
             assert len(s.token_ids) > 0
             assert all(np.isfinite(lp) for lp in s.logprobs)
 
+    def test_mixtral_generate_gpu(self):
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.engine import GenRequest, LLMEngine
+        from kllms_amd.engine.sampling import SamplingParams
+
+        eng = LLMEngine(EngineConfig(model="mid-mixtral", max_kv_blocks=256, use_hip_graphs=False, seed=9))
+        out = eng.generate([
+            GenRequest(prompt_ids=list(range(1, 60)), n=3,
+                       sampling=SamplingParams(temperature=1.0, max_tokens=16, seed=2))
+        ])[0]
+        assert len(out.streams) == 3
+        for s in out.streams:
+            assert len(s.token_ids) > 0
+            assert all(np.isfinite(lp) for lp in s.logprobs)
+
     def test_hipgraph_decode_matches_eager(self):
         """Graph-replayed decode step vs eager forward on IDENTICAL inputs.
         (Token-exact comparison is wrong by design: rocBLAS may pick different
