@@ -58,12 +58,16 @@ class DecodeGraphRunner:
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(2):
-                out = eng.model.forward_decode(buf["ids"], batch)
+                out = eng.model.forward_hidden(buf["ids"], batch)
         torch.cuda.current_stream().wait_stream(s)
 
+        # Capture the LAYERS only; the LM-head GEMM runs eagerly per step —
+        # under capture rocBLAS picks a stream-K algorithm (~3x slower, plus a
+        # workspace fill) for the [B,H]x[H,V] shape; eager picks the
+        # weight-streaming-bound one (measured in scripts/lmhead_probe.py).
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
-            out = eng.model.forward_decode(buf["ids"], batch)
+            out = eng.model.forward_hidden(buf["ids"], batch)
         buf["out"] = out
         self.graphs[bs] = g
         self.buffers[bs] = buf
@@ -107,4 +111,4 @@ class DecodeGraphRunner:
         buf["context_lens"][:B].copy_(batch.context_lens)
         buf["context_lens"][B:].fill_(1)
         self.graphs[bs].replay()
-        return buf["out"][:B]
+        return self.engine.model.compute_logits(buf["out"][:B])
