@@ -74,6 +74,8 @@ class LocalEngineClient:
         from .engine import GenRequest
 
         messages: List[Dict[str, Any]] = call_params["messages"]
+        if not messages:
+            raise ValueError("messages must be a non-empty list")
         model: str = call_params.get("model", self.config.model)
         n: int = int(call_params.get("n") or 1)
 
@@ -134,6 +136,21 @@ class LocalEngineClient:
                 ChatCompletionTokenLogprob(token=s, bytes=list(s.encode()), logprob=lp, top_logprobs=[])
             )
         return ChoiceLogprobs(content=toks)
+
+    def _mk_timings(self, out) -> Dict[str, Any]:
+        """Per-request serving metrics (tokens/s, per-phase wall
+        times) attached as an extra `timings` field on the response."""
+        tm = dict(getattr(self.engine, "last_timings", {}))
+        completion_tokens = sum(len(s.token_ids) for s in out.streams)
+        decode_s = tm.get("decode_ms", 0.0) / 1000.0
+        tm["prefill_ms_request"] = out.prefill_ms
+        tm["decode_ms_request"] = out.decode_ms
+        if decode_s > 0:
+            tm["decode_tokens_per_s_batch"] = round(
+                tm.get("decode_steps", 0) * tm.get("n_streams", 0) / decode_s, 2
+            )
+        tm["completion_tokens"] = completion_tokens
+        return tm
 
     def _mk_usage(self, out) -> CompletionUsage:
         completion_tokens = sum(len(s.token_ids) for s in out.streams)
@@ -217,6 +234,7 @@ class LocalEngineClient:
             created=int(time.time()),
             model=model,
             usage=self._mk_usage(out),
+            timings=self._mk_timings(out),
         )
 
     def chat_completions_parse(self, **call_params: Any) -> ParsedChatCompletion:
@@ -246,6 +264,7 @@ class LocalEngineClient:
             created=int(time.time()),
             model=model,
             usage=self._mk_usage(out),
+            timings=self._mk_timings(out),
         )
 
     # --- embeddings -------------------------------------------------------------
