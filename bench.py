@@ -35,6 +35,25 @@ def log(msg):
         print(msg, file=sys.stderr, flush=True)
 
 
+def _mean_likelihood(node) -> float:
+    """Average of all leaf confidences — the reference's 'consensus quality'
+    scale (README_TESTS.md:269-273: 0.8-0.9 good)."""
+    leaves = []
+
+    def walk(v):
+        if isinstance(v, dict):
+            for x in v.values():
+                walk(x)
+        elif isinstance(v, (list, tuple)):
+            for x in v:
+                walk(x)
+        elif isinstance(v, (int, float)):
+            leaves.append(float(v))
+
+    walk(node)
+    return sum(leaves) / len(leaves) if leaves else 0.0
+
+
 def make_prompt(rank: int, step: int, i: int, prompt_len_tokens: int) -> str:
     # deterministic synthetic prompt of roughly prompt_len_tokens byte-tokens
     seedtxt = f"Request {rank}-{step}-{i}: extract the entities. "
@@ -94,6 +113,8 @@ def main():
     def embeddings_wrapper(texts):
         return k.get_embeddings(texts, "text-embedding-3-small", 2048, False)
 
+    quality_scores: list = []
+
     def run_step(step_idx: int) -> None:
         call_params_list = [
             {
@@ -109,7 +130,8 @@ def main():
         completions = eng_client.chat_completions_create_many(call_params_list)
         t_gen = time.perf_counter()
         for comp in completions:
-            consolidate_chat_completions(comp, embeddings_wrapper, client=eng_client)
+            result = consolidate_chat_completions(comp, embeddings_wrapper, client=eng_client)
+            quality_scores.append(_mean_likelihood(result.likelihoods))
         if os.environ.get("KLLMS_BENCH_VERBOSE"):
             tm = getattr(eng_client.engine, "last_timings", {})
             log(f"[bench] step {step_idx}: engine={tm}, consensus={1000 * (time.perf_counter() - t_gen):.1f}ms")
@@ -166,6 +188,7 @@ def main():
                 "parallelism": f"dp{n_ranks} tp1",
                 "completions_per_s": round(value * args.n, 3),
                 "consensus_latency_s_per_request": round(elapsed / (args.steps * args.batch), 4),
+                "mean_consensus_quality": round(sum(quality_scores) / len(quality_scores), 4) if quality_scores else None,
             },
         }
         print(json.dumps(result), flush=True)
