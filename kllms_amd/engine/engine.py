@@ -375,30 +375,9 @@ class LLMEngine:
             any_done |= s.done
         return any_done
 
-    def _decode_step(self, active: List[_Stream]) -> torch.Tensor:
-        dev = self.device
-        B = len(active)
-        ids = torch.tensor([s.last_token for s in active], dtype=torch.long, device=dev)
-        positions = torch.tensor([s.seq.num_tokens for s in active], dtype=torch.long, device=dev)
-        slots = torch.tensor([self.kv.append_slot(s.seq) for s in active], dtype=torch.long, device=dev)
-        max_blocks = max(len(s.seq.blocks) for s in active)
-        bt = torch.zeros((B, max_blocks), dtype=torch.int32)
-        for i, s in enumerate(active):
-            bt[i, : len(s.seq.blocks)] = torch.tensor(s.seq.blocks, dtype=torch.int32)
-        ctx_lens = torch.tensor([s.seq.num_tokens for s in active], dtype=torch.int32, device=dev)
-        batch = ForwardBatch(
-            mode="decode",
-            positions=positions,
-            slot_mapping=slots,
-            kv_caches=self.kv.layer_caches(),
-            block_tables=bt.to(dev),
-            context_lens=ctx_lens,
-        )
-        if self._graph_runner is not None:
-            return self._graph_runner.run(ids, batch)
-        return self.model.forward_decode(ids, batch)
-
     def _sample_and_append(self, logits: torch.Tensor, streams: List[_Stream]) -> None:
+        """Sampling for the FIRST token (from the shared prefill logits);
+        subsequent steps go through the device-resident _sample_state path."""
         logits = self._apply_penalties(logits, streams)
         temps, top_ps, top_ks, seeds, steps = self._sampling_tensors(streams)
         mask = self._constraint_mask(streams)

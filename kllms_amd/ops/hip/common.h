@@ -99,4 +99,4 @@ __device__ __forceinline__ float u64_to_uniform(uint64_t h) {
 
 // decode-attention split-K geometry (shared by attn_decode.hip and bindings)
 #define ATTN_DECODE_CHUNK 256   // keys per workgroup
-#define ATTN_DECODE_TKV 128     // keys per LDS tile
+#define ATTN_DECODE_TKV 256     // keys per LDS tile (one tile per chunk)
