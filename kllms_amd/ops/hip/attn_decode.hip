@@ -51,7 +51,8 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
   float* mstate = red + 16;                               // [gqa]
   float* lstate = mstate + 8;                              // [gqa]
   float* alpha_lds = lstate + 8;                           // [gqa]
-  int64_t* rowoff = reinterpret_cast<int64_t*>(alpha_lds + 8);  // [TKV] cache row byte offsets
+  float* o_scratch = alpha_lds + 8;                        // [4][gqa][128] partition combine
+  int64_t* rowoff = reinterpret_cast<int64_t*>(o_scratch + 4 * gqa * 128);  // [TKV] row offsets
 
   for (int i = tid; i < gqa * 128; i += NTHREADS) {
     const int g = i >> 7, d = i & 127;
@@ -61,9 +62,14 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
   if (tid < 8) { mstate[tid] = -INFINITY; lstate[tid] = 0.0f; }
   __syncthreads();
 
-  float o_acc[2][2];
+  // V-accumulation state: threads partition the KEYS (4 partitions of 64
+  // dim-pair lanes), each thread accumulating ALL q-heads for its dim pair —
+  // so every V row is loaded exactly once per block. (The head-partitioned
+  // layout re-fetched the 64 KB V tile once per head-group wave: it exceeds
+  // the 32 KB L1, quadrupling V traffic at GQA=4.)
+  float o_part[8][2];
 #pragma unroll
-  for (int a = 0; a < 2; ++a) { o_acc[a][0] = 0.f; o_acc[a][1] = 0.f; }
+  for (int g = 0; g < 8; ++g) { o_part[g][0] = 0.f; o_part[g][1] = 0.f; }
 
   const int* bt = block_tables + (int64_t)b * max_blocks;
 
@@ -134,55 +140,70 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
     }
     __syncthreads();
 
-    // ---- V accumulation ---------------------------------------------------
-    const int d0 = (tid & 63) * 2;
-    for (int slot = 0; slot < 2; ++slot) {
-      const int g = (tid >> 6) + slot * 4;
-      if (g >= gqa) break;
-      const float alpha = alpha_lds[g];
-      float a0 = o_acc[slot][0] * alpha;
-      float a1 = o_acc[slot][1] * alpha;
-      const float* sg = s_lds + g * TKV;
-      // unroll-by-8 with all loads issued before use: the serial version is
-      // a per-key latency chain (~300 cy x nkeys); 8 independent loads in
-      // flight hide it.
-      int key = 0;
-      for (; key + 8 <= nkeys; key += 8) {
-        uint32_t pairs[8];
+    // ---- V accumulation (key-partitioned, V read once) --------------------
+    {
+      const int kpart = tid >> 6;        // wave id = key partition
+      const int d0 = (tid & 63) * 2;     // this thread's dim pair
+      for (int g = 0; g < gqa; ++g) {
+        const float alpha = alpha_lds[g];
+        o_part[g][0] *= alpha;
+        o_part[g][1] *= alpha;
+      }
+      const int nk4 = (nkeys + 3) >> 2;
+      const int kbeg = kpart * nk4;
+      const int kend = min(nkeys, kbeg + nk4);
+      // 4 keys in flight: loads issued before use (hides the per-key chain)
+      int key = kbeg;
+      for (; key + 4 <= kend; key += 4) {
+        uint32_t pairs[4];
 #pragma unroll
-        for (int u = 0; u < 8; ++u)
+        for (int u = 0; u < 4; ++u)
           pairs[u] = *reinterpret_cast<const uint32_t*>(
               (const short*)(v_cache + rowoff[key + u]) + d0);
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          const float p = sg[key + u];
-          a0 += p * bf16_to_f32((short)(pairs[u] & 0xffff));
-          a1 += p * bf16_to_f32((short)(pairs[u] >> 16));
+        for (int u = 0; u < 4; ++u) {
+          const float v0 = bf16_to_f32((short)(pairs[u] & 0xffff));
+          const float v1 = bf16_to_f32((short)(pairs[u] >> 16));
+          for (int g = 0; g < gqa; ++g) {
+            const float p = s_lds[g * TKV + key + u];  // wave-uniform: LDS broadcast
+            o_part[g][0] += p * v0;
+            o_part[g][1] += p * v1;
+          }
         }
       }
-      for (; key < nkeys; ++key) {
-        const float p = sg[key];
+      for (; key < kend; ++key) {
         const uint32_t pair = *reinterpret_cast<const uint32_t*>(
             (const short*)(v_cache + rowoff[key]) + d0);
-        a0 += p * bf16_to_f32((short)(pair & 0xffff));
-        a1 += p * bf16_to_f32((short)(pair >> 16));
+        const float v0 = bf16_to_f32((short)(pair & 0xffff));
+        const float v1 = bf16_to_f32((short)(pair >> 16));
+        for (int g = 0; g < gqa; ++g) {
+          const float p = s_lds[g * TKV + key];
+          o_part[g][0] += p * v0;
+          o_part[g][1] += p * v1;
+        }
       }
-      o_acc[slot][0] = a0;
-      o_acc[slot][1] = a1;
     }
     __syncthreads();
   }
 
-  // ---- write partial (m, l, o) -------------------------------------------
+  // ---- combine the 4 key partitions, write partial (m, l, o) --------------
+  {
+    const int kpart = tid >> 6;
+    const int d0 = (tid & 63) * 2;
+    for (int g = 0; g < gqa; ++g) {
+      o_scratch[((kpart * gqa + g) * 128) + d0] = o_part[g][0];
+      o_scratch[((kpart * gqa + g) * 128) + d0 + 1] = o_part[g][1];
+    }
+  }
+  __syncthreads();
   float* base = partials +
       ((((int64_t)b * num_kv_heads + g_kv) * max_chunks + chunk) * gqa) * PART_STRIDE;
-  const int d0 = (tid & 63) * 2;
-  for (int slot = 0; slot < 2; ++slot) {
-    const int g = (tid >> 6) + slot * 4;
-    if (g >= gqa) break;
-    float* pg = base + g * PART_STRIDE;
-    pg[d0] = o_acc[slot][0];
-    pg[d0 + 1] = o_acc[slot][1];
+  for (int i = tid; i < gqa * 128; i += NTHREADS) {
+    const int g = i >> 7, d = i & 127;
+    float acc = 0.f;
+#pragma unroll
+    for (int kp = 0; kp < 4; ++kp) acc += o_scratch[((kp * gqa + g) * 128) + d];
+    base[g * PART_STRIDE + d] = acc;
   }
   if (tid < 8 && tid < gqa) {
     float* pg = base + tid * PART_STRIDE;

@@ -103,7 +103,7 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache
   const int max_chunks = std::max(1, (max_blocks * BS + CHUNK_KEYS - 1) / CHUNK_KEYS);
   auto partials = torch::empty({(int64_t)B * KVH * max_chunks * gqa * 130},
                                torch::dtype(torch::kFloat).device(q.device()));
-  const size_t lds = (gqa * 128 + gqa * TILE_KEYS + 16 + 8 + 8) * sizeof(float) +
+  const size_t lds = (gqa * 128 + gqa * TILE_KEYS + 16 + 8 + 8 + 4 * gqa * 128) * sizeof(float) +
                      TILE_KEYS * sizeof(int64_t);
   hipLaunchKernelGGL(attn_decode_partial_kernel, dim3(B, KVH, max_chunks), dim3(256), lds, cur_stream(),
                      partials.data_ptr<float>(), cbf(q), cbf(k_cache), cbf(v_cache),
