@@ -6,13 +6,21 @@
 // split is what fills it. Each block emits a partial (m, l, o[gqa][128]);
 // a small reduce kernel merges chunks with the standard log-sum-exp combine.
 //
-// Memory behavior: each block streams its chunk of K/V exactly once
-// (16-byte lane loads); per-key cache-row byte offsets are staged in LDS once
-// per tile so the inner loops do no block-table math. All q-heads of the GQA
-// group are served by the same block (K/V read once per sequence-chunk).
+// Memory behavior: each block streams its chunk of K/V exactly ONCE:
+//   - scores: 8-lane thread groups per key, 16-byte K loads, in-wave reduce;
+//   - online softmax: one wave per head (in-wave shuffles, single barrier);
+//   - V accumulation: threads partition the KEYS (4 partitions x 64 dim-pair
+//     lanes), each thread accumulating ALL q-heads for its dim pair, so a V
+//     row is loaded once per block (a head-partitioned layout re-fetches the
+//     64 KB V tile per head-group wave — it exceeds the 32 KB L1);
+//   - per-key cache-row offsets staged in LDS once per tile.
+//
+// The kernel is TEMPLATED on the GQA group size: the per-head accumulator
+// array must be statically indexed (a runtime-gqa loop sends it to scratch —
+// cdna guide §5.4 rule 20) and the head loops fully unrolled.
 //
 // Layouts: q [B, H, D=128]; caches [NB, KVH, BS, D]; block_tables [B, MAXB];
-// context_lens INCLUDE the current token. GQA group <= 8.
+// context_lens INCLUDE the current token. GQA in {1, 2, 4, 8}.
 
 #include "common.h"
 
@@ -24,62 +32,55 @@
 // partials layout: [B, KVH, max_chunks, gqa, 130]: 128 o values + m + l
 #define PART_STRIDE 130
 
-extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kernel(
+template <int GQA>
+__global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
     float* __restrict__ partials,
     const bf16_t* __restrict__ q,        // [B, H, 128], row stride q_tstride
     const bf16_t* __restrict__ k_cache,  // [NB, KVH, BS, 128]
     const bf16_t* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [B, max_blocks]
     const int* __restrict__ context_lens,  // [B]
-    float scale, int num_q_heads, int num_kv_heads,
+    float scale, int num_kv_heads,
     int block_size, int max_blocks, int max_chunks, int q_tstride) {
   const int b = blockIdx.x;
   const int g_kv = blockIdx.y;
   const int chunk = blockIdx.z;
-  const int gqa = num_q_heads / num_kv_heads;
   const int L = context_lens[b];
   const int c0 = chunk * CHUNK;
   if (c0 >= L && chunk > 0) return;     // no keys for this chunk
   const int c1 = min(L, c0 + CHUNK);
   const int tid = threadIdx.x;
-  const int lane = tid & 63;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* q_lds = reinterpret_cast<float*>(smem);          // [gqa][128]
-  float* s_lds = q_lds + gqa * 128;                       // [gqa][TKV]
-  float* red = s_lds + gqa * TKV;                         // [16]
-  float* mstate = red + 16;                               // [gqa]
-  float* lstate = mstate + 8;                              // [gqa]
-  float* alpha_lds = lstate + 8;                           // [gqa]
-  float* o_scratch = alpha_lds + 8;                        // [4][gqa][128] partition combine
-  int64_t* rowoff = reinterpret_cast<int64_t*>(o_scratch + 4 * gqa * 128);  // [TKV] row offsets
+  float* q_lds = reinterpret_cast<float*>(smem);          // [GQA][128]
+  float* s_lds = q_lds + GQA * 128;                       // [GQA][TKV]
+  float* mstate = s_lds + GQA * TKV;                      // [GQA]
+  float* lstate = mstate + 8;                              // [GQA]
+  float* alpha_lds = lstate + 8;                           // [GQA]
+  float* o_scratch = alpha_lds + 8;                        // [4][GQA][128]
+  int64_t* rowoff = reinterpret_cast<int64_t*>(o_scratch + 4 * GQA * 128);  // [TKV]
 
-  for (int i = tid; i < gqa * 128; i += NTHREADS) {
+  for (int i = tid; i < GQA * 128; i += NTHREADS) {
     const int g = i >> 7, d = i & 127;
-    const bf16_t* qp = q + (int64_t)b * q_tstride + (g_kv * gqa + g) * 128;
+    const bf16_t* qp = q + (int64_t)b * q_tstride + (g_kv * GQA + g) * 128;
     q_lds[i] = bf16_to_f32(((const short*)qp)[d]) * scale;
   }
   if (tid < 8) { mstate[tid] = -INFINITY; lstate[tid] = 0.0f; }
   __syncthreads();
 
-  // V-accumulation state: threads partition the KEYS (4 partitions of 64
-  // dim-pair lanes), each thread accumulating ALL q-heads for its dim pair —
-  // so every V row is loaded exactly once per block. (The head-partitioned
-  // layout re-fetched the 64 KB V tile once per head-group wave: it exceeds
-  // the 32 KB L1, quadrupling V traffic at GQA=4.)
-  float o_part[8][2];
+  float o_part[GQA][2];
 #pragma unroll
-  for (int g = 0; g < 8; ++g) { o_part[g][0] = 0.f; o_part[g][1] = 0.f; }
+  for (int g = 0; g < GQA; ++g) { o_part[g][0] = 0.f; o_part[g][1] = 0.f; }
 
   const int* bt = block_tables + (int64_t)b * max_blocks;
 
   for (int tile = c0; tile < c1; tile += TKV) {
     const int nkeys = min(TKV, c1 - tile);
     // stage per-key cache-row element offsets (shared by K and V)
-    if (tid < nkeys) {
-      const int gk = tile + tid;
-      rowoff[tid] = (((int64_t)bt[gk / block_size] * num_kv_heads + g_kv) * block_size +
-                     (gk % block_size)) * 128;
+    for (int i = tid; i < nkeys; i += NTHREADS) {
+      const int gk = tile + i;
+      rowoff[i] = (((int64_t)bt[gk / block_size] * num_kv_heads + g_kv) * block_size +
+                   (gk % block_size)) * 128;
     }
     __syncthreads();
 
@@ -97,7 +98,8 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
         for (int j = 0; j < 8; ++j) af[j] = bf16_to_f32(ka[j]);
 #pragma unroll
         for (int j = 0; j < 8; ++j) af[8 + j] = bf16_to_f32(kb[j]);
-        for (int g = 0; g < gqa; ++g) {
+#pragma unroll
+        for (int g = 0; g < GQA; ++g) {
           const float* qg = q_lds + g * 128 + kl * 16;
           float acc = 0.f;
 #pragma unroll
@@ -110,14 +112,11 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
     }
     __syncthreads();
 
-    // ---- online softmax: one WAVE per head, no cross-wave reductions ------
-    // (the per-head block-wide reduce version cost ~4 barriers x gqa per
-    // tile; a wave owns a head's whole score row, so everything reduces
-    // with in-wave shuffles and ONE barrier publishes the state)
+    // ---- online softmax: one WAVE per head, in-wave reductions only -------
     {
       const int wave = tid >> 6;
       const int wlane = tid & 63;
-      for (int g = wave; g < gqa; g += NTHREADS / 64) {
+      for (int g = wave; g < GQA; g += NTHREADS / 64) {
         float lm = -INFINITY;
         for (int i = wlane; i < nkeys; i += 64) lm = fmaxf(lm, s_lds[g * TKV + i]);
         const float tile_max = wave_reduce_max(lm);
@@ -144,7 +143,8 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
     {
       const int kpart = tid >> 6;        // wave id = key partition
       const int d0 = (tid & 63) * 2;     // this thread's dim pair
-      for (int g = 0; g < gqa; ++g) {
+#pragma unroll
+      for (int g = 0; g < GQA; ++g) {
         const float alpha = alpha_lds[g];
         o_part[g][0] *= alpha;
         o_part[g][1] *= alpha;
@@ -152,8 +152,8 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
       const int nk4 = (nkeys + 3) >> 2;
       const int kbeg = kpart * nk4;
       const int kend = min(nkeys, kbeg + nk4);
-      // 4 keys in flight: loads issued before use (hides the per-key chain)
       int key = kbeg;
+      // 4 keys in flight: loads issued before use (hides the per-key chain)
       for (; key + 4 <= kend; key += 4) {
         uint32_t pairs[4];
 #pragma unroll
@@ -164,7 +164,8 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
         for (int u = 0; u < 4; ++u) {
           const float v0 = bf16_to_f32((short)(pairs[u] & 0xffff));
           const float v1 = bf16_to_f32((short)(pairs[u] >> 16));
-          for (int g = 0; g < gqa; ++g) {
+#pragma unroll
+          for (int g = 0; g < GQA; ++g) {
             const float p = s_lds[g * TKV + key + u];  // wave-uniform: LDS broadcast
             o_part[g][0] += p * v0;
             o_part[g][1] += p * v1;
@@ -176,7 +177,8 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
             (const short*)(v_cache + rowoff[key]) + d0);
         const float v0 = bf16_to_f32((short)(pair & 0xffff));
         const float v1 = bf16_to_f32((short)(pair >> 16));
-        for (int g = 0; g < gqa; ++g) {
+#pragma unroll
+        for (int g = 0; g < GQA; ++g) {
           const float p = s_lds[g * TKV + key];
           o_part[g][0] += p * v0;
           o_part[g][1] += p * v1;
@@ -190,26 +192,51 @@ extern "C" __global__ void __launch_bounds__(NTHREADS) attn_decode_partial_kerne
   {
     const int kpart = tid >> 6;
     const int d0 = (tid & 63) * 2;
-    for (int g = 0; g < gqa; ++g) {
-      o_scratch[((kpart * gqa + g) * 128) + d0] = o_part[g][0];
-      o_scratch[((kpart * gqa + g) * 128) + d0 + 1] = o_part[g][1];
+#pragma unroll
+    for (int g = 0; g < GQA; ++g) {
+      o_scratch[((kpart * GQA + g) * 128) + d0] = o_part[g][0];
+      o_scratch[((kpart * GQA + g) * 128) + d0 + 1] = o_part[g][1];
     }
   }
   __syncthreads();
   float* base = partials +
-      ((((int64_t)b * num_kv_heads + g_kv) * max_chunks + chunk) * gqa) * PART_STRIDE;
-  for (int i = tid; i < gqa * 128; i += NTHREADS) {
+      ((((int64_t)b * num_kv_heads + g_kv) * max_chunks + chunk) * GQA) * PART_STRIDE;
+  for (int i = tid; i < GQA * 128; i += NTHREADS) {
     const int g = i >> 7, d = i & 127;
     float acc = 0.f;
 #pragma unroll
-    for (int kp = 0; kp < 4; ++kp) acc += o_scratch[((kp * gqa + g) * 128) + d];
+    for (int kp = 0; kp < 4; ++kp) acc += o_scratch[((kp * GQA + g) * 128) + d];
     base[g * PART_STRIDE + d] = acc;
   }
-  if (tid < 8 && tid < gqa) {
+  if (tid < 8 && tid < GQA) {
     float* pg = base + tid * PART_STRIDE;
     pg[128] = mstate[tid];
     pg[129] = lstate[tid];
   }
+}
+
+// host-side dispatcher: picks the GQA template instantiation
+extern "C" void launch_attn_decode_partial(
+    float* partials, const bf16_t* q, const bf16_t* k_cache, const bf16_t* v_cache,
+    const int* block_tables, const int* context_lens, float scale,
+    int num_q_heads, int num_kv_heads, int block_size, int max_blocks,
+    int max_chunks, int q_tstride, int B, hipStream_t stream) {
+  const int gqa = num_q_heads / num_kv_heads;
+  const dim3 grid(B, num_kv_heads, max_chunks);
+  const size_t lds = (gqa * 128 + gqa * TKV + 24 + 4 * gqa * 128) * sizeof(float) +
+                     TKV * sizeof(int64_t);
+#define LAUNCH(G)                                                                  \
+  hipLaunchKernelGGL(attn_decode_partial_t<G>, grid, dim3(NTHREADS), lds, stream,  \
+                     partials, q, k_cache, v_cache, block_tables, context_lens,    \
+                     scale, num_kv_heads, block_size, max_blocks, max_chunks, q_tstride)
+  switch (gqa) {
+    case 1: LAUNCH(1); break;
+    case 2: LAUNCH(2); break;
+    case 4: LAUNCH(4); break;
+    case 8: LAUNCH(8); break;
+    default: LAUNCH(8); break;  // guarded by the binding's gqa<=8 check
+  }
+#undef LAUNCH
 }
 
 // merge the per-chunk partials: one 64-lane wave per (b, q_head)

@@ -12,7 +12,7 @@ extern "C" __global__ void fused_add_rmsnorm_kernel(bf16_t*, const bf16_t*, bf16
 extern "C" __global__ void silu_mul_kernel(bf16_t*, const bf16_t*, const bf16_t*, int64_t, int, int);
 extern "C" __global__ void rope_kernel(bf16_t*, bf16_t*, const int64_t*, const float*, int, int, int, int, int);
 extern "C" __global__ void store_kv_kernel(const bf16_t*, const bf16_t*, bf16_t*, bf16_t*, const int64_t*, int, int, int, int, int);
-extern "C" __global__ void attn_decode_partial_kernel(float*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, float, int, int, int, int, int, int);
+extern "C" void launch_attn_decode_partial(float*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, float, int, int, int, int, int, int, int, hipStream_t);
 extern "C" __global__ void attn_decode_reduce_kernel(bf16_t*, const float*, const int*, int, int, int);
 extern "C" __global__ void attn_prefill_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, const int*, float, int, int, int, int);
 extern "C" __global__ void sample_kernel(int64_t*, float*, const float*, const float*, const float*, const int*, const int64_t*, const int64_t*, const uint32_t*, int);
@@ -95,20 +95,19 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache
   const int B = q.size(0), H = q.size(1), D = q.size(2);
   const int KVH = k_cache.size(1), BS = k_cache.size(2);
   TORCH_CHECK(D == 128, "attn_decode: head_dim must be 128");
-  TORCH_CHECK(H % KVH == 0 && H / KVH <= 8, "attn_decode: GQA group must be <= 8");
+  TORCH_CHECK(H % KVH == 0, "attn_decode: H must be a multiple of KVH");
+  TORCH_CHECK(H / KVH == 1 || H / KVH == 2 || H / KVH == 4 || H / KVH == 8,
+              "attn_decode: GQA group must be 1, 2, 4 or 8");
   const int gqa = H / KVH;
   const int max_blocks = block_tables.size(1);
   const int CHUNK_KEYS = ATTN_DECODE_CHUNK;
-  const int TILE_KEYS = ATTN_DECODE_TKV;
   const int max_chunks = std::max(1, (max_blocks * BS + CHUNK_KEYS - 1) / CHUNK_KEYS);
   auto partials = torch::empty({(int64_t)B * KVH * max_chunks * gqa * 130},
                                torch::dtype(torch::kFloat).device(q.device()));
-  const size_t lds = (gqa * 128 + gqa * TILE_KEYS + 16 + 8 + 8 + 4 * gqa * 128) * sizeof(float) +
-                     TILE_KEYS * sizeof(int64_t);
-  hipLaunchKernelGGL(attn_decode_partial_kernel, dim3(B, KVH, max_chunks), dim3(256), lds, cur_stream(),
-                     partials.data_ptr<float>(), cbf(q), cbf(k_cache), cbf(v_cache),
-                     block_tables.data_ptr<int>(), context_lens.data_ptr<int>(),
-                     (float)scale, H, KVH, BS, max_blocks, max_chunks, (int)q.stride(0));
+  launch_attn_decode_partial(
+      partials.data_ptr<float>(), cbf(q), cbf(k_cache), cbf(v_cache),
+      block_tables.data_ptr<int>(), context_lens.data_ptr<int>(),
+      (float)scale, H, KVH, BS, max_blocks, max_chunks, (int)q.stride(0), B, cur_stream());
   hipLaunchKernelGGL(attn_decode_reduce_kernel, dim3(B, H), dim3(64), 0, cur_stream(),
                      bf(out), partials.data_ptr<float>(), context_lens.data_ptr<int>(),
                      H, KVH, max_chunks);
