@@ -59,9 +59,36 @@ class MixtralMoE(nn.Module):
         probs = torch.softmax(router_logits, dim=-1)
         topw, topi = torch.topk(probs, self.k, dim=-1)             # [T, k]
         topw = topw / topw.sum(dim=-1, keepdim=True)
+
+        if T <= 8 * self.E:
+            out = self._forward_dense(x, topw, topi)
+        else:
+            out = self._forward_grouped(x, topw, topi)
+        return self.ctx.all_reduce(out)
+
+    def _forward_dense(self, x: torch.Tensor, topw: torch.Tensor, topi: torch.Tensor) -> torch.Tensor:
+        """Decode-shape path: with few tokens and top-k routing, (nearly)
+        every expert's weights stream from HBM anyway, so computing ALL
+        experts with batched GEMMs costs the same memory traffic as exact
+        routing, needs no token bucketing, and replays cleanly in a hipGraph
+        (no data-dependent shapes)."""
+        T = x.shape[0]
+        xb = x.unsqueeze(0).expand(self.E, T, -1)                     # [E, T, H]
+        gu2 = torch.bmm(xb, self.w_gate_up.transpose(1, 2)).view(self.E * T, 2 * self.I)
+        act = ops.silu_mul(gu2[:, : self.I], gu2[:, self.I:]).view(self.E, T, self.I)
+        ye = torch.bmm(act, self.w_down.transpose(1, 2))              # [E, T, H]
+        # combine: weight[e, t] = topw where topi == e else 0
+        w = torch.zeros(self.E, T, device=x.device, dtype=torch.float32)
+        w.scatter_(0, topi.t().long(), topw.t().float())
+        return (ye.float() * w.unsqueeze(-1)).sum(dim=0).to(x.dtype)
+
+    def _forward_grouped(self, x: torch.Tensor, topw: torch.Tensor, topi: torch.Tensor) -> torch.Tensor:
+        """Prefill-shape path: tokens bucketed per expert, one GEMM group per
+        expert (rocBLAS); avoids the dense path's E/k x FLOP overhead when
+        T is large enough that compute, not weight streaming, dominates."""
+        T = x.shape[0]
         out = torch.zeros_like(x)
-        # grouped per-expert compute over token buckets
-        flat_exp = topi.reshape(-1)                                # [T*k]
+        flat_exp = topi.reshape(-1)
         flat_tok = torch.arange(T, device=x.device).repeat_interleave(self.k)
         flat_w = topw.reshape(-1)
         for e in range(self.E):
@@ -74,7 +101,7 @@ class MixtralMoE(nn.Module):
             g, u = gu.split([self.I, self.I], dim=-1)
             ye = torch.nn.functional.linear(ops.silu_mul(g, u), self.w_down[e])
             out.index_add_(0, toks, ye * flat_w[sel].unsqueeze(-1).to(ye.dtype))
-        return self.ctx.all_reduce(out)
+        return out
 
 
 class MixtralDecoderLayer(LlamaDecoderLayer):

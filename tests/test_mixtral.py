@@ -76,3 +76,22 @@ def test_embed_batched_matches_per_text(engine):
     solo, _ = engine.embed([texts[2]])
     assert vecs[2] == pytest.approx(solo[0], abs=1e-6)
     assert vecs[1] == [0.0] * len(vecs[1])
+
+
+def test_dense_path_matches_grouped():
+    """The decode-shape dense-all-experts path and the bucketed grouped path
+    must agree numerically."""
+    cfg = MODEL_PRESETS["tiny-mixtral"]
+    moe = MixtralMoE(cfg, ParallelContext(), dtype=torch.float32)
+    torch.manual_seed(3)
+    moe.gate.weight.copy_(torch.randn_like(moe.gate.weight) * 0.2)
+    moe.w_gate_up.copy_(torch.randn_like(moe.w_gate_up) * 0.1)
+    moe.w_down.copy_(torch.randn_like(moe.w_down) * 0.1)
+    x = torch.randn(11, cfg.hidden_size)
+    logits = moe.gate(x).float()
+    probs = torch.softmax(logits, dim=-1)
+    topw, topi = torch.topk(probs, moe.k, dim=-1)
+    topw = topw / topw.sum(dim=-1, keepdim=True)
+    dense = moe._forward_dense(x, topw, topi)
+    grouped = moe._forward_grouped(x, topw, topi)
+    assert torch.allclose(dense, grouped, atol=1e-4), (dense - grouped).abs().max()
