@@ -221,11 +221,46 @@ class LLMEngine:
     def generate(self, requests: List[GenRequest]) -> List[RequestOutput]:
         if not requests:
             return []
+        # admission control: bound concurrent decode streams; excess requests
+        # are served in sub-batches (same results, sequential batches)
+        total_streams = sum(max(1, r.n) for r in requests)
+        if total_streams > self.config.max_batch_size and len(requests) > 1:
+            outputs: List[RequestOutput] = []
+            sub: List[GenRequest] = []
+            sub_streams = 0
+            for r in requests:
+                rn = max(1, r.n)
+                if sub and sub_streams + rn > self.config.max_batch_size:
+                    outputs.extend(self.generate(sub))
+                    sub, sub_streams = [], 0
+                sub.append(r)
+                sub_streams += rn
+            if sub:
+                outputs.extend(self.generate(sub))
+            return outputs
+
+        parent_seqs: List[SequenceKV] = []
+        streams: List[_Stream] = []
+        try:
+            return self._generate_batch(requests, parent_seqs, streams)
+        except Exception:
+            # free every block this batch still holds (KV exhaustion mid-run,
+            # kernel failure, ...) so the engine stays usable
+            for seq in parent_seqs:
+                if seq.blocks:
+                    self.kv.free_sequence(seq)
+            for s in streams:
+                if s.seq.blocks:
+                    self.kv.free_sequence(s.seq)
+            raise
+
+    def _generate_batch(
+        self, requests: List[GenRequest], parent_seqs: List[SequenceKV], streams: List[_Stream]
+    ) -> List[RequestOutput]:
         dev = self.device
         t0 = time.perf_counter()
 
         # ---- shared prefill: one packed varlen batch over all prompts -------
-        parent_seqs: List[SequenceKV] = []
         all_ids: List[int] = []
         all_pos: List[int] = []
         all_slots: List[int] = []
@@ -254,7 +289,6 @@ class LLMEngine:
         t1 = time.perf_counter()
 
         # ---- fork: n streams per request share the prompt KV blocks ---------
-        streams: List[_Stream] = []
         outputs = [RequestOutput(prompt_tokens=len(r.prompt_ids)) for r in requests]
         for ri, req in enumerate(requests):
             base_seed = req.sampling.seed if req.sampling.seed is not None else (self.config.seed * 1000003 + ri)

@@ -88,19 +88,34 @@ class PagedKVCache:
     # --- sequence-level operations ------------------------------------------
     def alloc_sequence(self, num_tokens: int) -> SequenceKV:
         n_blocks = (num_tokens + self.block_size - 1) // self.block_size
-        return SequenceKV(blocks=[self.allocator.alloc() for _ in range(n_blocks)], num_tokens=num_tokens)
+        blocks: List[int] = []
+        try:
+            for _ in range(n_blocks):
+                blocks.append(self.allocator.alloc())
+        except Exception:
+            for b in blocks:
+                self.allocator.free(b)
+            raise
+        return SequenceKV(blocks=blocks, num_tokens=num_tokens)
 
     def fork(self, parent: SequenceKV) -> SequenceKV:
         """Share the parent's blocks by refcount. The block the child will
         write into next (a partially-filled tail block) is copied EAGERLY so
         the decode hot loop never needs copy-on-write bookkeeping."""
-        child = SequenceKV(blocks=list(parent.blocks), num_tokens=parent.num_tokens)
+        child = SequenceKV(blocks=[], num_tokens=parent.num_tokens)
         tail = parent.num_tokens % self.block_size
-        for i, b in enumerate(child.blocks):
-            if i == len(child.blocks) - 1 and tail != 0:
-                child.blocks[i] = self._copy_block(b)
-            else:
-                self.allocator.incref(b)
+        try:
+            for i, b in enumerate(parent.blocks):
+                if i == len(parent.blocks) - 1 and tail != 0:
+                    child.blocks.append(self._copy_block(b))
+                else:
+                    self.allocator.incref(b)
+                    child.blocks.append(b)
+        except Exception:
+            for b in child.blocks:
+                self.allocator.free(b)
+            child.blocks = []
+            raise
         return child
 
     def _copy_block(self, src: int) -> int:

@@ -141,3 +141,35 @@ class TestBatchedRequests:
         solo2 = engine.generate([r2()])[0]
         assert batch_out[0].streams[0].token_ids == solo1.streams[0].token_ids
         assert batch_out[1].streams[0].token_ids == solo2.streams[0].token_ids
+
+
+class TestRobustness:
+    def test_kv_exhaustion_frees_blocks(self):
+        eng = LLMEngine(EngineConfig(
+            model="tiny-llama", max_kv_blocks=8, use_hip_graphs=False, device="cpu", seed=0,
+        ))
+        free0 = eng.kv.allocator.num_free
+        with pytest.raises(RuntimeError):
+            # 8 blocks * 16 = 128 slots; this needs far more
+            eng.generate([GenRequest(prompt_ids=list(range(1, 100)), n=8,
+                                     sampling=greedy(64))])
+        assert eng.kv.allocator.num_free == free0, "blocks leaked after failure"
+        # engine still serves afterwards
+        out = eng.generate([GenRequest(prompt_ids=[1, 2, 3], n=1, sampling=greedy(4))])[0]
+        assert len(out.streams) == 1
+
+    def test_admission_splits_oversized_batches(self, engine):
+        small = EngineConfig(
+            model="tiny-llama", max_kv_blocks=512, use_hip_graphs=False,
+            device="cpu", seed=0, max_batch_size=4,
+        )
+        eng = LLMEngine(small)
+        reqs = [GenRequest(prompt_ids=[1 + i, 2, 3], n=3, sampling=greedy(4)) for i in range(4)]
+        outs = eng.generate(reqs)  # 12 streams > max 4 -> split into sub-batches
+        assert len(outs) == 4
+        assert all(len(o.streams) == 3 for o in outs)
+        # equal to serving them individually
+        solo = [eng.generate([GenRequest(prompt_ids=[1 + i, 2, 3], n=3, sampling=greedy(4))])[0]
+                for i in range(4)]
+        for a, b in zip(outs, solo):
+            assert [s.token_ids for s in a.streams] == [s.token_ids for s in b.streams]
