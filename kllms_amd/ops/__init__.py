@@ -95,20 +95,23 @@ def store_kv(
 
 
 def _prefill_tiles(cu_seqlens: torch.Tensor, device) -> tuple:
-    """Per-32-row q-tile metadata for the MFMA prefill kernel: each tile's
-    sequence start token, row0 position within the sequence, and seq length.
-    One host round-trip per prefill batch (once per generate() call)."""
+    """Grouped q-tile metadata for the MFMA prefill kernel: 32-row q-tiles of
+    each sequence packed 4 per workgroup (the 4 waves share the K/V LDS
+    staging); idle wave slots padded with -1. One host round-trip per prefill
+    batch (once per generate() call)."""
     cu = cu_seqlens.cpu().tolist()
-    seq_start, qpos0, seqlen = [], [], []
+    grp_start, grp_len, grp_qpos0 = [], [], []
     for i in range(len(cu) - 1):
         s, e = cu[i], cu[i + 1]
         L = e - s
-        for r0 in range(0, L, 32):
-            seq_start.append(s)
-            qpos0.append(r0)
-            seqlen.append(L)
+        tiles = list(range(0, L, 32))
+        for g0 in range(0, len(tiles), 4):
+            grp_start.append(s)
+            grp_len.append(L)
+            four = tiles[g0 : g0 + 4]
+            grp_qpos0.extend(four + [-1] * (4 - len(four)))
     t = lambda x: torch.tensor(x, dtype=torch.int32, device=device)
-    return t(seq_start), t(qpos0), t(seqlen)
+    return t(grp_start), t(grp_qpos0), t(grp_len)
 
 
 def attn_prefill_varlen(
@@ -116,8 +119,8 @@ def attn_prefill_varlen(
 ) -> torch.Tensor:
     if q.is_cuda and not _force_torch():
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-        ts, tq, tl = _prefill_tiles(cu_seqlens, q.device)
-        _hip_or_raise().attn_prefill(out, q, k, v, ts, tq, tl, scale)
+        gs, gq, gl = _prefill_tiles(cu_seqlens, q.device)
+        _hip_or_raise().attn_prefill(out, q, k, v, gs, gq, gl, scale)
         return out
     return torch_ref.attn_prefill_varlen(q, k, v, cu_seqlens, scale)
 
