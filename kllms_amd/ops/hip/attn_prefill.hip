@@ -30,11 +30,11 @@
 typedef __bf16 bf16x8_mfma __attribute__((ext_vector_type(8)));
 
 #define QBLK 32
-#define KVBLK 32
+#define KVBLK 64                   // staged keys per barrier round (2 MFMA sub-tiles)
 #define DHEAD 128
 #define NWAVES 4
 #define K_ROW_BYTES 256            // 128 bf16
-#define VT_ROW_SHORTS 40           // 32 keys + 8 pad (conflict-free b128 reads)
+#define VT_ROW_SHORTS 72           // 64 keys + 8 pad (conflict-free b128 reads)
 
 __device__ __forceinline__ int acc_row(int r, int half) {
   return (r & 3) + 8 * (r >> 2) + 4 * half;   // C/D row map
@@ -66,8 +66,8 @@ extern "C" __global__ void __launch_bounds__(64 * NWAVES) attn_prefill_kernel(
 
   __shared__ __attribute__((aligned(16))) char smem[KVBLK * K_ROW_BYTES +
                                                     DHEAD * VT_ROW_SHORTS * 2];
-  char* k_lds = smem;                                   // [32][256B], XOR-swizzled
-  short* vt_lds = reinterpret_cast<short*>(smem + KVBLK * K_ROW_BYTES);  // [128][40]
+  char* k_lds = smem;                                   // [64][256B], XOR-swizzled
+  short* vt_lds = reinterpret_cast<short*>(smem + KVBLK * K_ROW_BYTES);  // [128][72]
 
   // ---- Q fragments (B-operand): 8 d-slices of 16 --------------------------
   bf16x8_mfma q_frag[8];
@@ -92,15 +92,16 @@ extern "C" __global__ void __launch_bounds__(64 * NWAVES) attn_prefill_kernel(
   }
   const int kv_end_me = active ? min(seqlen, qpos0 + QBLK) : 0;
 
-  for (int kt0 = 0; kt0 < kv_end_grp; kt0 += KVBLK) {
-    const int nkeys = min(KVBLK, kv_end_grp - kt0);
+  for (int kt00 = 0; kt00 < kv_end_grp; kt00 += KVBLK) {
+    const int nkeys_blk = min(KVBLK, kv_end_grp - kt00);
 
     // ---- stage K (swizzled) + V^T, all 256 threads cooperatively ----------
-    {
-      const int key = tid & 31;          // one key per thread
-      const int piece = tid >> 5;        // 8 pieces of 16 dims
-      const bool valid = key < nkeys;
-      const bf16_t* kp = k + (int64_t)(seq_start + kt0 + min(key, nkeys - 1)) * kv_tstride + g_kv * DHEAD;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int key = (tid & 31) + kk * 32;   // two keys per thread
+      const int piece = tid >> 5;             // 8 pieces of 16 dims
+      const bool valid = key < nkeys_blk;
+      const bf16_t* kp = k + (int64_t)(seq_start + kt00 + min(key, nkeys_blk - 1)) * kv_tstride + g_kv * DHEAD;
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
         const int byte_off = piece * 32 + c * 16;
@@ -111,7 +112,7 @@ extern "C" __global__ void __launch_bounds__(64 * NWAVES) attn_prefill_kernel(
             k_lds + key * K_ROW_BYTES + (byte_off ^ ((key & 15) << 4))) = val;
       }
       // V^T: this thread transposes its [16 dims x 1 key] sliver
-      const bf16_t* vp = v + (int64_t)(seq_start + kt0 + min(key, nkeys - 1)) * kv_tstride + g_kv * DHEAD;
+      const bf16_t* vp = v + (int64_t)(seq_start + kt00 + min(key, nkeys_blk - 1)) * kv_tstride + g_kv * DHEAD;
       bf16x8_vec va = valid ? reinterpret_cast<const bf16x8_vec*>((const short*)vp)[piece * 2]
                             : bf16x8_vec{0, 0, 0, 0, 0, 0, 0, 0};
       bf16x8_vec vb = valid ? reinterpret_cast<const bf16x8_vec*>((const short*)vp)[piece * 2 + 1]
@@ -124,14 +125,19 @@ extern "C" __global__ void __launch_bounds__(64 * NWAVES) attn_prefill_kernel(
     }
     __syncthreads();
 
-    if (active && kt0 < kv_end_me) {
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+    const int kt0 = kt00 + sub * 32;
+    const int nkeys = min(32, nkeys_blk - sub * 32);
+    if (active && kt0 < kv_end_me && nkeys > 0) {
       // ---- QK^T: S[key][qrow] over 8 d-slices -----------------------------
       f32x16 s_acc = {};
 #pragma unroll
       for (int s = 0; s < 8; ++s) {
         const int byte_off = (s * 2 + half) * 16;
+        const int krow = col + sub * 32;
         bf16x8_mfma a_frag = *reinterpret_cast<const bf16x8_mfma*>(
-            k_lds + col * K_ROW_BYTES + (byte_off ^ ((col & 15) << 4)));
+            k_lds + krow * K_ROW_BYTES + (byte_off ^ ((krow & 15) << 4)));
         s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_frag, q_frag[s], s_acc, 0, 0, 0);
       }
 
@@ -195,7 +201,7 @@ extern "C" __global__ void __launch_bounds__(64 * NWAVES) attn_prefill_kernel(
 #pragma unroll
         for (int kt = 0; kt < 2; ++kt) {
           bf16x8_mfma vfrag = *reinterpret_cast<const bf16x8_mfma*>(
-              vt_lds + d_row * VT_ROW_SHORTS + kt * 16 + half * 8);
+              vt_lds + d_row * VT_ROW_SHORTS + sub * 32 + kt * 16 + half * 8);
           bf16x8_mfma pfrag;
           {
             uint32_t* pf = reinterpret_cast<uint32_t*>(&pfrag);
@@ -206,6 +212,7 @@ extern "C" __global__ void __launch_bounds__(64 * NWAVES) attn_prefill_kernel(
         }
       }
     }
+    }  // sub
     __syncthreads();
   }
 
