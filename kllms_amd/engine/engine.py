@@ -394,8 +394,11 @@ class LLMEngine:
         state.steps += 1
         state.positions += 1
         state.ctx += 1
-        tokens_l = tokens.tolist()
-        logprobs_l = logprobs.tolist()
+        # ONE device->host transfer per step (tokens + logprobs packed)
+        packed = torch.cat([tokens.to(torch.float64), logprobs.to(torch.float64)]).cpu()
+        B = tokens.shape[0]
+        tokens_l = [int(x) for x in packed[:B].tolist()]
+        logprobs_l = packed[B:].tolist()
         any_done = False
         for i, s in enumerate(streams):
             tok = tokens_l[i]
