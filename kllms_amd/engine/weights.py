@@ -14,7 +14,7 @@ from typing import Dict
 
 import torch
 
-from ..parallel.tp import ColumnParallelLinear, ParallelContext, RowParallelLinear
+from ..parallel.tp import ParallelContext
 
 
 def _shard(t: torch.Tensor, dim: int, rank: int, world: int) -> torch.Tensor:

@@ -6,18 +6,16 @@ expert-parallel — every rank holds a 1/tp slice of every expert, so routing
 needs no all-to-all and the layer keeps the dense model's two all-reduces
 ("experts TP-sharded so no all-to-all").
 
-Expert compute is grouped: tokens are bucketed per expert and each expert's
-gate/up/down GEMMs run on its token group (rocBLAS batched path on torch; the
-HIP grouped-GEMM kernel slots in behind ops.moe_grouped once profiling says
-the bucketing overhead dominates).
+Expert compute has two regimes: decode shapes run a DENSE batched GEMM over
+all experts (same HBM traffic as routed compute when nearly every expert is
+touched; hipGraph-friendly because shapes are data-independent), prefill
+shapes bucket tokens per expert and run grouped GEMMs.
 
 Router semantics match Mixtral: softmax over ALL expert logits (fp32), top-k
 (default 2), renormalize the selected weights.
 """
 
 from __future__ import annotations
-
-from typing import Optional
 
 import torch
 import torch.nn as nn

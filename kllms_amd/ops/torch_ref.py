@@ -14,7 +14,6 @@ Conventions:
 
 from __future__ import annotations
 
-import math
 from typing import Optional, Tuple
 
 import torch
