@@ -53,6 +53,17 @@ class LocalEngineClient:
 
     # --- engine lifecycle -----------------------------------------------------
     @property
+    def scheduler(self):
+        """Continuous-batching scheduler (lazy): concurrent submissions merge
+        into one running decode batch. Shares the engine lock with the direct
+        path so both can be used."""
+        if getattr(self, "_scheduler", None) is None:
+            from .scheduler import BatchScheduler
+
+            self._scheduler = BatchScheduler(self.engine, engine_lock=self._engine_lock)
+        return self._scheduler
+
+    @property
     def engine(self):
         if self._engine is None:
             with self._engine_lock:
@@ -70,7 +81,7 @@ class LocalEngineClient:
         return self.tokenizer.crop_to_tokens(text, max_tokens)
 
     # --- core generation ------------------------------------------------------
-    def _generate(self, call_params: Dict[str, Any], constrained: bool) -> tuple:
+    def _generate(self, call_params: Dict[str, Any], constrained: bool, scheduled: bool = False) -> tuple:
         from .engine import GenRequest
 
         messages: List[Dict[str, Any]] = call_params["messages"]
@@ -103,8 +114,11 @@ class LocalEngineClient:
         prompt_ids = prompt_ids[: max(1, max_prompt)]
 
         req = GenRequest(prompt_ids=prompt_ids, n=n, sampling=sampling, constraint=constraint)
-        with self._engine_lock:
-            out = eng.generate([req])[0]
+        if scheduled:
+            out = self.scheduler.submit(req).result()
+        else:
+            with self._engine_lock:
+                out = eng.generate([req])[0]
         return out, model, sampling
 
     def _build_constraint(self, response_format: Any, constrained: bool):
@@ -216,8 +230,8 @@ class LocalEngineClient:
             )
         return results
 
-    def chat_completions_create(self, **call_params: Any) -> ChatCompletion:
-        out, model, sampling = self._generate(call_params, constrained=False)
+    def chat_completions_create(self, _scheduled: bool = False, **call_params: Any) -> ChatCompletion:
+        out, model, sampling = self._generate(call_params, constrained=False, scheduled=_scheduled)
         choices = []
         for i, s in enumerate(out.streams):
             choices.append(
@@ -237,11 +251,11 @@ class LocalEngineClient:
             timings=self._mk_timings(out),
         )
 
-    def chat_completions_parse(self, **call_params: Any) -> ParsedChatCompletion:
+    def chat_completions_parse(self, _scheduled: bool = False, **call_params: Any) -> ParsedChatCompletion:
         import json
 
         response_format = call_params.get("response_format")
-        out, model, sampling = self._generate(call_params, constrained=True)
+        out, model, sampling = self._generate(call_params, constrained=True, scheduled=_scheduled)
         choices = []
         for i, s in enumerate(out.streams):
             parsed = None
@@ -286,13 +300,15 @@ class _CompletionsNS:
         return self._client.chat_completions_create(**kw)
 
     async def acreate(self, **kw) -> ChatCompletion:
-        return await asyncio.to_thread(self._client.chat_completions_create, **kw)
+        # async requests go through the continuous-batching scheduler so
+        # concurrent awaits merge into one decode batch
+        return await asyncio.to_thread(self._client.chat_completions_create, True, **kw)
 
     def parse(self, **kw) -> ParsedChatCompletion:
         return self._client.chat_completions_parse(**kw)
 
     async def aparse(self, **kw) -> ParsedChatCompletion:
-        return await asyncio.to_thread(self._client.chat_completions_parse, **kw)
+        return await asyncio.to_thread(self._client.chat_completions_parse, True, **kw)
 
 
 class _ChatNS:

@@ -41,6 +41,7 @@ class StreamOutput:
     logprobs: List[float] = field(default_factory=list)
     text: str = ""
     finish_reason: str = "length"
+    stream_idx: int = 0  # position among the request's n streams
 
 
 @dataclass
@@ -62,7 +63,7 @@ class _Stream:
         self.seq = seq
         self.sampling = sampling
         self.seed = seed
-        self.out = StreamOutput()
+        self.out = StreamOutput(stream_idx=stream_idx)
         self.done = False
         self.step = 0
         self.constraint = constraint
@@ -254,11 +255,15 @@ class LLMEngine:
                     self.kv.free_sequence(s.seq)
             raise
 
-    def _generate_batch(
+    def begin_requests(
         self, requests: List[GenRequest], parent_seqs: List[SequenceKV], streams: List[_Stream]
     ) -> List[RequestOutput]:
+        """Phase 1 of serving: shared prefill over all prompts, KV fork into n
+        streams per request, first-token sampling. Appends the new streams to
+        ``streams`` and returns one RequestOutput per request (streams are
+        attached as they finish). Used by generate() and by the continuous-
+        batching scheduler (which interleaves this with decode steps)."""
         dev = self.device
-        t0 = time.perf_counter()
 
         # ---- shared prefill: one packed varlen batch over all prompts -------
         all_ids: List[int] = []
@@ -284,21 +289,19 @@ class LLMEngine:
         )
         input_ids = torch.tensor(all_ids, dtype=torch.long, device=dev)
         prefill_logits = self.model.forward_prefill(input_ids, batch)  # [n_req, V]
-        if dev.type == "cuda":
-            torch.cuda.synchronize()
-        t1 = time.perf_counter()
 
         # ---- fork: n streams per request share the prompt KV blocks ---------
         outputs = [RequestOutput(prompt_tokens=len(r.prompt_ids)) for r in requests]
+        new_streams: List[_Stream] = []
         for ri, req in enumerate(requests):
             base_seed = req.sampling.seed if req.sampling.seed is not None else (self.config.seed * 1000003 + ri)
             for si in range(max(1, req.n)):
                 seq = self.kv.fork(parent_seqs[ri])
                 cstate = req.constraint.init_state() if req.constraint is not None else None
-                streams.append(
-                    _Stream(ri, si, seq, req.sampling, seed=base_seed + 7919 * si,
-                            constraint=req.constraint, constraint_state=cstate)
-                )
+                st = _Stream(ri, si, seq, req.sampling, seed=base_seed + 7919 * si,
+                             constraint=req.constraint, constraint_state=cstate)
+                streams.append(st)
+                new_streams.append(st)
             self.kv.free_sequence(parent_seqs[ri])  # streams hold their own refs
 
         # ---- first token: sample n times from each request's prefill logits -
@@ -306,7 +309,26 @@ class LLMEngine:
             prefill_logits[ri].unsqueeze(0).expand(max(1, requests[ri].n), -1)
             for ri in range(len(requests))
         ])
-        self._sample_and_append(rep_logits.contiguous(), streams)
+        self._sample_and_append(rep_logits.contiguous(), new_streams)
+        return outputs
+
+    def finish_stream(self, s: _Stream) -> StreamOutput:
+        """Phase 3: detokenize (unless a stop-string already trimmed the
+        text) and release the stream's KV blocks."""
+        if not s.out.text:
+            s.out.text = self.tokenizer.decode(s.out.token_ids)
+        self.kv.free_sequence(s.seq)
+        return s.out
+
+    def _generate_batch(
+        self, requests: List[GenRequest], parent_seqs: List[SequenceKV], streams: List[_Stream]
+    ) -> List[RequestOutput]:
+        dev = self.device
+        t0 = time.perf_counter()
+        outputs = self.begin_requests(requests, parent_seqs, streams)
+        if dev.type == "cuda":
+            torch.cuda.synchronize()
+        t1 = time.perf_counter()
 
         # ---- decode loop (device-resident batch state) -----------------------
         active = [s for s in streams if not s.done]
@@ -342,8 +364,7 @@ class LLMEngine:
 
         # ---- collect ---------------------------------------------------------
         for s in streams:
-            s.out.text = self.tokenizer.decode(s.out.token_ids)
-            self.kv.free_sequence(s.seq)
+            self.finish_stream(s)
         for s in streams:
             outputs[s.req_idx].streams.append(s.out)
         for o in outputs:

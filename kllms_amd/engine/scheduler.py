@@ -1,0 +1,184 @@
+"""Continuous-batching scheduler.
+
+The production serving layer over LLMEngine: concurrent callers submit
+requests that MERGE into the running decode batch instead of waiting for the
+whole previous batch to finish. One worker thread owns the engine:
+
+    loop:
+      - drain newly-submitted requests (respecting the stream budget)
+      - prefill + fork them (engine.begin_requests) and add their streams
+        to the active set
+      - run ONE decode step over all active streams
+      - retire finished streams; resolve a request's future when its last
+        stream completes
+
+Merging is RESULT-TRANSPARENT: sampling is counter-based per (seed, step),
+so a stream draws the same random sequence regardless of which other
+requests share its batch (greedy decodes are bitwise-insensitive to batch
+composition up to GEMM reduction-order noise).
+
+The async client (AsyncKLLMs) routes through this scheduler, which is what
+makes many concurrent ``await client.chat.completions.create(...)`` calls
+share prefill batches and decode steps.
+"""
+
+from __future__ import annotations
+
+import queue
+import threading
+from concurrent.futures import Future
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+from .engine import GenRequest, LLMEngine, RequestOutput, _DecodeBatchState, _Stream
+
+
+class _WorkerContext:
+    """Mutable state owned by the scheduler worker thread."""
+
+    def __init__(self):
+        self.active: List[_Stream] = []
+        self.stream_ticket: Dict[int, "_Ticket"] = {}   # id(stream) -> ticket
+        self.state: Optional[_DecodeBatchState] = None
+
+
+@dataclass
+class _Ticket:
+    request: GenRequest
+    future: Future
+    output: Optional[RequestOutput] = None
+    remaining: int = 0
+
+
+class BatchScheduler:
+    def __init__(self, engine: LLMEngine, admit_wait_s: float = 0.002, engine_lock=None):
+        self.engine = engine
+        self.admit_wait_s = admit_wait_s
+        # serializes engine access against direct (non-scheduled) callers
+        self.engine_lock = engine_lock or threading.Lock()
+        self._queue: "queue.Queue[_Ticket]" = queue.Queue()
+        self._thread: Optional[threading.Thread] = None
+        self._lock = threading.Lock()
+        self._stop = threading.Event()
+        # observability
+        self.steps = 0
+        self.admitted_batches = 0
+
+    # --- public -----------------------------------------------------------
+    def submit(self, request: GenRequest) -> Future:
+        fut: Future = Future()
+        self._queue.put(_Ticket(request=request, future=fut))
+        self._ensure_thread()
+        return fut
+
+    def shutdown(self) -> None:
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=30)
+
+    # --- worker -----------------------------------------------------------
+    def _ensure_thread(self) -> None:
+        with self._lock:
+            if self._thread is None or not self._thread.is_alive():
+                self._stop.clear()
+                self._thread = threading.Thread(target=self._loop, name="kllms-scheduler", daemon=True)
+                self._thread.start()
+
+    def _drain(self, active_streams: int, block: bool) -> List[_Ticket]:
+        """Admit pending tickets up to the engine's stream budget."""
+        budget = self.engine.config.max_batch_size - active_streams
+        tickets: List[_Ticket] = []
+        try:
+            while budget > 0:
+                t = self._queue.get(timeout=self.admit_wait_s) if (block and not tickets) else self._queue.get_nowait()
+                need = max(1, t.request.n)
+                if tickets and need > budget:
+                    # keep order: put it back and stop admitting this round
+                    self._queue.put(t)
+                    break
+                tickets.append(t)
+                budget -= need
+        except queue.Empty:
+            pass
+        return tickets
+
+    def _loop(self) -> None:
+        import torch
+
+        ctx = _WorkerContext()
+        with torch.inference_mode():
+            while not self._stop.is_set():
+                tickets = self._drain(len(ctx.active), block=not ctx.active)
+                with self.engine_lock:
+                    if tickets:
+                        self._admit(ctx, tickets)
+                    if ctx.active:
+                        self._step(ctx)
+
+    def _admit(self, ctx: "_WorkerContext", tickets: List[_Ticket]) -> None:
+        eng = self.engine
+        reqs = [t.request for t in tickets]
+        parent_seqs: List = []
+        new_streams: List[_Stream] = []
+        try:
+            outputs = eng.begin_requests(reqs, parent_seqs, new_streams)
+        except Exception as e:
+            for seq in parent_seqs:
+                if seq.blocks:
+                    eng.kv.free_sequence(seq)
+            for st in new_streams:
+                if st.seq.blocks:
+                    eng.kv.free_sequence(st.seq)
+            for t in tickets:
+                t.future.set_exception(e)
+            return
+        for t, out, req in zip(tickets, outputs, reqs):
+            t.output = out
+            t.remaining = max(1, req.n)
+        expanded = [t for t in tickets for _ in range(max(1, t.request.n))]
+        for st, t in zip(new_streams, expanded):
+            ctx.stream_ticket[id(st)] = t
+        # first-token sampling may already have finished streams
+        self._retire(eng, new_streams, ctx.stream_ticket)
+        ctx.active.extend(st for st in new_streams if not st.done)
+        ctx.state = None
+        self.admitted_batches += 1
+
+    def _step(self, ctx: "_WorkerContext") -> None:
+        eng = self.engine
+        try:
+            if ctx.state is None:
+                ctx.state = _DecodeBatchState(eng, ctx.active)
+            logits = eng._decode_step_state(ctx.state)
+            any_done = eng._sample_state(ctx.state, logits)
+            self.steps += 1
+        except Exception as e:
+            # poisoned batch: fail every in-flight request, free KV
+            for st in ctx.active:
+                if st.seq.blocks:
+                    eng.kv.free_sequence(st.seq)
+                t = ctx.stream_ticket.pop(id(st), None)
+                if t is not None and not t.future.done():
+                    t.future.set_exception(e)
+            ctx.active = []
+            ctx.state = None
+            return
+        if any_done:
+            self._retire(eng, ctx.active, ctx.stream_ticket)
+            ctx.active = [st for st in ctx.active if not st.done]
+            ctx.state = None
+
+    def _retire(self, eng: LLMEngine, streams: List[_Stream], stream_ticket: Dict[int, "_Ticket"]) -> None:
+        for s in streams:
+            if not s.done:
+                continue
+            t = stream_ticket.pop(id(s), None)
+            if t is None:
+                continue
+            eng.finish_stream(s)
+            t.output.streams.append(s.out)
+            t.remaining -= 1
+            if t.remaining == 0:
+                # order by stream index for API parity with generate()
+                t.output.streams.sort(key=lambda o: o.stream_idx)
+                t.future.set_result(t.output)
