@@ -127,6 +127,11 @@ class LLMEngine:
         if self.device.type == "cpu" and self.dtype == torch.bfloat16:
             self.dtype = torch.float32
 
+        # the RoPE cos/sin table (and thus every position) is bounded by the
+        # architecture's max_position_embeddings
+        if config.max_seq_len > self.arch.max_position_embeddings:
+            config.max_seq_len = self.arch.max_position_embeddings
+
         self.tokenizer: BaseTokenizer = load_tokenizer(config.model, config.weights_path, self.arch.vocab_size)
         self.eos_token_id = config.eos_token_id if config.eos_token_id is not None else self.tokenizer.eos_id
 
@@ -464,7 +469,7 @@ class LLMEngine:
             s.out.finish_reason = "stop"
             s.done = True
             return
-        if len(s.out.token_ids) >= max_new:
+        if len(s.out.token_ids) >= max_new or s.seq.num_tokens >= self.config.max_seq_len - 1:
             s.out.finish_reason = "length"
             s.done = True
             return
