@@ -187,7 +187,10 @@ def main():
                 "max_new_tokens": args.max_new,
                 "parallelism": f"dp{n_ranks} tp1",
                 "completions_per_s": round(value * args.n, 3),
+                # amortized (throughput) latency and the user-perceived batch
+                # wall time (a sync batch's requests all complete together)
                 "consensus_latency_s_per_request": round(elapsed / (args.steps * args.batch), 4),
+                "batch_wall_s": round(ms_per_step / 1000.0, 4),
                 "mean_consensus_quality": round(sum(quality_scores) / len(quality_scores), 4) if quality_scores else None,
             },
         }

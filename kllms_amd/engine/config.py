@@ -101,7 +101,7 @@ class EngineConfig(BaseModel):
 
     # Decode-step hipGraph capture
     use_hip_graphs: bool = True
-    hip_graph_batch_sizes: list[int] = Field(default_factory=lambda: [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128])
+    hip_graph_batch_sizes: list[int] = Field(default_factory=lambda: [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256])
 
     # Generation defaults
     default_max_new_tokens: int = 512
