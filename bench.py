@@ -68,7 +68,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--model", type=str, default="llama-3-8b")
     ap.add_argument("--n", type=int, default=5)
-    ap.add_argument("--batch", type=int, default=16, help="consensus requests per step per GPU")
+    ap.add_argument("--batch", type=int, default=24, help="consensus requests per step per GPU")
     ap.add_argument("--prompt-len", type=int, default=512, help="approx prompt tokens")
     ap.add_argument("--max-new", type=int, default=64, help="decode tokens per stream")
     ap.add_argument("--no-graphs", action="store_true")
