@@ -71,16 +71,32 @@ _HEX = set(b"0123456789abcdefABCDEF")
 _STR_CHAR = {b for b in range(0x20, 0x7F) if b not in (0x22, 0x5C)}
 
 
-def _json_string_ir() -> _Node:
-    escape = Seq([
-        Lit(b"\\"),
-        Alt([
-            Cls(set(b'"\\/bfnrt')),
-            Seq([Lit(b"u"), Cls(_HEX), Cls(_HEX), Cls(_HEX), Cls(_HEX)]),
-        ]),
-    ])
-    body = Star(Alt([Cls(_STR_CHAR), escape]))
-    return Seq([Lit(b'"'), body, Lit(b'"')])
+def _bounded(inner_factory, min_n: int, max_n: Optional[int]) -> _Node:
+    """min_n required occurrences then (max_n - min_n) optional ones, or a
+    Star tail when max_n is None."""
+    parts: List[_Node] = [inner_factory() for _ in range(min_n)]
+    if max_n is None:
+        parts.append(Star(inner_factory()))
+    else:
+        for _ in range(max_n - min_n):
+            parts.append(Opt(inner_factory()))
+    return Seq(parts)
+
+
+def _json_string_ir(min_len: int = 0, max_len: Optional[int] = None) -> _Node:
+    def char() -> _Node:
+        return Alt([
+            Cls(_STR_CHAR),
+            Seq([
+                Lit(b"\\"),
+                Alt([
+                    Cls(set(b'"\\/bfnrt')),
+                    Seq([Lit(b"u"), Cls(_HEX), Cls(_HEX), Cls(_HEX), Cls(_HEX)]),
+                ]),
+            ]),
+        ])
+
+    return Seq([Lit(b'"'), _bounded(char, min_len, max_len), Lit(b'"')])
 
 
 def _integer_ir() -> _Node:
@@ -121,7 +137,7 @@ def schema_to_ir(schema: Dict[str, Any], defs: Dict[str, Any], depth: int = 0) -
     if isinstance(t, list):
         return Alt([schema_to_ir({**schema, "type": ti}, defs, depth + 1) for ti in t])
     if t == "string":
-        return _json_string_ir()
+        return _json_string_ir(schema.get("minLength", 0), schema.get("maxLength"))
     if t == "integer":
         return _integer_ir()
     if t == "number":
@@ -132,9 +148,20 @@ def schema_to_ir(schema: Dict[str, Any], defs: Dict[str, Any], depth: int = 0) -
         return Lit(b"null")
     if t == "array":
         item = schema.get("items", {})
-        item_ir = schema_to_ir(item, defs, depth + 1) if item else _any_value_ir(defs, depth + 1)
-        more = Star(Seq([Lit(b","), item_ir]))
-        return Seq([Lit(b"["), Opt(Seq([item_ir, more])), Lit(b"]")])
+
+        def item_ir() -> _Node:
+            return schema_to_ir(item, defs, depth + 1) if item else _any_value_ir(defs, depth + 1)
+
+        min_items = schema.get("minItems", 0)
+        max_items = schema.get("maxItems")
+        if min_items == 0:
+            more = _bounded(lambda: Seq([Lit(b","), item_ir()]), 0,
+                            None if max_items is None else max(0, max_items - 1))
+            return Seq([Lit(b"["), Opt(Seq([item_ir(), more])), Lit(b"]")])
+        head = [item_ir()] + [Seq([Lit(b","), item_ir()]) for _ in range(min_items - 1)]
+        more = _bounded(lambda: Seq([Lit(b","), item_ir()]), 0,
+                        None if max_items is None else max(0, max_items - min_items))
+        return Seq([Lit(b"[")] + head + [more, Lit(b"]")])
     if t == "object" or "properties" in schema:
         props = schema.get("properties", {})
         if not props:

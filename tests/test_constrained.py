@@ -162,3 +162,37 @@ class TestEndToEnd:
                 # length-capped stream: prefix of valid JSON by construction
                 assert s.text.startswith("{")
         assert n_valid >= 1  # with 300 tokens most streams close the object
+
+
+class TestBounds:
+    def test_string_max_length(self):
+        sch = {"type": "string", "maxLength": 3}
+        assert accepts(sch, '"ab"')
+        assert accepts(sch, '"abc"')
+        assert not accepts(sch, '"abcd"')
+
+    def test_string_min_length(self):
+        sch = {"type": "string", "minLength": 2, "maxLength": 4}
+        assert not accepts(sch, '"a"')
+        assert accepts(sch, '"ab"')
+        assert accepts(sch, '"abcd"')
+        assert not accepts(sch, '"abcde"')
+
+    def test_array_bounds(self):
+        sch = {"type": "array", "items": {"type": "integer"}, "minItems": 1, "maxItems": 3}
+        assert not accepts(sch, "[]")
+        assert accepts(sch, "[1]")
+        assert accepts(sch, "[1,2,3]")
+        assert not accepts(sch, "[1,2,3,4]")
+
+    def test_pydantic_field_constraints_flow_through(self):
+        from pydantic import BaseModel, Field
+
+        class Doc(BaseModel):
+            tag: str = Field(max_length=4)
+            nums: list[int] = Field(max_length=2)  # -> maxItems
+
+        sch = Doc.model_json_schema()
+        assert accepts(sch, '{"tag":"abcd","nums":[1,2]}')
+        assert not accepts(sch, '{"tag":"abcde","nums":[1,2]}')
+        assert not accepts(sch, '{"tag":"ab","nums":[1,2,3]}')
