@@ -1,0 +1,123 @@
+"""Weight-loading tests: HF-style safetensors -> fused/sharded layout, and
+the HF tokenizers-file path."""
+
+import json
+import os
+
+import pytest
+import torch
+
+from kllms_amd.engine.config import MODEL_PRESETS
+from kllms_amd.engine.weights import load_safetensors_weights
+from kllms_amd.models.llama import LlamaForCausalLM
+from kllms_amd.parallel.tp import ParallelContext
+
+
+def _make_hf_llama_checkpoint(tmp_path, cfg, seed=0):
+    """Write a random HF-style (unfused q/k/v, gate/up) checkpoint."""
+    from safetensors.torch import save_file
+
+    g = torch.Generator().manual_seed(seed)
+    H = cfg.hidden_size
+    D = cfg.head_dim_
+    tensors = {
+        "model.embed_tokens.weight": torch.randn(cfg.vocab_size, H, generator=g),
+        "model.norm.weight": torch.randn(H, generator=g),
+        "lm_head.weight": torch.randn(cfg.vocab_size, H, generator=g),
+    }
+    for L in range(cfg.num_layers):
+        p = f"model.layers.{L}"
+        tensors[f"{p}.self_attn.q_proj.weight"] = torch.randn(cfg.num_heads * D, H, generator=g)
+        tensors[f"{p}.self_attn.k_proj.weight"] = torch.randn(cfg.num_kv_heads * D, H, generator=g)
+        tensors[f"{p}.self_attn.v_proj.weight"] = torch.randn(cfg.num_kv_heads * D, H, generator=g)
+        tensors[f"{p}.self_attn.o_proj.weight"] = torch.randn(H, cfg.num_heads * D, generator=g)
+        tensors[f"{p}.mlp.gate_proj.weight"] = torch.randn(cfg.intermediate_size, H, generator=g)
+        tensors[f"{p}.mlp.up_proj.weight"] = torch.randn(cfg.intermediate_size, H, generator=g)
+        tensors[f"{p}.mlp.down_proj.weight"] = torch.randn(H, cfg.intermediate_size, generator=g)
+        tensors[f"{p}.input_layernorm.weight"] = torch.randn(H, generator=g)
+        tensors[f"{p}.post_attention_layernorm.weight"] = torch.randn(H, generator=g)
+    save_file({k: v.contiguous() for k, v in tensors.items()}, str(tmp_path / "model.safetensors"))
+    return tensors
+
+
+def test_safetensors_load_fuses_correctly(tmp_path):
+    pytest.importorskip("safetensors")
+    cfg = MODEL_PRESETS["tiny-llama"]
+    tensors = _make_hf_llama_checkpoint(tmp_path, cfg)
+
+    model = LlamaForCausalLM(cfg, ParallelContext(), dtype=torch.float32)
+    load_safetensors_weights(model, str(tmp_path), ParallelContext())
+
+    # fused qkv = [q; k; v]
+    got = model.layers[0].self_attn.qkv_proj.weight
+    want = torch.cat([
+        tensors["model.layers.0.self_attn.q_proj.weight"],
+        tensors["model.layers.0.self_attn.k_proj.weight"],
+        tensors["model.layers.0.self_attn.v_proj.weight"],
+    ], dim=0)
+    assert torch.equal(got, want)
+    # fused gate_up = [gate; up]
+    got = model.layers[1].mlp.gate_up_proj.weight
+    want = torch.cat([
+        tensors["model.layers.1.mlp.gate_proj.weight"],
+        tensors["model.layers.1.mlp.up_proj.weight"],
+    ], dim=0)
+    assert torch.equal(got, want)
+    assert torch.equal(model.lm_head.weight, tensors["lm_head.weight"])
+    assert torch.equal(model.norm.weight, tensors["model.norm.weight"])
+
+
+def test_engine_boots_from_weights_dir(tmp_path):
+    pytest.importorskip("safetensors")
+    cfg = MODEL_PRESETS["tiny-llama"]
+    _make_hf_llama_checkpoint(tmp_path, cfg)
+    # HF-style config.json so resolve_arch() reads the architecture
+    (tmp_path / "config.json").write_text(json.dumps({
+        "model_type": "llama",
+        "vocab_size": cfg.vocab_size,
+        "hidden_size": cfg.hidden_size,
+        "intermediate_size": cfg.intermediate_size,
+        "num_hidden_layers": cfg.num_layers,
+        "num_attention_heads": cfg.num_heads,
+        "num_key_value_heads": cfg.num_kv_heads,
+        "rope_theta": cfg.rope_theta,
+        "rms_norm_eps": cfg.rms_norm_eps,
+        "max_position_embeddings": cfg.max_position_embeddings,
+    }))
+
+    from kllms_amd.engine.config import EngineConfig
+    from kllms_amd.engine.engine import GenRequest, LLMEngine
+    from kllms_amd.engine.sampling import SamplingParams
+
+    eng = LLMEngine(EngineConfig(
+        model=str(tmp_path), weights_path=str(tmp_path),
+        max_kv_blocks=128, use_hip_graphs=False, device="cpu",
+    ))
+    out = eng.generate([GenRequest(prompt_ids=[1, 2, 3], n=1,
+                                   sampling=SamplingParams(temperature=0.0, max_tokens=4))])[0]
+    assert len(out.streams[0].token_ids) > 0
+
+
+def test_hf_tokenizer_roundtrip(tmp_path):
+    tokenizers = pytest.importorskip("tokenizers")
+    from tokenizers import Tokenizer, models, pre_tokenizers, trainers
+
+    tok = Tokenizer(models.BPE(unk_token="<unk>"))
+    tok.pre_tokenizer = pre_tokenizers.Whitespace()
+    trainer = trainers.BpeTrainer(special_tokens=["<unk>", "<|begin_of_text|>", "<|eot_id|>"], vocab_size=200)
+    tok.train_from_iterator(["hello world", "the quick brown fox", "hello fox"] * 20, trainer)
+    path = tmp_path / "tokenizer.json"
+    tok.save(str(path))
+
+    from kllms_amd.engine.tokenizer import HFTokenizer, load_tokenizer
+
+    ht = HFTokenizer(str(path))
+    ids = ht.encode("hello world")
+    assert ids
+    assert "hello" in ht.decode(ids)
+    assert ht.crop_to_tokens("hello world", 1)  # crops without error
+    assert ht.token_bytes(ids[0])
+
+    # load_tokenizer picks the file up from a weights dir
+    lt = load_tokenizer(str(tmp_path), None, 512)
+    assert isinstance(lt, HFTokenizer)
