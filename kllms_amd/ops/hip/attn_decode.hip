@@ -26,7 +26,7 @@
 
 #define TKV ATTN_DECODE_TKV
 #define NTHREADS 256
-#define KLANES 8           // lanes cooperating on one key's dot product
+#define KLANES 4           // lanes cooperating on one key's dot product
 #define CHUNK ATTN_DECODE_CHUNK
 
 // partials layout: [B, KVH, max_chunks, gqa, 130]: 128 o values + m + l
@@ -84,29 +84,33 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
     }
     __syncthreads();
 
-    // ---- scores: KLANES lanes per key, 16 bf16 per lane ------------------
-    const int kg = tid / KLANES;
-    const int kl = tid % KLANES;
-    for (int key0 = 0; key0 < nkeys; key0 += NTHREADS / KLANES) {
-      const int key = key0 + kg;
-      if (key < nkeys) {
-        const bf16x8_vec* kv8 = reinterpret_cast<const bf16x8_vec*>(k_cache + rowoff[key]) + kl * 2;
-        bf16x8_vec ka = kv8[0];
-        bf16x8_vec kb = kv8[1];
-        float af[16];
+    // ---- scores: KLANES lanes per key, 128/KLANES bf16 per lane -----------
+    {
+      constexpr int ELEMS = 128 / KLANES;     // per-lane K elements
+      constexpr int VECS = ELEMS / 8;         // 16-byte pieces per lane
+      const int kg = tid / KLANES;
+      const int kl = tid % KLANES;
+      for (int key0 = 0; key0 < nkeys; key0 += NTHREADS / KLANES) {
+        const int key = key0 + kg;
+        if (key < nkeys) {
+          const bf16x8_vec* kv8 = reinterpret_cast<const bf16x8_vec*>(k_cache + rowoff[key]) + kl * VECS;
+          float af[ELEMS];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) af[j] = bf16_to_f32(ka[j]);
+          for (int c = 0; c < VECS; ++c) {
+            bf16x8_vec kv = kv8[c];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) af[8 + j] = bf16_to_f32(kb[j]);
+            for (int j = 0; j < 8; ++j) af[c * 8 + j] = bf16_to_f32(kv[j]);
+          }
 #pragma unroll
-        for (int g = 0; g < GQA; ++g) {
-          const float* qg = q_lds + g * 128 + kl * 16;
-          float acc = 0.f;
+          for (int g = 0; g < GQA; ++g) {
+            const float* qg = q_lds + g * 128 + kl * ELEMS;
+            float acc = 0.f;
 #pragma unroll
-          for (int j = 0; j < 16; ++j) acc += af[j] * qg[j];
+            for (int j = 0; j < ELEMS; ++j) acc += af[j] * qg[j];
 #pragma unroll
-          for (int off = 4; off > 0; off >>= 1) acc += __shfl_xor(acc, off);
-          if (kl == 0) s_lds[g * TKV + key] = acc;
+            for (int off = KLANES / 2; off > 0; off >>= 1) acc += __shfl_xor(acc, off);
+            if (kl == 0) s_lds[g * TKV + key] = acc;
+          }
         }
       }
     }
