@@ -27,7 +27,7 @@
 #define TKV ATTN_DECODE_TKV
 #define NTHREADS 256
 #define KLANES 4           // lanes cooperating on one key's dot product
-#define CHUNK ATTN_DECODE_CHUNK
+// chunk size is a runtime arg (adaptive split-K: big batches need no split)
 
 // partials layout: [B, KVH, max_chunks, gqa, 130]: 128 o values + m + l
 #define PART_STRIDE 130
@@ -41,14 +41,14 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
     const int* __restrict__ block_tables,  // [B, max_blocks]
     const int* __restrict__ context_lens,  // [B]
     float scale, int num_kv_heads,
-    int block_size, int max_blocks, int max_chunks, int q_tstride) {
+    int block_size, int max_blocks, int max_chunks, int q_tstride, int chunk_keys) {
   const int b = blockIdx.x;
   const int g_kv = blockIdx.y;
   const int chunk = blockIdx.z;
   const int L = context_lens[b];
-  const int c0 = chunk * CHUNK;
+  const int c0 = chunk * chunk_keys;
   if (c0 >= L && chunk > 0) return;     // no keys for this chunk
-  const int c1 = min(L, c0 + CHUNK);
+  const int c1 = min(L, c0 + chunk_keys);
   const int tid = threadIdx.x;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -224,7 +224,7 @@ extern "C" void launch_attn_decode_partial(
     float* partials, const bf16_t* q, const bf16_t* k_cache, const bf16_t* v_cache,
     const int* block_tables, const int* context_lens, float scale,
     int num_q_heads, int num_kv_heads, int block_size, int max_blocks,
-    int max_chunks, int q_tstride, int B, hipStream_t stream) {
+    int max_chunks, int q_tstride, int chunk_keys, int B, hipStream_t stream) {
   const int gqa = num_q_heads / num_kv_heads;
   const dim3 grid(B, num_kv_heads, max_chunks);
   const size_t lds = (gqa * 128 + gqa * TKV + 24 + 4 * gqa * 128) * sizeof(float) +
@@ -232,7 +232,7 @@ extern "C" void launch_attn_decode_partial(
 #define LAUNCH(G)                                                                  \
   hipLaunchKernelGGL(attn_decode_partial_t<G>, grid, dim3(NTHREADS), lds, stream,  \
                      partials, q, k_cache, v_cache, block_tables, context_lens,    \
-                     scale, num_kv_heads, block_size, max_blocks, max_chunks, q_tstride)
+                     scale, num_kv_heads, block_size, max_blocks, max_chunks, q_tstride, chunk_keys)
   switch (gqa) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;
@@ -248,14 +248,14 @@ extern "C" __global__ void __launch_bounds__(64) attn_decode_reduce_kernel(
     bf16_t* __restrict__ out,            // [B, H, 128]
     const float* __restrict__ partials,  // [B, KVH, max_chunks, gqa, 130]
     const int* __restrict__ context_lens,
-    int num_q_heads, int num_kv_heads, int max_chunks) {
+    int num_q_heads, int num_kv_heads, int max_chunks, int chunk_keys) {
   const int b = blockIdx.x;
   const int h = blockIdx.y;
   const int gqa = num_q_heads / num_kv_heads;
   const int g_kv = h / gqa;
   const int g = h % gqa;
   const int lane = threadIdx.x;
-  const int nchunks = min(max_chunks, (context_lens[b] + CHUNK - 1) / CHUNK);
+  const int nchunks = min(max_chunks, (context_lens[b] + chunk_keys - 1) / chunk_keys);
 
   const float* base = partials +
       ((((int64_t)b * num_kv_heads + g_kv) * max_chunks) * gqa + g) * PART_STRIDE;

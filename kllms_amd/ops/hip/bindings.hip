@@ -12,8 +12,8 @@ extern "C" __global__ void fused_add_rmsnorm_kernel(bf16_t*, const bf16_t*, bf16
 extern "C" __global__ void silu_mul_kernel(bf16_t*, const bf16_t*, const bf16_t*, int64_t, int, int);
 extern "C" __global__ void rope_kernel(bf16_t*, bf16_t*, const int64_t*, const float*, int, int, int, int, int);
 extern "C" __global__ void store_kv_kernel(const bf16_t*, const bf16_t*, bf16_t*, bf16_t*, const int64_t*, int, int, int, int, int);
-extern "C" void launch_attn_decode_partial(float*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, float, int, int, int, int, int, int, int, hipStream_t);
-extern "C" __global__ void attn_decode_reduce_kernel(bf16_t*, const float*, const int*, int, int, int);
+extern "C" void launch_attn_decode_partial(float*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, float, int, int, int, int, int, int, int, int, hipStream_t);
+extern "C" __global__ void attn_decode_reduce_kernel(bf16_t*, const float*, const int*, int, int, int, int);
 extern "C" __global__ void attn_prefill_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, const int*, float, int, int, int, int);  // grouped: [G], [G], [G*4]
 extern "C" __global__ void sample_kernel(int64_t*, float*, const float*, const float*, const float*, const int*, const int64_t*, const int64_t*, const uint32_t*, int);
 extern "C" __global__ void mfma_selftest_kernel(float*, const bf16_t*, const bf16_t*);
@@ -100,17 +100,24 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache
               "attn_decode: GQA group must be 1, 2, 4 or 8");
   const int gqa = H / KVH;
   const int max_blocks = block_tables.size(1);
-  const int CHUNK_KEYS = ATTN_DECODE_CHUNK;
-  const int max_chunks = std::max(1, (max_blocks * BS + CHUNK_KEYS - 1) / CHUNK_KEYS);
+  // adaptive split-K: split only as far as needed to fill the chip
+  // (~1024 workgroups); large batches get one chunk per (seq, kv-head) and
+  // pay no partial-combine traffic beyond the normalize pass
+  const int TILE = ATTN_DECODE_TKV;
+  const int max_ctx = max_blocks * BS;
+  const int per_seq = std::max(1, 1024 / std::max(1, B * KVH));
+  int CHUNK_KEYS = (max_ctx + per_seq - 1) / per_seq;
+  CHUNK_KEYS = std::max(TILE, ((CHUNK_KEYS + TILE - 1) / TILE) * TILE);
+  const int max_chunks = std::max(1, (max_ctx + CHUNK_KEYS - 1) / CHUNK_KEYS);
   auto partials = torch::empty({(int64_t)B * KVH * max_chunks * gqa * 130},
                                torch::dtype(torch::kFloat).device(q.device()));
   launch_attn_decode_partial(
       partials.data_ptr<float>(), cbf(q), cbf(k_cache), cbf(v_cache),
       block_tables.data_ptr<int>(), context_lens.data_ptr<int>(),
-      (float)scale, H, KVH, BS, max_blocks, max_chunks, (int)q.stride(0), B, cur_stream());
+      (float)scale, H, KVH, BS, max_blocks, max_chunks, (int)q.stride(0), CHUNK_KEYS, B, cur_stream());
   hipLaunchKernelGGL(attn_decode_reduce_kernel, dim3(B, H), dim3(64), 0, cur_stream(),
                      bf(out), partials.data_ptr<float>(), context_lens.data_ptr<int>(),
-                     H, KVH, max_chunks);
+                     H, KVH, max_chunks, CHUNK_KEYS);
 }
 
 void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k, torch::Tensor v,
