@@ -100,14 +100,11 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache
               "attn_decode: GQA group must be 1, 2, 4 or 8");
   const int gqa = H / KVH;
   const int max_blocks = block_tables.size(1);
-  // adaptive split-K: split only as far as needed to fill the chip
-  // (~4096 workgroups = 16 per CU keeps latency hidden; measured: dropping
-  // to 1024 bigger blocks LOSES 5% at B=240 despite less combine traffic)
-  const int TILE = ATTN_DECODE_TKV;
+  // split-K granule: fixed 256 keys measured best across B=40..240 and
+  // ctx 576..2048 (adaptive coarser chunks trade combine traffic for lost
+  // block-level parallelism and net 0..-5%; chunk size stays a runtime arg)
   const int max_ctx = max_blocks * BS;
-  const int per_seq = std::max(1, 4096 / std::max(1, B * KVH));
-  int CHUNK_KEYS = (max_ctx + per_seq - 1) / per_seq;
-  CHUNK_KEYS = std::max(TILE, ((CHUNK_KEYS + TILE - 1) / TILE) * TILE);
+  const int CHUNK_KEYS = 2 * ATTN_DECODE_TKV;
   const int max_chunks = std::max(1, (max_ctx + CHUNK_KEYS - 1) / CHUNK_KEYS);
   auto partials = torch::empty({(int64_t)B * KVH * max_chunks * gqa * 130},
                                torch::dtype(torch::kFloat).device(q.device()));
