@@ -104,7 +104,7 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache
   // ctx 576..2048 (adaptive coarser chunks trade combine traffic for lost
   // block-level parallelism and net 0..-5%; chunk size stays a runtime arg)
   const int max_ctx = max_blocks * BS;
-  const int CHUNK_KEYS = 2 * ATTN_DECODE_TKV;
+  const int CHUNK_KEYS = ATTN_DECODE_TKV;  // 256
   const int max_chunks = std::max(1, (max_ctx + CHUNK_KEYS - 1) / CHUNK_KEYS);
   auto partials = torch::empty({(int64_t)B * KVH * max_chunks * gqa * 130},
                                torch::dtype(torch::kFloat).device(q.device()));
