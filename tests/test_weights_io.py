@@ -121,3 +121,34 @@ def test_hf_tokenizer_roundtrip(tmp_path):
     # load_tokenizer picks the file up from a weights dir
     lt = load_tokenizer(str(tmp_path), None, 512)
     assert isinstance(lt, HFTokenizer)
+
+
+def test_safetensors_tp2_shards(tmp_path):
+    """TP=2 loading: each rank's fused qkv holds its q/k/v HEAD shards (not a
+    contiguous slice of the fused dim)."""
+    pytest.importorskip("safetensors")
+    cfg = MODEL_PRESETS["tiny-llama"]
+    tensors = _make_hf_llama_checkpoint(tmp_path, cfg, seed=5)
+    D = cfg.head_dim_
+    q_full = tensors["model.layers.0.self_attn.q_proj.weight"]
+    k_full = tensors["model.layers.0.self_attn.k_proj.weight"]
+    v_full = tensors["model.layers.0.self_attn.v_proj.weight"]
+
+    for rank in (0, 1):
+        ctx = ParallelContext(world_size=2, rank=rank)
+        model = LlamaForCausalLM(cfg, ctx, dtype=torch.float32)
+        load_safetensors_weights(model, str(tmp_path), ctx)
+        got = model.layers[0].self_attn.qkv_proj.weight
+        qh = cfg.num_heads // 2 * D
+        kh = cfg.num_kv_heads // 2 * D
+        want = torch.cat([
+            q_full[rank * qh:(rank + 1) * qh],
+            k_full[rank * kh:(rank + 1) * kh],
+            v_full[rank * kh:(rank + 1) * kh],
+        ], dim=0)
+        assert torch.equal(got, want), f"rank {rank} qkv shard mismatch"
+        # row-parallel down_proj: columns sharded
+        down_full = tensors["model.layers.0.mlp.down_proj.weight"]
+        got_down = model.layers[0].mlp.down_proj.weight
+        half = cfg.intermediate_size // 2
+        assert torch.equal(got_down, down_full[:, rank * half:(rank + 1) * half])
