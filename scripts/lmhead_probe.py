@@ -26,7 +26,7 @@ def bench(fn, iters=50):
 def main():
     V, H = 128256, 4096
     W = torch.randn(V, H, dtype=torch.bfloat16, device="cuda") * 0.02
-    for B in (8, 20, 40, 80):
+    for B in (40, 80, 120, 160, 240):
         x = torch.randn(B, H, dtype=torch.bfloat16, device="cuda")
         t_full = bench(lambda: torch.nn.functional.linear(x, W))
         results = {"full": t_full}
@@ -40,9 +40,11 @@ def main():
                     out[:, i * cs:(i + 1) * cs] = torch.nn.functional.linear(x, c)
 
             results[f"chunk{nchunk}"] = bench(chunked)
-        # matmul with pre-transposed weight (different Tensile path)
-        Wt = W.t().contiguous()
-        results["x@Wt"] = bench(lambda: x @ Wt)
+        # row-split halves (each M/2 may dodge the stream-K pick)
+        def rowsplit():
+            torch.nn.functional.linear(x[: B // 2], W)
+            torch.nn.functional.linear(x[B // 2:], W)
+        results["rowsplit"] = bench(rowsplit)
         print(f"B={B}: " + ", ".join(f"{k}={v:.0f}us" for k, v in results.items()), flush=True)
 
 
