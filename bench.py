@@ -54,6 +54,30 @@ def _mean_likelihood(node) -> float:
     return sum(leaves) / len(leaves) if leaves else 0.0
 
 
+def _enable_tunableop(local_rank: int) -> None:
+    """Load the shipped offline GEMM tuning results (read-only). The decode-
+    shape GEMMs (M = graph bucket) are ~30% faster under the tuned algorithm
+    picks than the default heuristics (see kllms_amd/tunableop/)."""
+    import shutil
+    import tempfile
+
+    shipped = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                           "kllms_amd", "tunableop", "tunableop_gfx9500.csv")
+    if not os.path.exists(shipped):
+        return
+    tmp = tempfile.mkdtemp(prefix="kllms_tunableop_")
+    # TunableOp inserts the device ordinal before ".csv"; provide every ordinal
+    for i in range(8):
+        shutil.copy(shipped, os.path.join(tmp, f"tunableop_gfx950{i}.csv"))
+    try:
+        torch.cuda.tunable.enable(True)
+        torch.cuda.tunable.tuning_enable(False)
+        torch.cuda.tunable.set_filename(os.path.join(tmp, "tunableop_gfx950.csv"), insert_device_ordinal=True)
+        torch.cuda.tunable.read_file()
+    except Exception as e:
+        log(f"[bench] tunableop disabled ({e})")
+
+
 def make_prompt(rank: int, step: int, i: int, prompt_len_tokens: int) -> str:
     # deterministic synthetic prompt of roughly prompt_len_tokens byte-tokens
     seedtxt = f"Request {rank}-{step}-{i}: extract the entities. "
@@ -89,6 +113,9 @@ def main():
             torch.cuda.set_device(local_rank)
 
     device = f"cuda:{local_rank}" if on_gpu else "cpu"
+
+    if on_gpu:
+        _enable_tunableop(local_rank)
 
     from kllms_amd import KLLMs
     from kllms_amd.consensus.consolidation import consolidate_chat_completions
