@@ -71,9 +71,18 @@ def _enable_tunableop(local_rank: int) -> None:
         shutil.copy(shipped, os.path.join(tmp, f"tunableop_gfx950{i}.csv"))
     try:
         torch.cuda.tunable.enable(True)
-        torch.cuda.tunable.tuning_enable(False)
+        # KLLMS_TUNE=1 re-runs the offline tuning and leaves results in the
+        # temp dir (printed) for refreshing kllms_amd/tunableop/.
+        tune = os.environ.get("KLLMS_TUNE", "0") == "1"
+        torch.cuda.tunable.tuning_enable(tune)
         torch.cuda.tunable.set_filename(os.path.join(tmp, "tunableop_gfx950.csv"), insert_device_ordinal=True)
         torch.cuda.tunable.read_file()
+        if tune:
+            log(f"[bench] tunableop TUNING, results dir: {tmp}")
+            import atexit
+
+            atexit.register(torch.cuda.tunable.write_file)
+            atexit.register(lambda: log(f"[bench] tuned results in {tmp}"))
     except Exception as e:
         log(f"[bench] tunableop disabled ({e})")
 
