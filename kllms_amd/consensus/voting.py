@@ -33,7 +33,17 @@ def voting_consensus(
     is_boolean = isinstance(first_non_none, bool)
 
     if is_boolean:
-        processed_values = [v or False for v in values]
+        processed_values = []
+        for v in values:
+            v2 = v or False  # None counts as False (ref :954-958)
+            try:
+                hash(v2)
+            except TypeError:
+                # hardening over the reference (which crashes in Counter):
+                # a truthy unhashable candidate (list/dict mixed in with
+                # bools) votes as True
+                v2 = True
+            processed_values.append(v2)
         counts = Counter(processed_values)
         best_val, best_count = counts.most_common(1)[0]
     else:
