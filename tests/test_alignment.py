@@ -120,3 +120,65 @@ def test_sort_by_original_majority_basic():
     sorted_lists, idx = sort_by_original_majority(aligned, originals)
     assert sorted_lists[0] == ["a", "b", "c"]
     assert idx[0] == [0, 1, 2]
+
+
+class TestAlignmentInternals:
+    """Targeted tests for the alignment internals' observable quirks
+    (reference consensus_utils.py:185-333)."""
+
+    def test_dynamic_threshold_single_list(self):
+        from kllms_amd.consensus.alignment import SimilarityCache, _compute_dynamic_threshold
+
+        cache = SimilarityCache(sim_fn, [["a"]])
+        assert _compute_dynamic_threshold(cache) == 0.5  # <2 lists -> base
+
+    def test_dynamic_threshold_tracks_best_matches(self):
+        from kllms_amd.consensus.alignment import SimilarityCache, _compute_dynamic_threshold
+
+        # identical elements across lists -> best-match sims all 1.0 ->
+        # threshold = max(0.5, 0.95 * 1.0)
+        lists = [["alpha", "beta"], ["alpha", "beta"], ["alpha", "beta"]]
+        thr = _compute_dynamic_threshold(SimilarityCache(sim_fn, lists))
+        assert thr == pytest.approx(0.95)
+
+    def test_reference_build_groups_by_support(self):
+        from kllms_amd.consensus.alignment import SimilarityCache, _build_reference_list
+
+        lists = [["apple"], ["apple"], ["apple", "zebra"]]
+        cache = SimilarityCache(sim_fn, lists)
+        refs = _build_reference_list(cache, None, min_support_ratio=0.5, threshold=0.8)
+        # "apple" group has support 3/3 >= 0.5; "zebra" 1/3 < 0.5 pruned
+        assert len(refs) == 1
+        li, pos = refs[0]
+        assert lists[li][pos] == "apple"
+
+    def test_reference_reelection_uses_index_medoid(self):
+        from kllms_amd.consensus.alignment import SimilarityCache, _build_reference_list
+
+        # three near-identical strings: the re-elected representative is the
+        # medoid of the (list_idx, pos) INDEX TUPLES (reference quirk :308-318)
+        lists = [["colour"], ["colour"], ["colour"]]
+        cache = SimilarityCache(sim_fn, lists)
+        refs = _build_reference_list(cache, None, min_support_ratio=0.5, threshold=0.8)
+        assert len(refs) == 1
+        assert refs[0] in [(0, 0), (1, 0), (2, 0)]
+
+    def test_hungarian_respects_threshold(self):
+        from kllms_amd.consensus.alignment import (
+            SimilarityCache,
+            _align_lists_to_reference_hungarian,
+        )
+
+        lists = [["aaaa"], ["zzzz"]]
+        cache = SimilarityCache(sim_fn, lists)
+        aligned = _align_lists_to_reference_hungarian(cache, [(0, 0)], threshold=0.9)
+        assert aligned[0] == ["aaaa"]      # the reference element itself (sim 1.0)
+        assert aligned[1] == [None]        # "zzzz" below threshold
+
+    def test_prune_keeps_highest_when_all_below(self):
+        from kllms_amd.consensus.alignment import _prune_low_support_elements
+
+        # both columns below 0.9 -> threshold relaxes to the max support
+        aligned = [["a", None], [None, None], ["a", None]]
+        out = _prune_low_support_elements(aligned, 0.9)
+        assert out == [["a"], [None], ["a"]]
