@@ -92,6 +92,10 @@ class EngineConfig(BaseModel):
 
     # KV cache
     kv_block_size: int = 16
+    # "bf16" (default, matches compute dtype) or "fp8_e4m3" (half the KV
+    # traffic/footprint; OCP e4m3 with static scale 1.0 — opt-in, slightly
+    # reduced KV precision)
+    kv_cache_dtype: str = "bf16"
     # Fraction of free HBM given to the KV cache after weights are resident
     kv_memory_fraction: float = 0.70
     max_kv_blocks: Optional[int] = None  # explicit cap (used on CPU/tests)

@@ -166,6 +166,8 @@ class LLMEngine:
         tp = self.ctx.world_size
         kv_heads_local = max(1, a.num_kv_heads // tp)
         elem = torch.tensor([], dtype=self.dtype).element_size()
+        if self.config.kv_cache_dtype == "fp8_e4m3":
+            elem = 1
         block_bytes = 2 * a.num_layers * kv_heads_local * self.config.kv_block_size * a.head_dim_ * elem
         if self.config.max_kv_blocks is not None:
             num_blocks = self.config.max_kv_blocks
@@ -176,9 +178,12 @@ class LLMEngine:
             # cap the decode block-table width implied by max_seq_len anyway
         else:
             num_blocks = 512
+        cache_dtype = self.dtype
+        if self.config.kv_cache_dtype == "fp8_e4m3":
+            cache_dtype = torch.float8_e4m3fn
         return PagedKVCache(
             a.num_layers, kv_heads_local, a.head_dim_, self.config.kv_block_size,
-            int(num_blocks), self.device, self.dtype,
+            int(num_blocks), self.device, cache_dtype,
         )
 
     # --- helpers --------------------------------------------------------------

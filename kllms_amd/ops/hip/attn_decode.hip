@@ -32,12 +32,12 @@
 // partials layout: [B, KVH, max_chunks, gqa, 130]: 128 o values + m + l
 #define PART_STRIDE 130
 
-template <int GQA>
+template <int GQA, bool FP8>
 __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
     float* __restrict__ partials,
     const bf16_t* __restrict__ q,        // [B, H, 128], row stride q_tstride
-    const bf16_t* __restrict__ k_cache,  // [NB, KVH, BS, 128]
-    const bf16_t* __restrict__ v_cache,
+    const char* __restrict__ k_cache,    // [NB, KVH, BS, 128], bf16 or fp8
+    const char* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [B, max_blocks]
     const int* __restrict__ context_lens,  // [B]
     float scale, int num_kv_heads,
@@ -93,13 +93,24 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
       for (int key0 = 0; key0 < nkeys; key0 += NTHREADS / KLANES) {
         const int key = key0 + kg;
         if (key < nkeys) {
-          const bf16x8_vec* kv8 = reinterpret_cast<const bf16x8_vec*>(k_cache + rowoff[key]) + kl * VECS;
           float af[ELEMS];
+          if constexpr (FP8) {
+            const u8x8_vec* kv8 = reinterpret_cast<const u8x8_vec*>(k_cache + rowoff[key]) + kl * VECS;
 #pragma unroll
-          for (int c = 0; c < VECS; ++c) {
-            bf16x8_vec kv = kv8[c];
+            for (int c = 0; c < VECS; ++c) {
+              u8x8_vec kv = kv8[c];
 #pragma unroll
-            for (int j = 0; j < 8; ++j) af[c * 8 + j] = bf16_to_f32(kv[j]);
+              for (int j = 0; j < 8; ++j) af[c * 8 + j] = fp8_to_f32(kv[j]);
+            }
+          } else {
+            const bf16x8_vec* kv8 =
+                reinterpret_cast<const bf16x8_vec*>(k_cache + 2 * rowoff[key]) + kl * VECS;
+#pragma unroll
+            for (int c = 0; c < VECS; ++c) {
+              bf16x8_vec kv = kv8[c];
+#pragma unroll
+              for (int j = 0; j < 8; ++j) af[c * 8 + j] = bf16_to_f32(kv[j]);
+            }
           }
 #pragma unroll
           for (int g = 0; g < GQA; ++g) {
@@ -161,13 +172,18 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
       for (; key + 4 <= kend; key += 4) {
         uint32_t pairs[4];
 #pragma unroll
-        for (int u = 0; u < 4; ++u)
-          pairs[u] = *reinterpret_cast<const uint32_t*>(
-              (const short*)(v_cache + rowoff[key + u]) + d0);
+        for (int u = 0; u < 4; ++u) {
+          if constexpr (FP8)
+            pairs[u] = *reinterpret_cast<const unsigned short*>(v_cache + rowoff[key + u] + d0);
+          else
+            pairs[u] = *reinterpret_cast<const uint32_t*>(v_cache + 2 * (rowoff[key + u] + d0));
+        }
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
-          const float v0 = bf16_to_f32((short)(pairs[u] & 0xffff));
-          const float v1 = bf16_to_f32((short)(pairs[u] >> 16));
+          const float v0 = FP8 ? fp8_to_f32((unsigned char)(pairs[u] & 0xff))
+                               : bf16_to_f32((short)(pairs[u] & 0xffff));
+          const float v1 = FP8 ? fp8_to_f32((unsigned char)((pairs[u] >> 8) & 0xff))
+                               : bf16_to_f32((short)(pairs[u] >> 16));
 #pragma unroll
           for (int g = 0; g < GQA; ++g) {
             const float p = s_lds[g * TKV + key + u];  // wave-uniform: LDS broadcast
@@ -177,10 +193,15 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
         }
       }
       for (; key < kend; ++key) {
-        const uint32_t pair = *reinterpret_cast<const uint32_t*>(
-            (const short*)(v_cache + rowoff[key]) + d0);
-        const float v0 = bf16_to_f32((short)(pair & 0xffff));
-        const float v1 = bf16_to_f32((short)(pair >> 16));
+        uint32_t pair;
+        if constexpr (FP8)
+          pair = *reinterpret_cast<const unsigned short*>(v_cache + rowoff[key] + d0);
+        else
+          pair = *reinterpret_cast<const uint32_t*>(v_cache + 2 * (rowoff[key] + d0));
+        const float v0 = FP8 ? fp8_to_f32((unsigned char)(pair & 0xff))
+                             : bf16_to_f32((short)(pair & 0xffff));
+        const float v1 = FP8 ? fp8_to_f32((unsigned char)((pair >> 8) & 0xff))
+                             : bf16_to_f32((short)(pair >> 16));
 #pragma unroll
         for (int g = 0; g < GQA; ++g) {
           const float p = s_lds[g * TKV + key];
@@ -221,25 +242,29 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
 
 // host-side dispatcher: picks the GQA template instantiation
 extern "C" void launch_attn_decode_partial(
-    float* partials, const bf16_t* q, const bf16_t* k_cache, const bf16_t* v_cache,
+    float* partials, const bf16_t* q, const void* k_cache, const void* v_cache,
     const int* block_tables, const int* context_lens, float scale,
     int num_q_heads, int num_kv_heads, int block_size, int max_blocks,
-    int max_chunks, int q_tstride, int chunk_keys, int B, hipStream_t stream) {
+    int max_chunks, int q_tstride, int chunk_keys, int B, int cache_fp8,
+    hipStream_t stream) {
   const int gqa = num_q_heads / num_kv_heads;
   const dim3 grid(B, num_kv_heads, max_chunks);
   const size_t lds = (gqa * 128 + gqa * TKV + 24 + 4 * gqa * 128) * sizeof(float) +
                      TKV * sizeof(int64_t);
-#define LAUNCH(G)                                                                  \
-  hipLaunchKernelGGL(attn_decode_partial_t<G>, grid, dim3(NTHREADS), lds, stream,  \
-                     partials, q, k_cache, v_cache, block_tables, context_lens,    \
-                     scale, num_kv_heads, block_size, max_blocks, max_chunks, q_tstride, chunk_keys)
+#define LAUNCH(G, F)                                                                  \
+  hipLaunchKernelGGL((attn_decode_partial_t<G, F>), grid, dim3(NTHREADS), lds, stream, \
+                     partials, q, (const char*)k_cache, (const char*)v_cache,          \
+                     block_tables, context_lens, scale, num_kv_heads, block_size,      \
+                     max_blocks, max_chunks, q_tstride, chunk_keys)
+#define DISPATCH(G) do { if (cache_fp8) LAUNCH(G, true); else LAUNCH(G, false); } while (0)
   switch (gqa) {
-    case 1: LAUNCH(1); break;
-    case 2: LAUNCH(2); break;
-    case 4: LAUNCH(4); break;
-    case 8: LAUNCH(8); break;
-    default: LAUNCH(8); break;  // guarded by the binding's gqa<=8 check
+    case 1: DISPATCH(1); break;
+    case 2: DISPATCH(2); break;
+    case 4: DISPATCH(4); break;
+    case 8: DISPATCH(8); break;
+    default: DISPATCH(8); break;  // guarded by the binding's gqa<=8 check
   }
+#undef DISPATCH
 #undef LAUNCH
 }
 

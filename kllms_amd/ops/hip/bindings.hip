@@ -12,7 +12,8 @@ extern "C" __global__ void fused_add_rmsnorm_kernel(bf16_t*, const bf16_t*, bf16
 extern "C" __global__ void silu_mul_kernel(bf16_t*, const bf16_t*, const bf16_t*, int64_t, int, int);
 extern "C" __global__ void rope_kernel(bf16_t*, bf16_t*, const int64_t*, const float*, int, int, int, int, int);
 extern "C" __global__ void store_kv_kernel(const bf16_t*, const bf16_t*, bf16_t*, bf16_t*, const int64_t*, int, int, int, int, int);
-extern "C" void launch_attn_decode_partial(float*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, float, int, int, int, int, int, int, int, int, hipStream_t);
+extern "C" __global__ void store_kv_fp8_kernel(const bf16_t*, const bf16_t*, unsigned char*, unsigned char*, const int64_t*, int, int, int, int, int);
+extern "C" void launch_attn_decode_partial(float*, const bf16_t*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int, int, int, hipStream_t);
 extern "C" __global__ void attn_decode_reduce_kernel(bf16_t*, const float*, const int*, int, int, int, int);
 extern "C" __global__ void attn_prefill_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, const int*, float, int, int, int, int);  // grouped: [G], [G], [G*4]
 extern "C" __global__ void sample_kernel(int64_t*, float*, const float*, const float*, const float*, const int*, const int64_t*, const int64_t*, const uint32_t*, int);
@@ -23,6 +24,11 @@ extern "C" __global__ void mfma_selftest_kernel(float*, const bf16_t*, const bf1
 
 // [T, H, D] activation views: head/dim dims packed, token stride free (so the
 // fused-QKV splits need no .contiguous() copies)
+#define CHECK_KV_CACHE(t) \
+  TORCH_CHECK((t).is_cuda() && (t).is_contiguous() && \
+              ((t).scalar_type() == at::kBFloat16 || (t).scalar_type() == at::kFloat8_e4m3fn), \
+              #t " must be a contiguous bf16 or fp8-e4m3 KV cache on GPU")
+
 #define CHECK_BF16_ROWS(t) \
   TORCH_CHECK((t).is_cuda() && (t).scalar_type() == at::kBFloat16 && (t).dim() == 3 && \
               (t).stride(2) == 1 && (t).stride(1) == (t).size(2), #t " must be a bf16 [T,H,D] row view")
@@ -77,21 +83,29 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions, tor
 
 void store_kv(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache, torch::Tensor v_cache,
               torch::Tensor slot_mapping) {
-  CHECK_BF16_ROWS(k); CHECK_BF16_ROWS(v); CHECK_BF16_CONTIG(k_cache); CHECK_BF16_CONTIG(v_cache);
+  CHECK_BF16_ROWS(k); CHECK_BF16_ROWS(v); CHECK_KV_CACHE(k_cache); CHECK_KV_CACHE(v_cache);
   TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share the token stride");
   const int T = k.size(0), KVH = k.size(1), D = k.size(2);
   const int BS = k_cache.size(2);
   const int64_t total = (int64_t)T * KVH * (D / 8);
   const int grid = (int)std::min<int64_t>((total + 255) / 256, 2048);
-  hipLaunchKernelGGL(store_kv_kernel, dim3(grid), dim3(256), 0, cur_stream(),
-                     cbf(k), cbf(v), bf(k_cache), bf(v_cache),
-                     slot_mapping.data_ptr<int64_t>(), T, KVH, D, BS, (int)k.stride(0));
+  if (k_cache.scalar_type() == at::kFloat8_e4m3fn) {
+    hipLaunchKernelGGL(store_kv_fp8_kernel, dim3(grid), dim3(256), 0, cur_stream(),
+                       cbf(k), cbf(v),
+                       reinterpret_cast<unsigned char*>(k_cache.data_ptr()),
+                       reinterpret_cast<unsigned char*>(v_cache.data_ptr()),
+                       slot_mapping.data_ptr<int64_t>(), T, KVH, D, BS, (int)k.stride(0));
+  } else {
+    hipLaunchKernelGGL(store_kv_kernel, dim3(grid), dim3(256), 0, cur_stream(),
+                       cbf(k), cbf(v), bf(k_cache), bf(v_cache),
+                       slot_mapping.data_ptr<int64_t>(), T, KVH, D, BS, (int)k.stride(0));
+  }
 }
 
 void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                        torch::Tensor v_cache, torch::Tensor block_tables,
                        torch::Tensor context_lens, double scale) {
-  CHECK_BF16_CONTIG(out); CHECK_BF16_ROWS(q); CHECK_BF16_CONTIG(k_cache); CHECK_BF16_CONTIG(v_cache);
+  CHECK_BF16_CONTIG(out); CHECK_BF16_ROWS(q); CHECK_KV_CACHE(k_cache); CHECK_KV_CACHE(v_cache);
   const int B = q.size(0), H = q.size(1), D = q.size(2);
   const int KVH = k_cache.size(1), BS = k_cache.size(2);
   TORCH_CHECK(D == 128, "attn_decode: head_dim must be 128");
@@ -108,10 +122,12 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache
   const int max_chunks = std::max(1, (max_ctx + CHUNK_KEYS - 1) / CHUNK_KEYS);
   auto partials = torch::empty({(int64_t)B * KVH * max_chunks * gqa * 130},
                                torch::dtype(torch::kFloat).device(q.device()));
+  const int cache_fp8 = k_cache.scalar_type() == at::kFloat8_e4m3fn ? 1 : 0;
   launch_attn_decode_partial(
-      partials.data_ptr<float>(), cbf(q), cbf(k_cache), cbf(v_cache),
+      partials.data_ptr<float>(), cbf(q), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr<int>(), context_lens.data_ptr<int>(),
-      (float)scale, H, KVH, BS, max_blocks, max_chunks, (int)q.stride(0), CHUNK_KEYS, B, cur_stream());
+      (float)scale, H, KVH, BS, max_blocks, max_chunks, (int)q.stride(0), CHUNK_KEYS, B,
+      cache_fp8, cur_stream());
   hipLaunchKernelGGL(attn_decode_reduce_kernel, dim3(B, H), dim3(64), 0, cur_stream(),
                      bf(out), partials.data_ptr<float>(), context_lens.data_ptr<int>(),
                      H, KVH, max_chunks, CHUNK_KEYS);

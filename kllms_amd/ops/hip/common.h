@@ -4,6 +4,7 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <hip/hip_fp8.h>
 #include <hip/hip_fp16.h>
 #include <stdint.h>
 
@@ -74,6 +75,22 @@ __device__ __forceinline__ float block_reduce_max(float v, float* lds_scratch) {
   v = wave_reduce_max(v);
   __syncthreads();
   return v;
+}
+
+// fp8 (OCP e4m3) conversions for the optional fp8 KV cache (gfx950 native
+// OCP format, NOT the MI300X fnuz variant)
+typedef unsigned char u8x16_vec __attribute__((ext_vector_type(16)));
+typedef unsigned char u8x8_vec __attribute__((ext_vector_type(8)));
+
+__device__ __forceinline__ float fp8_to_f32(unsigned char u) {
+  __hip_fp8_e4m3 v;
+  v.__x = u;
+  return (float)v;
+}
+
+__device__ __forceinline__ unsigned char f32_to_fp8(float f) {
+  __hip_fp8_e4m3 v(f);
+  return v.__x;
 }
 
 // splitmix64: per-(seed, step, token) counter-based RNG for the sampler

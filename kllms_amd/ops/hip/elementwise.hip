@@ -171,3 +171,37 @@ extern "C" __global__ void __launch_bounds__(256) store_kv_kernel(
     reinterpret_cast<bf16x8_vec*>(v_cache + dst_off)[dv] = *src_v;
   }
 }
+
+// fp8 (e4m3) variant of the paged-KV scatter: bf16 activations quantized
+// into 1-byte cache elements (static scale 1.0; e4m3 covers +-448 which the
+// post-RoPE K / V magnitudes fit comfortably).
+extern "C" __global__ void __launch_bounds__(256) store_kv_fp8_kernel(
+    const bf16_t* __restrict__ k, const bf16_t* __restrict__ v,
+    unsigned char* __restrict__ k_cache, unsigned char* __restrict__ v_cache,
+    const int64_t* __restrict__ slots, int num_tokens, int kv_heads,
+    int head_dim, int block_size, int kv_tstride) {
+  const int dvec = head_dim / 8;
+  const int64_t total = (int64_t)num_tokens * kv_heads * dvec;
+  for (int64_t idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int dv = idx % dvec;
+    const int g = (idx / dvec) % kv_heads;
+    const int t = idx / ((int64_t)dvec * kv_heads);
+    const int64_t slot = slots[t];
+    const int64_t blk = slot / block_size;
+    const int off = slot % block_size;
+    const bf16x8_vec src_k =
+        reinterpret_cast<const bf16x8_vec*>(k + (int64_t)t * kv_tstride + g * head_dim)[dv];
+    const bf16x8_vec src_v =
+        reinterpret_cast<const bf16x8_vec*>(v + (int64_t)t * kv_tstride + g * head_dim)[dv];
+    u8x8_vec qk, qv;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      qk[j] = f32_to_fp8(bf16_to_f32(src_k[j]));
+      qv[j] = f32_to_fp8(bf16_to_f32(src_v[j]));
+    }
+    const int64_t dst_off = ((blk * kv_heads + g) * block_size + off) * head_dim;
+    reinterpret_cast<u8x8_vec*>(k_cache + dst_off)[dv] = qk;
+    reinterpret_cast<u8x8_vec*>(v_cache + dst_off)[dv] = qv;
+  }
+}

@@ -350,3 +350,50 @@ class TestEngineGPU:
         scale = eager_logits.abs().max().item()
         eng.kv.free_sequence(seq)
         assert diff < 0.05 * max(scale, 1.0), f"graph vs eager logits: {diff} (scale {scale})"
+
+
+class TestFp8KVCacheGPU:
+    def test_store_and_decode_fp8(self):
+        torch.manual_seed(17)
+        KVH, D, BS, NB, B = 2, 128, 16, 64, 5
+        H = KVH * 4
+        ctx = [7, 33, 257, 120, 300]
+        kc = torch.zeros(NB, KVH, BS, D, dtype=torch.float8_e4m3fn, device=DEV)
+        vc = torch.zeros_like(kc)
+        # fill via the store kernel (quantization on device)
+        T = NB * BS
+        k = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=DEV) * 0.3
+        v = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=DEV) * 0.3
+        slots = torch.arange(T, device=DEV)
+        ops.store_kv(k, v, kc, vc, slots)
+        # quantized store matches the torch reference
+        ref_kc = torch.zeros_like(kc)
+        ref_vc = torch.zeros_like(vc)
+        torch_ref.store_kv(k, v, ref_kc, ref_vc, slots)
+        assert torch.equal(kc.view(torch.uint8), ref_kc.view(torch.uint8))
+
+        max_blocks = max((c + BS - 1) // BS for c in ctx)
+        perm = torch.randperm(NB).tolist()
+        bt = torch.zeros(B, max_blocks, dtype=torch.int32, device=DEV)
+        it = iter(perm * 4)
+        for b in range(B):
+            for j in range((ctx[b] + BS - 1) // BS):
+                bt[b, j] = next(it)
+        lens = torch.tensor(ctx, dtype=torch.int32, device=DEV)
+        q = torch.randn(B, H, D, dtype=torch.bfloat16, device=DEV) * 0.5
+        scale = D ** -0.5
+        out = ops.attn_decode_paged(q, kc, vc, bt, lens, scale)
+        ref = torch_ref.attn_decode_paged(q, kc, vc, bt, lens, scale)
+        assert_close_bf16(out, ref, rtol=4e-2, atol=4e-2, msg="fp8 decode")
+
+    def test_engine_fp8_cache_generates(self):
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.engine import GenRequest, LLMEngine
+        from kllms_amd.engine.sampling import SamplingParams
+
+        eng = LLMEngine(EngineConfig(model="mid-llama", max_kv_blocks=512, use_hip_graphs=False,
+                                     seed=3, kv_cache_dtype="fp8_e4m3"))
+        out = eng.generate([GenRequest(prompt_ids=list(range(1, 80)), n=2,
+                                       sampling=SamplingParams(temperature=0.0, max_tokens=12))])[0]
+        assert out.streams[0].token_ids == out.streams[1].token_ids
+        assert all(np.isfinite(lp) for lp in out.streams[0].logprobs)
