@@ -72,6 +72,27 @@ class DecodeGraphRunner:
         self.graphs[bs] = g
         self.buffers[bs] = buf
 
+    def _recover_generator_state(self) -> None:
+        """Un-stick the default Philox generator after a failed capture.
+
+        hipGraph capture registers the device's default RNG generator at
+        capture_begin; only a successful capture_end runs the epilogue that
+        clears the generator's 'capturing' flag. A capture that dies between
+        the two (e.g. cudaStreamEndCapture on an invalidated capture) leaves
+        the flag set, and the next RNG op ANYWHERE in the process then raises
+        "Offset increment outside graph capture encountered unexpectedly".
+        Swapping in a fresh clone of the state (same seed/offset, flag clear)
+        repairs it.
+        """
+        try:
+            idx = self.engine.device.index
+            if idx is None:
+                idx = torch.cuda.current_device()
+            gen = torch.cuda.default_generators[idx]
+            gen.graphsafe_set_state(gen.clone_state())
+        except Exception:
+            pass  # best effort: eager fallback works regardless for non-RNG ops
+
     def _bucket(self, b: int) -> Optional[int]:
         for bs in self.batch_sizes:
             if bs >= b:
@@ -92,9 +113,11 @@ class DecodeGraphRunner:
                 import logging
 
                 logging.getLogger("kllms_amd.engine").error(
-                    "hipGraph capture failed for batch %d (%s); falling back to eager", bs, e
+                    "hipGraph capture failed for batch %d (%s); falling back to eager",
+                    bs, e, exc_info=True,
                 )
                 self._enabled = False
+                self._recover_generator_state()
                 return self.engine.model.forward_decode(ids, batch)
         buf = self.buffers[bs]
         scratch_slot = self._scratch_block * self.engine.config.kv_block_size

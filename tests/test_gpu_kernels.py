@@ -351,6 +351,41 @@ class TestEngineGPU:
         eng.kv.free_sequence(seq)
         assert diff < 0.05 * max(scale, 1.0), f"graph vs eager logits: {diff} (scale {scale})"
 
+    def test_failed_capture_repairs_rng_state(self):
+        """A capture that dies mid-flight (illegal sync inside the captured
+        region) must fall back to eager AND un-stick the default Philox
+        generator — otherwise the next RNG op anywhere in the process raises
+        'Offset increment outside graph capture encountered unexpectedly'
+        (capture_begin runs the generator prologue; only a successful
+        capture_end runs the epilogue that clears the capturing flag)."""
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.engine import GenRequest, LLMEngine
+        from kllms_amd.engine.sampling import SamplingParams
+
+        eng = LLMEngine(EngineConfig(
+            model="tiny-llama", max_kv_blocks=128, use_hip_graphs=True,
+            hip_graph_batch_sizes=[1, 2, 4], max_seq_len=256, seed=11,
+        ))
+        orig = eng.model.forward_hidden
+
+        def poisoned(ids, batch):
+            torch.cuda.synchronize()   # prohibited during stream capture
+            _ = ids.cpu()              # sync D2H copy: also prohibited
+            return orig(ids, batch)
+
+        eng.model.forward_hidden = poisoned
+        try:
+            out = eng.generate([GenRequest(
+                prompt_ids=list(range(1, 20)), n=2,
+                sampling=SamplingParams(temperature=0.8, max_tokens=4, seed=3))])[0]
+        finally:
+            eng.model.forward_hidden = orig
+        assert eng._graph_runner is not None and not eng._graph_runner._enabled, \
+            "poisoned capture unexpectedly succeeded"
+        assert all(len(s.token_ids) > 0 for s in out.streams)  # eager fallback served
+        t = torch.randn(64, device=DEV)  # repaired generator must serve RNG again
+        assert torch.isfinite(t).all()
+
 
 class TestFp8KVCacheGPU:
     def test_store_and_decode_fp8(self):
