@@ -363,8 +363,8 @@ class TestEngineGPU:
         from kllms_amd.engine.sampling import SamplingParams
 
         eng = LLMEngine(EngineConfig(
-            model="tiny-llama", max_kv_blocks=128, use_hip_graphs=True,
-            hip_graph_batch_sizes=[1, 2, 4], max_seq_len=256, seed=11,
+            model="mid-llama", max_kv_blocks=256, use_hip_graphs=True,
+            hip_graph_batch_sizes=[1, 2, 4], max_seq_len=512, seed=11,
         ))
         orig = eng.model.forward_hidden
 
