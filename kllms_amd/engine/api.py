@@ -22,6 +22,7 @@ from ..types.openai_compat import (
     ChatCompletion,
     ChatCompletionMessage,
     ChatCompletionTokenLogprob,
+    TopLogprob,
     Choice,
     ChoiceLogprobs,
     CompletionUsage,
@@ -100,7 +101,12 @@ class LocalEngineClient:
             frequency_penalty=call_params.get("frequency_penalty", 0.0),
             presence_penalty=call_params.get("presence_penalty", 0.0),
             logprobs=bool(call_params.get("logprobs", False)),
+            top_logprobs=int(call_params.get("top_logprobs") or 0),
         )
+        if sampling.top_logprobs and not sampling.logprobs:
+            raise ValueError("top_logprobs requires logprobs=True")
+        if not 0 <= sampling.top_logprobs <= 20:
+            raise ValueError("top_logprobs must be between 0 and 20")
 
         constraint = None
         response_format = call_params.get("response_format")
@@ -144,10 +150,16 @@ class LocalEngineClient:
 
     def _mk_logprobs(self, stream) -> Optional[ChoiceLogprobs]:
         toks = []
-        for tid, lp in zip(stream.token_ids, stream.logprobs):
+        alt_lists = getattr(stream, "top_logprobs", []) or []
+        for pos, (tid, lp) in enumerate(zip(stream.token_ids, stream.logprobs)):
             s = self.engine.tokenizer.decode([tid])
+            tops = []
+            if pos < len(alt_lists):
+                for a_tid, a_lp in alt_lists[pos]:
+                    a_s = self.engine.tokenizer.decode([a_tid])
+                    tops.append(TopLogprob(token=a_s, bytes=list(a_s.encode()), logprob=a_lp))
             toks.append(
-                ChatCompletionTokenLogprob(token=s, bytes=list(s.encode()), logprob=lp, top_logprobs=[])
+                ChatCompletionTokenLogprob(token=s, bytes=list(s.encode()), logprob=lp, top_logprobs=tops)
             )
         return ChoiceLogprobs(content=toks)
 
@@ -197,6 +209,7 @@ class LocalEngineClient:
                 stop=call_params.get("stop"),
                 seed=call_params.get("seed"),
                 logprobs=bool(call_params.get("logprobs", False)),
+                top_logprobs=int(call_params.get("top_logprobs") or 0),
             )
             constraint = None
             rf = call_params.get("response_format")

@@ -39,6 +39,9 @@ class GenRequest:
 class StreamOutput:
     token_ids: List[int] = field(default_factory=list)
     logprobs: List[float] = field(default_factory=list)
+    # per emitted token: top-K (token_id, logprob) alternatives — only
+    # populated for streams whose sampling.top_logprobs > 0
+    top_logprobs: List[List[Tuple[int, float]]] = field(default_factory=list)
     text: str = ""
     finish_reason: str = "length"
     stream_idx: int = 0  # position among the request's n streams
@@ -414,6 +417,29 @@ class LLMEngine:
             return self._graph_runner.run(state.ids, batch)
         return self.model.forward_decode(state.ids, batch)
 
+    def _topk_logprobs(self, logits: torch.Tensor, mask: Optional[torch.Tensor], K: int):
+        """Top-K alternatives from the same distribution the sampler scores:
+        log-softmax of the MASKED, UNtempered logits (OpenAI `top_logprobs`
+        semantics, matching ops.sample's chosen-token logprob)."""
+        lf = logits.float()
+        if mask is not None:
+            V = lf.shape[1]
+            bit = torch.arange(V, device=lf.device)
+            allowed = (mask[:, torch.div(bit, 32, rounding_mode="floor")] >> (bit % 32).to(torch.int64)) & 1
+            lf = torch.where(allowed.bool(), lf, torch.full_like(lf, float("-inf")))
+        lp = torch.log_softmax(lf, dim=-1)
+        vals, idx = lp.topk(K, dim=-1)
+        return idx, vals
+
+    def _append_topk(self, streams: List["_Stream"], K: int,
+                     ids_flat: List[float], lps_flat: List[float]) -> None:
+        for i, s in enumerate(streams):
+            ks = s.sampling.top_logprobs
+            if ks > 0:
+                s.out.top_logprobs.append(
+                    [(int(ids_flat[i * K + j]), lps_flat[i * K + j]) for j in range(min(ks, K))]
+                )
+
     def _sample_state(self, state: "_DecodeBatchState", logits: torch.Tensor) -> bool:
         streams = state.streams
         logits = self._apply_penalties(logits, streams)
@@ -421,15 +447,23 @@ class LLMEngine:
         tokens, logprobs = ops.sample(
             logits, state.temps, state.top_ps, state.top_ks, state.seeds, state.steps, mask
         )
+        K = max((s.sampling.top_logprobs for s in streams), default=0)
+        parts = [tokens.to(torch.float64), logprobs.to(torch.float64)]
+        if K > 0:
+            tl_idx, tl_vals = self._topk_logprobs(logits, mask, K)
+            parts += [tl_idx.reshape(-1).to(torch.float64), tl_vals.reshape(-1).to(torch.float64)]
         state.ids = tokens
         state.steps += 1
         state.positions += 1
         state.ctx += 1
-        # ONE device->host transfer per step (tokens + logprobs packed)
-        packed = torch.cat([tokens.to(torch.float64), logprobs.to(torch.float64)]).cpu()
+        # ONE device->host transfer per step (tokens + logprobs [+ topk] packed)
+        packed = torch.cat(parts).cpu()
         B = tokens.shape[0]
         tokens_l = [int(x) for x in packed[:B].tolist()]
-        logprobs_l = packed[B:].tolist()
+        logprobs_l = packed[B : 2 * B].tolist()
+        if K > 0:
+            flat = packed[2 * B :].tolist()
+            self._append_topk(streams, K, flat[: B * K], flat[B * K :])
         any_done = False
         for i, s in enumerate(streams):
             tok = tokens_l[i]
@@ -450,6 +484,10 @@ class LLMEngine:
         temps, top_ps, top_ks, seeds, steps = self._sampling_tensors(streams)
         mask = self._constraint_mask(streams)
         tokens, logprobs = ops.sample(logits, temps, top_ps, top_ks, seeds, steps, mask)
+        K = max((s.sampling.top_logprobs for s in streams), default=0)
+        if K > 0:
+            tl_idx, tl_vals = self._topk_logprobs(logits, mask, K)
+            self._append_topk(streams, K, tl_idx.reshape(-1).tolist(), tl_vals.reshape(-1).tolist())
         tokens_l = tokens.tolist()
         logprobs_l = logprobs.tolist()
         for i, s in enumerate(streams):
@@ -467,6 +505,8 @@ class LLMEngine:
         if s.last_token == self.eos_token_id:
             s.out.token_ids.pop()  # EOS itself is not part of the content
             s.out.logprobs.pop()
+            if len(s.out.top_logprobs) > len(s.out.token_ids):
+                s.out.top_logprobs.pop()
             s.out.finish_reason = "stop"
             s.done = True
             return

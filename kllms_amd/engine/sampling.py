@@ -19,6 +19,9 @@ class SamplingParams(BaseModel):
     frequency_penalty: float = 0.0
     presence_penalty: float = 0.0
     logprobs: bool = False
+    # number of top alternative tokens to report per position (OpenAI
+    # `top_logprobs`, 0 = off; requires logprobs=True at the API layer)
+    top_logprobs: int = 0
 
     @property
     def stop_list(self) -> List[str]:

@@ -58,6 +58,31 @@ class TestCreate:
         for t in orig.logprobs.content:
             assert t.logprob <= 0.0
 
+    def test_top_logprobs(self, client):
+        r = client.chat.completions.create(
+            messages=[{"role": "user", "content": "x"}], model="tiny-llama", n=2,
+            max_tokens=4, logprobs=True, top_logprobs=3, seed=1, temperature=0.0,
+        )
+        for orig in r.choices[1:]:
+            assert orig.logprobs is not None and orig.logprobs.content
+            for t in orig.logprobs.content:
+                assert len(t.top_logprobs) == 3
+                lps = [a.logprob for a in t.top_logprobs]
+                assert lps == sorted(lps, reverse=True)
+                # greedy decode: the chosen token IS the top alternative
+                assert t.top_logprobs[0].token == t.token
+                assert t.top_logprobs[0].logprob == pytest.approx(t.logprob, abs=1e-4)
+        # consensus choice carries the FIRST original's logprobs (reference
+        # consolidation.py:129 copies completion.choices[0].logprobs)
+        assert r.choices[0].logprobs == r.choices[1].logprobs
+
+    def test_top_logprobs_requires_logprobs(self, client):
+        with pytest.raises(ValueError):
+            client.chat.completions.create(
+                messages=[{"role": "user", "content": "x"}], model="tiny-llama",
+                n=1, max_tokens=2, top_logprobs=3,
+            )
+
 
 class Extraction(BaseModel):
     label: str
