@@ -142,7 +142,10 @@ class LocalEngineClient:
         if schema is None:
             return None
         try:
-            return JsonSchemaConstraint(schema, self.engine.tokenizer)
+            return JsonSchemaConstraint(
+                schema, self.engine.tokenizer,
+                whitespace=getattr(self.config, "constrained_whitespace", False),
+            )
         except Exception:
             if constrained:
                 raise

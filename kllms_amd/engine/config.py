@@ -105,6 +105,9 @@ class EngineConfig(BaseModel):
 
     # Decode-step hipGraph capture
     use_hip_graphs: bool = True
+    # constrained decoding: allow optional JSON whitespace between tokens
+    # (default emits compact JSON — smaller DFA, fewer wasted tokens)
+    constrained_whitespace: bool = False
     hip_graph_batch_sizes: list[int] = Field(default_factory=lambda: [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256])
 
     # Generation defaults

@@ -65,6 +65,12 @@ class Opt(_Node):
 
 _DIGITS = set(b"0123456789")
 _HEX = set(b"0123456789abcdefABCDEF")
+_WS_SET = set(b" \t\n\r")
+
+
+def _lead_ws(node: _Node, ws: bool) -> _Node:
+    """Optional JSON whitespace before a token (whitespace-tolerant mode)."""
+    return Seq([Star(Cls(_WS_SET)), node]) if ws else node
 # JSON string body chars: printable ASCII except '"' and '\\'. Non-ASCII
 # content is expressible via \uXXXX escapes, which keeps every constrained
 # output valid UTF-8 regardless of how the tokenizer splits bytes.
@@ -116,7 +122,14 @@ class SchemaCompileError(ValueError):
     pass
 
 
-def schema_to_ir(schema: Dict[str, Any], defs: Dict[str, Any], depth: int = 0) -> _Node:
+def schema_to_ir(schema: Dict[str, Any], defs: Dict[str, Any], depth: int = 0, ws: bool = False) -> _Node:
+    """Schema -> IR. With ws=True every JSON token (value start, key, comma,
+    colon, closer) admits optional leading whitespace; ws inside atoms
+    (strings, numbers) stays forbidden, as in the JSON grammar."""
+    return _lead_ws(_schema_ir_body(schema, defs, depth, ws), ws)
+
+
+def _schema_ir_body(schema: Dict[str, Any], defs: Dict[str, Any], depth: int, ws: bool) -> _Node:
     if depth > 32:
         raise SchemaCompileError("schema nesting too deep (recursive $ref?)")
     if "$ref" in schema:
@@ -124,18 +137,18 @@ def schema_to_ir(schema: Dict[str, Any], defs: Dict[str, Any], depth: int = 0) -
         name = ref.split("/")[-1]
         if name not in defs:
             raise SchemaCompileError(f"unresolved $ref {ref}")
-        return schema_to_ir(defs[name], defs, depth + 1)
+        return _schema_ir_body(defs[name], defs, depth + 1, ws)
     if "enum" in schema:
         return Alt([Lit(json.dumps(v).encode()) for v in schema["enum"]])
     if "const" in schema:
         return Lit(json.dumps(schema["const"]).encode())
     if "anyOf" in schema or "oneOf" in schema:
         opts = schema.get("anyOf") or schema.get("oneOf")
-        return Alt([schema_to_ir(s, defs, depth + 1) for s in opts])
+        return Alt([_schema_ir_body(s, defs, depth + 1, ws) for s in opts])
 
     t = schema.get("type")
     if isinstance(t, list):
-        return Alt([schema_to_ir({**schema, "type": ti}, defs, depth + 1) for ti in t])
+        return Alt([_schema_ir_body({**schema, "type": ti}, defs, depth + 1, ws) for ti in t])
     if t == "string":
         return _json_string_ir(schema.get("minLength", 0), schema.get("maxLength"))
     if t == "integer":
@@ -150,31 +163,32 @@ def schema_to_ir(schema: Dict[str, Any], defs: Dict[str, Any], depth: int = 0) -
         item = schema.get("items", {})
 
         def item_ir() -> _Node:
-            return schema_to_ir(item, defs, depth + 1) if item else _any_value_ir(defs, depth + 1)
+            return schema_to_ir(item, defs, depth + 1, ws) if item else _any_value_ir(defs, depth + 1, ws=ws)
 
         min_items = schema.get("minItems", 0)
         max_items = schema.get("maxItems")
+        comma = lambda: _lead_ws(Lit(b","), ws)
         if min_items == 0:
-            more = _bounded(lambda: Seq([Lit(b","), item_ir()]), 0,
+            more = _bounded(lambda: Seq([comma(), item_ir()]), 0,
                             None if max_items is None else max(0, max_items - 1))
-            return Seq([Lit(b"["), Opt(Seq([item_ir(), more])), Lit(b"]")])
-        head = [item_ir()] + [Seq([Lit(b","), item_ir()]) for _ in range(min_items - 1)]
-        more = _bounded(lambda: Seq([Lit(b","), item_ir()]), 0,
+            return Seq([Lit(b"["), Opt(Seq([item_ir(), more])), _lead_ws(Lit(b"]"), ws)])
+        head = [item_ir()] + [Seq([comma(), item_ir()]) for _ in range(min_items - 1)]
+        more = _bounded(lambda: Seq([comma(), item_ir()]), 0,
                         None if max_items is None else max(0, max_items - min_items))
-        return Seq([Lit(b"[")] + head + [more, Lit(b"]")])
+        return Seq([Lit(b"[")] + head + [more, _lead_ws(Lit(b"]"), ws)])
     if t == "object" or "properties" in schema:
         props = schema.get("properties", {})
         if not props:
-            return _any_object_ir(defs, depth + 1)
+            return _any_object_ir(defs, depth + 1, ws)
         parts: List[_Node] = [Lit(b"{")]
         required = set(schema.get("required", list(props.keys())))
         first = True
         for key, sub in props.items():
             field = Seq([
-                Lit(b"" if first else b","),
-                Lit(json.dumps(key).encode()),
-                Lit(b":"),
-                schema_to_ir(sub, defs, depth + 1),
+                Lit(b"") if first else _lead_ws(Lit(b","), ws),
+                _lead_ws(Lit(json.dumps(key).encode()), ws),
+                _lead_ws(Lit(b":"), ws),
+                schema_to_ir(sub, defs, depth + 1, ws),
             ])
             if key in required or first:
                 # fields emitted in schema order; the first field is always
@@ -183,34 +197,41 @@ def schema_to_ir(schema: Dict[str, Any], defs: Dict[str, Any], depth: int = 0) -
             else:
                 parts.append(Opt(field))
             first = False
-        parts.append(Lit(b"}"))
+        parts.append(_lead_ws(Lit(b"}"), ws))
         return Seq(parts)
     # untyped: any JSON value (bounded nesting)
-    return _any_value_ir(defs, depth + 1)
+    return _any_value_ir(defs, depth + 1, ws=ws)
 
 
-def _any_value_ir(defs, depth: int, max_depth: int = 4) -> _Node:
-    """'Any JSON value' with nesting bounded at max_depth levels."""
+def _any_value_ir(defs, depth: int, max_depth: int = 3, ws: bool = False) -> _Node:
+    """'Any JSON value' with nesting bounded at max_depth levels.
+
+    max_depth=3 keeps the free-form DFA ~7k states (measured; depth 4
+    exceeds the 20k compile cap in both compact and ws modes). Typed
+    schemas never hit this path and stay small."""
     scalar = Alt([_json_string_ir(), _number_ir(), Alt([Lit(b"true"), Lit(b"false")]), Lit(b"null")])
     node = scalar
     for _ in range(max_depth):
-        arr = Seq([Lit(b"["), Opt(Seq([node, Star(Seq([Lit(b","), node]))])), Lit(b"]")])
+        val = _lead_ws(node, ws)
+        key = _lead_ws(_json_string_ir(), ws)
+        colon = _lead_ws(Lit(b":"), ws)
+        arr = Seq([Lit(b"["), Opt(Seq([val, Star(Seq([_lead_ws(Lit(b","), ws), val]))])), _lead_ws(Lit(b"]"), ws)])
         obj = Seq([
             Lit(b"{"),
             Opt(Seq([
-                _json_string_ir(), Lit(b":"), node,
-                Star(Seq([Lit(b","), _json_string_ir(), Lit(b":"), node])),
+                key, colon, val,
+                Star(Seq([_lead_ws(Lit(b","), ws), key, colon, val])),
             ])),
-            Lit(b"}"),
+            _lead_ws(Lit(b"}"), ws),
         ])
         node = Alt([scalar, arr, obj])
     return node
 
 
-def _any_object_ir(defs, depth: int) -> _Node:
-    v = _any_value_ir(defs, depth)
-    member = Seq([_json_string_ir(), Lit(b":"), v])
-    return Seq([Lit(b"{"), Opt(Seq([member, Star(Seq([Lit(b","), member]))])), Lit(b"}")])
+def _any_object_ir(defs, depth: int, ws: bool = False) -> _Node:
+    v = _any_value_ir(defs, depth, ws=ws)
+    member = Seq([_lead_ws(_json_string_ir(), ws), _lead_ws(Lit(b":"), ws), _lead_ws(v, ws)])
+    return Seq([Lit(b"{"), Opt(Seq([member, Star(Seq([_lead_ws(Lit(b","), ws), member]))])), _lead_ws(Lit(b"}"), ws)])
 
 
 # ---------------------------------------------------------------------------
@@ -341,16 +362,16 @@ class JsonSchemaConstraint:
     """Per-request constraint handle. State is an int DFA state; tables are
     cached per (schema, tokenizer vocab)."""
 
-    def __init__(self, schema: Dict[str, Any], tokenizer):
+    def __init__(self, schema: Dict[str, Any], tokenizer, whitespace: bool = False):
         self.schema = schema
-        key = (json.dumps(schema, sort_keys=True), id(type(tokenizer)), tokenizer.vocab_size)
+        key = (json.dumps(schema, sort_keys=True), id(type(tokenizer)), tokenizer.vocab_size, whitespace)
         cached = _TABLE_CACHE.get(key)  # type: ignore[arg-type]
         if cached is not None:
             self.__dict__.update(cached.__dict__)
             return
 
         defs = schema.get("$defs", schema.get("definitions", {}))
-        ir = schema_to_ir(schema, defs)
+        ir = schema_to_ir(schema, defs, ws=whitespace)
         trans, accepting, start = compile_dfa(ir)
         S = trans.shape[0]
         V = tokenizer.vocab_size

@@ -16,8 +16,8 @@ from kllms_amd.engine.constrained import (
 from kllms_amd.engine.tokenizer import ByteTokenizer
 
 
-def accepts(schema: dict, text: str) -> bool:
-    ir = schema_to_ir(schema, schema.get("$defs", {}))
+def accepts(schema: dict, text: str, ws: bool = False) -> bool:
+    ir = schema_to_ir(schema, schema.get("$defs", {}), ws=ws)
     trans, accepting, start = compile_dfa(ir)
     s = start
     for b in text.encode():
@@ -196,3 +196,74 @@ class TestBounds:
         assert accepts(sch, '{"tag":"abcd","nums":[1,2]}')
         assert not accepts(sch, '{"tag":"abcde","nums":[1,2]}')
         assert not accepts(sch, '{"tag":"ab","nums":[1,2,3]}')
+
+
+class TestWhitespaceTolerant:
+    """ws=True: optional JSON whitespace between tokens (never inside atoms)."""
+
+    SCH = {
+        "type": "object",
+        "properties": {
+            "name": {"type": "string"},
+            "age": {"type": "integer"},
+            "tags": {"type": "array", "items": {"type": "string"}},
+        },
+        "required": ["name", "age", "tags"],
+    }
+
+    def test_pretty_printed_accepted(self):
+        obj = {"name": "Ann", "age": 30, "tags": ["x", "y"]}
+        for indent in (None, 1, 2, 4):
+            assert accepts(self.SCH, json.dumps(obj, indent=indent), ws=True)
+
+    def test_mixed_whitespace_accepted(self):
+        assert accepts(self.SCH, '{ "name" : "A" ,\n\t"age":1 , "tags" : [ ] }', ws=True)
+        assert accepts(self.SCH, '  {"name":"A","age":1,"tags":["z"]}', ws=True)
+
+    def test_ws_inside_atoms_rejected(self):
+        assert not accepts({"type": "integer"}, "1 2", ws=True)
+        assert not accepts({"type": "number"}, "1. 5", ws=True)
+        assert not accepts({"type": "boolean"}, "tr ue", ws=True)
+
+    def test_compact_mode_rejects_pretty(self):
+        obj = {"name": "Ann", "age": 30, "tags": []}
+        assert not accepts(self.SCH, json.dumps(obj, indent=2), ws=False)
+        assert accepts(self.SCH, json.dumps(obj, separators=(",", ":")), ws=False)
+
+    def test_enum_array_anyof_ws(self):
+        sch = {
+            "type": "array",
+            "items": {"anyOf": [{"enum": ["a", "b"]}, {"type": "integer"}]},
+            "minItems": 1,
+            "maxItems": 3,
+        }
+        assert accepts(sch, '[ "a" , 2 ]', ws=True)
+        assert accepts(sch, '["a",2]', ws=True)
+        assert not accepts(sch, '[ "a" , 2 , 3 , 4 ]', ws=True)
+
+    def test_untyped_any_value_ws(self):
+        sch = {"type": "object"}  # free-form object
+        assert accepts(sch, '{ "k" : [ 1 , { "n" : null } ] }', ws=True)
+
+    def test_end_to_end_ws_constrained_generation(self):
+        """Engine with constrained_whitespace=True still produces parseable,
+        schema-valid JSON on every completed stream."""
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.api import LocalEngineClient
+
+        class Rec(BaseModel):
+            label: str
+            score: int
+
+        client = LocalEngineClient(
+            model="tiny-llama", max_kv_blocks=512, use_hip_graphs=False,
+            device="cpu", constrained_whitespace=True,
+        )
+        r = client.chat_completions_parse(
+            messages=[{"role": "user", "content": "extract"}],
+            model="tiny-llama", response_format=Rec, n=2, max_tokens=200, seed=4,
+        )
+        for c in r.choices:
+            if c.finish_reason == "stop" and c.message.content:
+                obj = json.loads(c.message.content)
+                Rec.model_validate(obj)
