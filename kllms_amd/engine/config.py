@@ -108,6 +108,10 @@ class EngineConfig(BaseModel):
     # constrained decoding: allow optional JSON whitespace between tokens
     # (default emits compact JSON — smaller DFA, fewer wasted tokens)
     constrained_whitespace: bool = False
+    # scheduler: prefill prompts longer than this in slices interleaved with
+    # decode steps, so a long prompt doesn't stall running streams
+    # (None = whole-prompt prefill at admission)
+    prefill_chunk_tokens: Optional[int] = None
     hip_graph_batch_sizes: list[int] = Field(default_factory=lambda: [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256])
 
     # Generation defaults

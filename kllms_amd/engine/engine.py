@@ -303,8 +303,21 @@ class LLMEngine:
         input_ids = torch.tensor(all_ids, dtype=torch.long, device=dev)
         prefill_logits = self.model.forward_prefill(input_ids, batch)  # [n_req, V]
 
-        # ---- fork: n streams per request share the prompt KV blocks ---------
+        # ---- fork + first-token sampling ------------------------------------
         outputs = [RequestOutput(prompt_tokens=len(r.prompt_ids)) for r in requests]
+        self._fork_and_sample(requests, parent_seqs, prefill_logits, streams)
+        return outputs
+
+    def _fork_and_sample(
+        self,
+        requests: List[GenRequest],
+        parent_seqs: List[SequenceKV],
+        prefill_logits: torch.Tensor,
+        streams: List[_Stream],
+    ) -> List[_Stream]:
+        """Fork each request's prompt KV into n streams (copy-on-write block
+        refs) and sample the first token of every stream from that request's
+        prefill logits row."""
         new_streams: List[_Stream] = []
         for ri, req in enumerate(requests):
             base_seed = req.sampling.seed if req.sampling.seed is not None else (self.config.seed * 1000003 + ri)
@@ -317,13 +330,49 @@ class LLMEngine:
                 new_streams.append(st)
             self.kv.free_sequence(parent_seqs[ri])  # streams hold their own refs
 
-        # ---- first token: sample n times from each request's prefill logits -
         rep_logits = torch.cat([
             prefill_logits[ri].unsqueeze(0).expand(max(1, requests[ri].n), -1)
             for ri in range(len(requests))
         ])
         self._sample_and_append(rep_logits.contiguous(), new_streams)
-        return outputs
+        return new_streams
+
+    def prefill_chunk(
+        self, seq: SequenceKV, prompt_ids: List[int], start: int, end: int, want_logits: bool = False
+    ) -> Optional[torch.Tensor]:
+        """Prefill positions [start, end) of a prompt into an already-allocated
+        sequence (chunked prefill: the scheduler slices long prompts between
+        decode steps so running streams aren't stalled for a whole prefill).
+
+        Chunks after the first must attend to KV written by earlier chunks,
+        which lives in the paged cache — so this runs the DECODE forward path
+        with one row per chunk token (row i: position start+i, context
+        start+i+1). Identical math to the packed prefill (RoPE positions and
+        causal structure preserved); the LM head runs only on the last token
+        of the final chunk (want_logits=True)."""
+        dev = self.device
+        ids = torch.tensor(prompt_ids[start:end], dtype=torch.long, device=dev)
+        pos = torch.arange(start, end, dtype=torch.long, device=dev)
+        slots_all = self.kv.prefill_slot_mapping(seq)
+        slots = torch.tensor(slots_all[start:end], dtype=torch.long, device=dev)
+        nb = len(seq.blocks)
+        bt = (
+            torch.tensor(seq.blocks, dtype=torch.int32, device=dev)
+            .unsqueeze(0).expand(end - start, nb).contiguous()
+        )
+        ctx = torch.arange(start + 1, end + 1, dtype=torch.int32, device=dev)
+        batch = ForwardBatch(
+            mode="decode",
+            positions=pos,
+            slot_mapping=slots,
+            kv_caches=self.kv.layer_caches(),
+            block_tables=bt,
+            context_lens=ctx,
+        )
+        hidden = self.model.forward_hidden(ids, batch)
+        if want_logits:
+            return self.model.compute_logits(hidden[-1:])
+        return None
 
     def finish_stream(self, s: _Stream) -> StreamOutput:
         """Phase 3: detokenize (unless a stop-string already trimmed the

@@ -40,6 +40,7 @@ class _WorkerContext:
         self.active: List[_Stream] = []
         self.stream_ticket: Dict[int, "_Ticket"] = {}   # id(stream) -> ticket
         self.state: Optional[_DecodeBatchState] = None
+        self.pending: List["_PendingPrefill"] = []       # chunked prefills in flight
 
 
 @dataclass
@@ -48,6 +49,15 @@ class _Ticket:
     future: Future
     output: Optional[RequestOutput] = None
     remaining: int = 0
+
+
+@dataclass
+class _PendingPrefill:
+    """A long prompt being prefilled in slices (config.prefill_chunk_tokens):
+    one chunk advances per scheduler loop iteration, between decode steps."""
+    ticket: _Ticket
+    seq: object          # SequenceKV holding the whole prompt's blocks
+    next_pos: int = 0
 
 
 class BatchScheduler:
@@ -108,14 +118,72 @@ class BatchScheduler:
         ctx = _WorkerContext()
         with torch.inference_mode():
             while not self._stop.is_set():
-                tickets = self._drain(len(ctx.active), block=not ctx.active)
+                idle = not ctx.active and not ctx.pending
+                # pending chunked prefills reserve their future stream slots
+                reserved = sum(max(1, p.ticket.request.n) for p in ctx.pending)
+                tickets = self._drain(len(ctx.active) + reserved, block=idle)
                 with self.engine_lock:
                     if tickets:
                         self._admit(ctx, tickets)
+                    if ctx.pending:
+                        self._advance_prefill(ctx)
                     if ctx.active:
                         self._step(ctx)
 
     def _admit(self, ctx: "_WorkerContext", tickets: List[_Ticket]) -> None:
+        chunk = self.engine.config.prefill_chunk_tokens
+        if chunk:
+            long_tickets = [t for t in tickets if len(t.request.prompt_ids) > chunk]
+            tickets = [t for t in tickets if len(t.request.prompt_ids) <= chunk]
+            for t in long_tickets:
+                self._start_chunked(ctx, t)
+        if tickets:
+            self._admit_whole(ctx, tickets)
+
+    def _start_chunked(self, ctx: "_WorkerContext", t: _Ticket) -> None:
+        eng = self.engine
+        try:
+            seq = eng.kv.alloc_sequence(len(t.request.prompt_ids))
+        except Exception as e:
+            t.future.set_exception(e)
+            return
+        t.output = RequestOutput(prompt_tokens=len(t.request.prompt_ids))
+        t.remaining = max(1, t.request.n)
+        ctx.pending.append(_PendingPrefill(ticket=t, seq=seq))
+        self.admitted_batches += 1
+
+    def _advance_prefill(self, ctx: "_WorkerContext") -> None:
+        """Run ONE chunk of the head-of-line pending prefill; on the final
+        chunk, fork the n streams and sample their first token."""
+        eng = self.engine
+        p = ctx.pending[0]
+        t = p.ticket
+        ids = t.request.prompt_ids
+        chunk = eng.config.prefill_chunk_tokens or len(ids)
+        end = min(p.next_pos + chunk, len(ids))
+        final = end == len(ids)
+        try:
+            logits = eng.prefill_chunk(p.seq, ids, p.next_pos, end, want_logits=final)
+            p.next_pos = end
+            if not final:
+                return
+            ctx.pending.pop(0)
+            new_streams: List[_Stream] = []
+            eng._fork_and_sample([t.request], [p.seq], logits, new_streams)
+        except Exception as e:
+            ctx.pending.pop(0)
+            if p.seq.blocks:
+                eng.kv.free_sequence(p.seq)
+            if not t.future.done():
+                t.future.set_exception(e)
+            return
+        for st in new_streams:
+            ctx.stream_ticket[id(st)] = t
+        self._retire(eng, new_streams, ctx.stream_ticket)
+        ctx.active.extend(st for st in new_streams if not st.done)
+        ctx.state = None
+
+    def _admit_whole(self, ctx: "_WorkerContext", tickets: List[_Ticket]) -> None:
         eng = self.engine
         reqs = [t.request for t in tickets]
         parent_seqs: List = []
