@@ -289,6 +289,44 @@ class TestEngineGPU:
         eng.kv.free_sequence(seq)
         assert int(logits[0].argmax()) == toks[-1], "decode diverged from prefill at long context"
 
+    def test_chunked_prefill_matches_packed_gpu(self):
+        """Chunked prefill (decode-path slices over the paged cache) must
+        reproduce the packed varlen prefill's next-token logits on the HIP
+        kernel path (per-row context lengths in attn_decode, store_kv)."""
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.engine import LLMEngine
+        from kllms_amd.models.llama import ForwardBatch
+
+        eng = LLMEngine(EngineConfig(
+            model="mid-llama", max_kv_blocks=512, use_hip_graphs=False,
+            max_seq_len=1024, seed=3,
+        ))
+        ids = [(i * 17) % 200 + 1 for i in range(101)]  # not a chunk multiple
+
+        seq_a = eng.kv.alloc_sequence(len(ids))
+        batch = ForwardBatch(
+            mode="prefill",
+            positions=torch.arange(len(ids), device=DEV),
+            slot_mapping=torch.tensor(eng.kv.prefill_slot_mapping(seq_a), device=DEV),
+            kv_caches=eng.kv.layer_caches(),
+            cu_seqlens=torch.tensor([0, len(ids)], dtype=torch.int32, device=DEV),
+        )
+        logits_a = eng.model.forward_prefill(torch.tensor(ids, device=DEV), batch)
+
+        seq_b = eng.kv.alloc_sequence(len(ids))
+        logits_b = None
+        for a in range(0, len(ids), 32):
+            e = min(a + 32, len(ids))
+            r = eng.prefill_chunk(seq_b, ids, a, e, want_logits=(e == len(ids)))
+            if r is not None:
+                logits_b = r
+        la, lb = logits_a[0].float(), logits_b[0].float()
+        scale = la.abs().max().item()
+        diff = (la - lb).abs().max().item()
+        eng.kv.free_sequence(seq_a)
+        eng.kv.free_sequence(seq_b)
+        assert diff < 0.05 * max(scale, 1.0), f"chunked vs packed logits: {diff} (scale {scale})"
+
     def test_mixtral_generate_gpu(self):
         from kllms_amd.engine.config import EngineConfig
         from kllms_amd.engine.engine import GenRequest, LLMEngine
