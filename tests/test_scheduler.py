@@ -182,3 +182,29 @@ class TestChunkedPrefill:
         sched.submit(greedy_req([7] * 21, 6, n=3)).result(timeout=120)
         sched.shutdown()
         assert eng.kv.allocator.num_free == free0
+
+    def test_async_client_over_chunked_scheduler(self):
+        """AsyncKLLMs concurrent calls through a chunked-prefill scheduler:
+        long and short prompts complete with full consensus responses."""
+        import asyncio
+        from kllms_amd import AsyncKLLMs
+
+        client = AsyncKLLMs(
+            model="tiny-llama", max_kv_blocks=512, use_hip_graphs=False,
+            device="cpu", seed=0, max_batch_size=64, prefill_chunk_tokens=8,
+        )
+
+        async def run():
+            long_msg = "word " * 60
+            return await asyncio.gather(
+                client.chat.completions.create(
+                    messages=[{"role": "user", "content": long_msg}],
+                    model="tiny-llama", n=3, max_tokens=6, seed=1),
+                client.chat.completions.create(
+                    messages=[{"role": "user", "content": "hi"}],
+                    model="tiny-llama", n=2, max_tokens=6, seed=2),
+            )
+
+        r_long, r_short = asyncio.run(run())
+        assert len(r_long.choices) == 4 and len(r_short.choices) == 3
+        assert r_long.usage.prompt_tokens > 8  # actually went through chunking
