@@ -12,6 +12,12 @@ decode steps.
 
 Prints one JSON-ish dict: short-request p50/p95/max latency + long-request
 latency + aggregate req/s.
+
+Scale note: on MI355X an 8B packed prefill runs ~17 us/token (profiles/), so
+a 2.8k-token prompt stalls decode for only ~50 ms — chunking pays at much
+longer prompts (16k+) or bigger models, and costs a little extra arithmetic
+per slice otherwise. On CPU (torch oracle path) the per-row decode forward is
+slow, so CPU runs of this script exaggerate the chunked cost.
 """
 
 import argparse
