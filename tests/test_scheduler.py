@@ -208,3 +208,35 @@ class TestChunkedPrefill:
         r_long, r_short = asyncio.run(run())
         assert len(r_long.choices) == 4 and len(r_short.choices) == 3
         assert r_long.usage.prompt_tokens > 8  # actually went through chunking
+
+    def test_mixtral_chunked_prefill(self):
+        """Chunked prefill through the MoE decode forward path (router +
+        experts per row) matches the packed MoE prefill end to end."""
+        eng = LLMEngine(EngineConfig(
+            model="tiny-mixtral", max_kv_blocks=512, use_hip_graphs=False,
+            device="cpu", seed=0, max_batch_size=64, prefill_chunk_tokens=8,
+        ))
+        prompt = [(i * 11) % 120 + 1 for i in range(30)]
+        direct = eng.generate([greedy_req(prompt, 8, n=2)])[0]
+        sched = BatchScheduler(eng)
+        out = sched.submit(greedy_req(prompt, 8, n=2)).result(timeout=120)
+        sched.shutdown()
+        assert [s.token_ids for s in out.streams] == [s.token_ids for s in direct.streams]
+
+    def test_kv_exhaustion_fails_chunked_request_cleanly(self):
+        """A chunked prefill that cannot allocate KV must fail its future
+        (not hang) and leave the allocator balanced for later requests."""
+        eng = LLMEngine(EngineConfig(
+            model="tiny-llama", max_kv_blocks=8, use_hip_graphs=False,
+            device="cpu", seed=0, prefill_chunk_tokens=4, max_seq_len=4096,
+        ))
+        free0 = eng.kv.allocator.num_free
+        sched = BatchScheduler(eng)
+        f = sched.submit(greedy_req([5] * 400, 4))  # needs 25 blocks, only 8 exist
+        with pytest.raises(Exception):
+            f.result(timeout=60)
+        # engine still serves a small request afterwards
+        ok = sched.submit(greedy_req([1, 2, 3], 4)).result(timeout=60)
+        sched.shutdown()
+        assert len(ok.streams) == 1
+        assert eng.kv.allocator.num_free == free0
