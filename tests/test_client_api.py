@@ -134,3 +134,88 @@ class TestAsync:
 
         r = asyncio.run(run())
         assert len(r.choices) == 3
+
+
+class TestReferenceScenarios:
+    """Mirrors the reference's example/comprehensive test themes
+    (README_TESTS.md: nested models, multi-turn, error handling,
+    temperature effects)."""
+
+    def test_deep_nested_parse(self, client):
+        class Employee(BaseModel):
+            name: str
+            role: str
+
+        class Department(BaseModel):
+            dept: str
+            staff: list[Employee]
+
+        class Company(BaseModel):
+            company: str
+            departments: list[Department]
+
+        r = client.chat.completions.parse(
+            messages=[{"role": "user", "content": "extract the org chart"}],
+            model="tiny-llama", response_format=Company, n=3, max_tokens=400, seed=7,
+        )
+        assert len(r.choices) == 4
+        for c in r.choices[1:]:
+            if c.finish_reason == "stop":
+                Company.model_validate(json.loads(c.message.content))
+        # consensus likelihoods mirror the nested structure when parseable
+        if r.choices[0].message.parsed is not None:
+            assert isinstance(r.choices[0].message.parsed, Company)
+            assert "company" in r.likelihoods
+
+    def test_multi_turn_conversation(self, client):
+        r = client.chat.completions.create(
+            messages=[
+                {"role": "system", "content": "You are terse."},
+                {"role": "user", "content": "What is 2+2?"},
+                {"role": "assistant", "content": "4"},
+                {"role": "user", "content": "And doubled?"},
+            ],
+            model="tiny-llama", n=3, max_tokens=6, seed=3,
+        )
+        assert len(r.choices) == 4
+        # multi-turn prompt is longer than the last message alone
+        solo = client.chat.completions.create(
+            messages=[{"role": "user", "content": "And doubled?"}],
+            model="tiny-llama", n=1, max_tokens=2,
+        )
+        assert r.usage.prompt_tokens > solo.usage.prompt_tokens
+
+    def test_empty_messages_raises(self, client):
+        with pytest.raises(ValueError):
+            client.chat.completions.create(messages=[], model="tiny-llama", n=1)
+
+    def test_unknown_model_raises(self):
+        from kllms_amd import KLLMs
+        with pytest.raises(Exception):
+            KLLMs(model="no-such-model-xyz").chat.completions.create(
+                messages=[{"role": "user", "content": "x"}], model="no-such-model-xyz",
+            )
+
+    def test_greedy_consensus_is_unanimous(self, client):
+        """temperature=0: all n streams identical -> consensus == the unique
+        completion and every leaf likelihood is 1.0."""
+        r = client.chat.completions.create(
+            messages=[{"role": "user", "content": "deterministic"}],
+            model="tiny-llama", n=4, temperature=0.0, max_tokens=6, seed=9,
+        )
+        texts = {c.message.content for c in r.choices[1:]}
+        assert len(texts) == 1
+        assert r.choices[0].message.content == texts.pop()
+
+        def leaves(x):
+            if isinstance(x, dict):
+                for v in x.values():
+                    yield from leaves(v)
+            elif isinstance(x, list):
+                for v in x:
+                    yield from leaves(v)
+            elif isinstance(x, (int, float)):
+                yield x
+
+        lv = list(leaves(r.likelihoods))
+        assert lv and all(v == pytest.approx(1.0) for v in lv)
