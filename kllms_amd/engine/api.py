@@ -220,6 +220,8 @@ class LocalEngineClient:
                 constraint = self._build_constraint(rf, constrained=False)
             prompt = eng.tokenizer.apply_chat_template(messages)
             prompt_ids = eng.tokenizer.encode(prompt)
+            max_prompt = eng.config.max_seq_len - (sampling.max_tokens or self.config.default_max_new_tokens)
+            prompt_ids = prompt_ids[: max(1, max_prompt)]
             reqs.append(GenRequest(prompt_ids=prompt_ids, n=n, sampling=sampling, constraint=constraint))
             samplings.append(sampling)
         with self._engine_lock:

@@ -122,8 +122,21 @@ class LLMEngine:
             self.device = torch.device(config.device)
         else:
             self.device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
-        self.ctx = parallel_ctx or ParallelContext.from_env_or_single()
-        assert self.ctx.world_size == config.tp_size or parallel_ctx is None or True
+        # Parallel context is tied to config.tp_size, NOT to whether
+        # torch.distributed happens to be initialized: under a data-parallel
+        # launch (torchrun, one independent engine per rank — the bench's
+        # weak-scaling mode) a tp_size=1 engine must NOT adopt the world as a
+        # TP group, or its layers would all-reduce across ranks that are
+        # processing different batches.
+        if parallel_ctx is not None:
+            self.ctx = parallel_ctx
+        elif config.tp_size > 1:
+            self.ctx = ParallelContext.from_env_or_single()
+            assert self.ctx.world_size == config.tp_size, (
+                f"tp_size={config.tp_size} but torch.distributed world is {self.ctx.world_size}"
+            )
+        else:
+            self.ctx = ParallelContext()
 
         self.dtype = {"bfloat16": torch.bfloat16, "float16": torch.float16, "float32": torch.float32}[config.dtype]
         # CPU bf16 matmuls are slow and torch CPU attention paths prefer f32

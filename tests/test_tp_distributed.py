@@ -101,3 +101,51 @@ def test_tp2_matches_tp1_logits():
 
     out = eng.generate([GenRequest(prompt_ids=ids, n=2, sampling=SamplingParams(temperature=0.0, max_tokens=8))])[0]
     assert out.streams[0].token_ids == results[0][1], "TP=2 greedy diverged from TP=1"
+
+
+def _dp_worker(rank: int, world_size: int, q):
+    """tp_size=1 engines under an INITIALIZED process group must be fully
+    independent (the bench's data-parallel weak-scaling mode): different
+    per-rank prompts, no collectives. Before the ctx/tp_size fix, the engine
+    adopted the world as a TP group and gloo crashed on mismatched sizes."""
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(PORT + 3)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.engine import GenRequest, LLMEngine
+        from kllms_amd.engine.sampling import SamplingParams
+
+        eng = LLMEngine(EngineConfig(model="tiny-llama", max_kv_blocks=128,
+                                     use_hip_graphs=False, device="cpu", seed=rank))
+        assert eng.ctx.world_size == 1  # NOT the dist world
+        # per-rank different work: different prompt lengths and counts
+        prompts = [[1 + rank + i] * (5 + 3 * rank + i) for i in range(2 + rank)]
+        outs = eng.generate([
+            GenRequest(prompt_ids=p, n=1 + rank,
+                       sampling=SamplingParams(temperature=0.8, max_tokens=6, seed=rank))
+            for p in prompts
+        ])
+        q.put((rank, [len(o.streams) for o in outs]))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp_engines_independent_under_dist():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_dp_worker, args=(r, 2, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, counts = q.get(timeout=240)
+        results[rank] = counts
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert results[0] == [1, 1]
+    assert results[1] == [2, 2, 2]
