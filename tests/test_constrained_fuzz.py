@@ -1,0 +1,123 @@
+"""Property-based tests for the constrained-decoding FSM.
+
+For ANY schema the generator produces and ANY instance conforming to it,
+the compiled byte DFA must accept the canonical compact serialization —
+and in whitespace mode, any re-indented serialization. This is the
+soundness direction (valid JSON is never rejected); the completeness
+direction (the FSM never lets an invalid byte through) is covered by the
+deterministic tests in test_constrained.py.
+"""
+
+import json
+
+import pytest
+
+hypothesis = pytest.importorskip("hypothesis")
+from hypothesis import HealthCheck, given, settings, strategies as st  # noqa: E402
+
+from kllms_amd.engine.constrained import compile_dfa, schema_to_ir  # noqa: E402
+
+
+# ---------------------------------------------------------------------------
+# (schema, conforming instance) strategy
+# ---------------------------------------------------------------------------
+
+# printable-ASCII strings excluding chars the FSM expresses via escapes
+_safe_text = st.text(
+    alphabet=st.characters(min_codepoint=0x20, max_codepoint=0x7E, exclude_characters='"\\'),
+    max_size=8,
+)
+
+_keys = st.sampled_from(["a", "b", "c", "name", "value", "k1", "k2"])
+
+
+def _leaf():
+    return st.one_of(
+        st.tuples(st.just({"type": "boolean"}), st.booleans()),
+        st.tuples(st.just({"type": "null"}), st.none()),
+        st.tuples(st.just({"type": "integer"}), st.integers(min_value=-10**9, max_value=10**9)),
+        st.tuples(st.just({"type": "string"}), _safe_text),
+        _safe_text.flatmap(
+            lambda v: st.just(({"enum": [v, v + "x"]}, v))
+        ),
+        st.tuples(
+            st.just({"type": "number"}),
+            st.floats(allow_nan=False, allow_infinity=False, width=32),
+        ),
+    )
+
+
+def _extend(children):
+    def arr(pair_strategy):
+        return pair_strategy.flatmap(
+            lambda pair: st.lists(st.just(pair[1]), min_size=0, max_size=3).map(
+                lambda items: ({"type": "array", "items": pair[0]}, items)
+            )
+        )
+
+    def obj(pair_strategy):
+        return st.lists(
+            st.tuples(_keys, pair_strategy), min_size=1, max_size=3,
+            unique_by=lambda kv: kv[0],
+        ).map(
+            lambda kvs: (
+                {
+                    "type": "object",
+                    "properties": {k: s for k, (s, _) in kvs},
+                    "required": [k for k, _ in kvs],
+                },
+                {k: v for k, (_, v) in kvs},
+            )
+        )
+
+    return st.one_of(arr(children), obj(children))
+
+
+schema_and_instance = st.recursive(_leaf(), _extend, max_leaves=6)
+
+
+def _accepts(schema: dict, text: str, ws: bool) -> bool:
+    ir = schema_to_ir(schema, schema.get("$defs", {}), ws=ws)
+    trans, accepting, s = compile_dfa(ir)
+    for b in text.encode():
+        nxt = int(trans[s, b])
+        if nxt == 0xFFFF:
+            return False
+        s = nxt
+    return bool(accepting[s])
+
+
+def _dumps_compact(value) -> str:
+    return json.dumps(value, separators=(",", ":"))
+
+
+@settings(max_examples=120, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(schema_and_instance)
+def test_compact_serialization_accepted(pair):
+    schema, value = pair
+    text = _dumps_compact(value)
+    # the FSM's number grammar is JSON's; python float repr is JSON-compatible
+    assert _accepts(schema, text, ws=False), f"{schema} rejected {text!r}"
+
+
+@settings(max_examples=60, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(schema_and_instance, st.sampled_from([1, 2, 4]))
+def test_ws_mode_accepts_pretty(pair, indent):
+    schema, value = pair
+    text = json.dumps(value, indent=indent)
+    assert _accepts(schema, text, ws=True), f"{schema} (ws) rejected {text!r}"
+
+
+@settings(max_examples=60, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(schema_and_instance)
+def test_mutated_byte_never_reaches_accept_silently(pair):
+    """Flipping a structural byte must either be rejected mid-walk or leave
+    the DFA in a non-accepting state (never 'accepted garbage')."""
+    schema, value = pair
+    text = _dumps_compact(value)
+    if len(text) < 2:
+        return
+    # corrupt one byte to an impossible structural character
+    mid = len(text) // 2
+    corrupted = text[:mid] + "\x01" + text[mid + 1:]
+    assert not _accepts(schema, corrupted, ws=False)
