@@ -92,3 +92,29 @@ def test_consolidation_invariants_total(contents):
     assert isinstance(r.choices[0].message.content, str)
     if r.likelihoods is not None:
         check_likelihood_tree(r.likelihoods)
+
+
+@settings(max_examples=120, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(st.lists(json_values, min_size=1, max_size=5))
+def test_key_aligner_total_and_shape_preserving(values):
+    """The key-based aligner (L1b) must be total over arbitrary JSON and
+    return one aligned value per input plus a path-mapping table whose
+    per-path lists have one entry per source."""
+    from kllms_amd.consensus.key_based_alignment import recursive_align
+
+    aligned, km = recursive_align(values, "levenshtein", no_embed, None, 0.5)
+    assert len(aligned) == len(values)
+    for path, sources in km.items():
+        assert len(sources) == len(values), (path, sources)
+
+
+@settings(max_examples=100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(st.lists(json_values, min_size=2, max_size=4))
+def test_key_aligner_consensus_pipeline(values):
+    """consensus over key-aligned values never raises and produces a
+    range-valid likelihood tree (the 'key' aligner end-to-end)."""
+    from kllms_amd.consensus.key_based_alignment import recursive_align
+
+    aligned, _ = recursive_align(values, "levenshtein", no_embed, None, 0.5)
+    consensus, conf = consensus_values(list(aligned), SETTINGS, no_embed)
+    check_likelihood_tree(conf)
