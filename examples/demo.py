@@ -88,6 +88,27 @@ def main():
     except Exception as e:
         print("  raised:", type(e).__name__)
 
+    print("== 8. per-token top_logprobs ==")
+    r = client.chat.completions.create(
+        model=args.model, messages=[{"role": "user", "content": "alternatives"}],
+        n=1, max_tokens=3, temperature=0.0, logprobs=True, top_logprobs=3, seed=2,
+    )
+    tk = r.choices[0].logprobs.content[0]
+    print(f"  token {tk.token!r} lp={tk.logprob:.3f}; "
+          f"top-3: {[(t.token, round(t.logprob, 3)) for t in tk.top_logprobs]}")
+
+    print("== 9. serving opt-ins (fp8 KV cache, whitespace-tolerant JSON, chunked prefill) ==")
+    opted = KLLMs(model=args.model, device=client.client.config.device,
+                  use_hip_graphs=False, max_kv_blocks=256,
+                  kv_cache_dtype="fp8_e4m3", constrained_whitespace=True,
+                  prefill_chunk_tokens=64)
+    r = opted.chat.completions.parse(
+        model=args.model, messages=[{"role": "user", "content": "extract"}],
+        response_format=Invoice, n=2, max_tokens=200, seed=5,
+    )
+    print("  fp8-cache constrained parse choices:", len(r.choices),
+          "parsed:", r.choices[0].message.parsed is not None)
+
 
 if __name__ == "__main__":
     main()
