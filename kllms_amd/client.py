@@ -56,8 +56,27 @@ class BaseEngineWrapper:
 
         engine = self._extra_kwargs.pop("engine", None)
         if engine is not None and isinstance(engine, LocalEngineClient):
+            self._owns_engine = False
             return engine
+        self._owns_engine = True
         return LocalEngineClient(**self._extra_kwargs)
+
+    def close(self) -> None:
+        """Stop the continuous-batching scheduler thread, if one was started
+        (OpenAI clients expose close(); engines shared via ``engine=`` are
+        left running for their owner)."""
+        client = getattr(self, "_client", None)
+        if client is not None and getattr(self, "_owns_engine", False):
+            sched = getattr(client, "_scheduler", None)
+            if sched is not None:
+                sched.shutdown()
+                client._scheduler = None
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc) -> None:
+        self.close()
 
 
 class KLLMs(BaseEngineWrapper):
@@ -94,6 +113,15 @@ class AsyncKLLMs(BaseEngineWrapper):
     @property
     def client(self):
         return self._client
+
+    async def aclose(self) -> None:
+        self.close()
+
+    async def __aenter__(self):
+        return self
+
+    async def __aexit__(self, *exc) -> None:
+        self.close()
 
 
 class Chat:

@@ -219,3 +219,34 @@ class TestReferenceScenarios:
 
         lv = list(leaves(r.likelihoods))
         assert lv and all(v == pytest.approx(1.0) for v in lv)
+
+    def test_close_and_context_manager(self):
+        from kllms_amd import KLLMs
+
+        with KLLMs(**TINY) as c:
+            r = c.chat.completions.create(
+                messages=[{"role": "user", "content": "x"}], model="tiny-llama", n=1, max_tokens=2)
+            assert r.choices
+            # force a scheduler into existence, then ensure close() stops it
+            sched = c.client.scheduler
+            assert sched is not None
+        assert c.client._scheduler is None  # shut down on exit
+
+    def test_async_context_manager(self):
+        async def run():
+            async with AsyncKLLMs(**TINY) as ac:
+                r = await ac.chat.completions.create(
+                    messages=[{"role": "user", "content": "x"}], model="tiny-llama", n=2, max_tokens=2)
+                return r
+        r = asyncio.run(run())
+        assert len(r.choices) == 3
+
+    def test_shared_engine_not_closed_by_borrower(self):
+        from kllms_amd import KLLMs
+
+        owner = KLLMs(**TINY)
+        _ = owner.client.scheduler
+        with KLLMs(engine=owner.client) as borrower:
+            pass
+        assert owner.client._scheduler is not None  # borrower must not kill it
+        owner.close()
