@@ -149,3 +149,87 @@ def test_dp_engines_independent_under_dist():
         assert p.exitcode == 0
     assert results[0] == [1, 1]
     assert results[1] == [2, 2, 2]
+
+
+def _tp4_worker(rank: int, world_size: int, q):
+    """TP=4 (per-rank kv_heads=1, heads=2): each rank compares its TP-sharded
+    prefill logits and greedy decode against a locally-built TP=1 engine —
+    valid because random init is TP-degree-invariant by construction."""
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(PORT + 7)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        from kllms_amd.engine import config as cfgmod
+        from kllms_amd.engine.config import EngineConfig, ModelArchConfig
+        from kllms_amd.engine.engine import GenRequest, LLMEngine
+        from kllms_amd.engine.sampling import SamplingParams
+        from kllms_amd.models.llama import ForwardBatch
+        from kllms_amd.parallel.tp import ParallelContext
+
+        cfgmod.MODEL_PRESETS["test-tp4"] = ModelArchConfig(
+            arch="llama", vocab_size=512, hidden_size=256, intermediate_size=512,
+            num_layers=2, num_heads=8, num_kv_heads=4, rope_theta=10000.0,
+            max_position_embeddings=512,
+        )
+
+        def build(tp, ctx):
+            return LLMEngine(
+                EngineConfig(model="test-tp4", tp_size=tp, max_kv_blocks=128,
+                             use_hip_graphs=False, device="cpu", seed=0),
+                parallel_ctx=ctx,
+            )
+
+        eng4 = build(world_size, ParallelContext(world_size=world_size, rank=rank))
+        eng1 = build(1, None)
+
+        def prefill_logits(eng, ids):
+            seq = eng.kv.alloc_sequence(len(ids))
+            batch = ForwardBatch(
+                mode="prefill",
+                positions=torch.arange(len(ids)),
+                slot_mapping=torch.tensor(eng.kv.prefill_slot_mapping(seq)),
+                kv_caches=eng.kv.layer_caches(),
+                cu_seqlens=torch.tensor([0, len(ids)], dtype=torch.int32),
+            )
+            out = eng.model.forward_prefill(torch.tensor(ids), batch)
+            eng.kv.free_sequence(seq)
+            return out
+
+        ids = list(range(1, 41))
+        l4 = prefill_logits(eng4, ids)
+        l1 = prefill_logits(eng1, ids)
+        logits_close = torch.allclose(l4.float(), l1.float(), rtol=2e-2, atol=2e-2)
+
+        g4 = eng4.generate([GenRequest(prompt_ids=ids, n=2,
+                                       sampling=SamplingParams(temperature=0.0, max_tokens=8))])[0]
+        g1 = eng1.generate([GenRequest(prompt_ids=ids, n=2,
+                                       sampling=SamplingParams(temperature=0.0, max_tokens=8))])[0]
+        tokens_equal = [s.token_ids for s in g4.streams] == [s.token_ids for s in g1.streams]
+        q.put((rank, bool(logits_close), bool(tokens_equal),
+               [s.token_ids for s in g4.streams]))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp4_matches_tp1():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tp4_worker, args=(r, 4, q)) for r in range(4)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(4):
+        rank, lc, te, toks = q.get(timeout=240)
+        results[rank] = (lc, te, toks)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    for rank, (lc, te, toks) in results.items():
+        assert lc, f"rank {rank}: TP4 prefill logits diverged from TP1"
+        assert te, f"rank {rank}: TP4 greedy decode diverged from TP1"
+    # all ranks produced the identical decode (same model, same batch)
+    tok_sets = {json.dumps(v[2]) for v in results.values()}
+    assert len(tok_sets) == 1
