@@ -193,8 +193,12 @@ class LlamaForCausalLM(nn.Module):
             elif isinstance(mod, RowParallelLinear):
                 shard_dim = 1
                 full_shape[1] *= tp
-            elif leaf in spec:
-                shard_dim = spec[leaf]
+            n_parts = 1
+            if leaf in spec:
+                entry = spec[leaf]
+                # spec value: dim, or (dim, n_parts) for FUSED dims (e.g. the
+                # MoE [gate;up] stack) where each equal part shards separately
+                shard_dim, n_parts = entry if isinstance(entry, tuple) else (entry, 1)
                 full_shape[shard_dim] *= tp
             if "layernorm" in name or name.endswith("norm.weight"):
                 full = torch.ones(full_shape, dtype=torch.float32, device=p.device)
@@ -203,7 +207,19 @@ class LlamaForCausalLM(nn.Module):
                 full = torch.randn(full_shape, generator=g, dtype=torch.float32, device=p.device) * std
             if shard_dim is not None and tp > 1:
                 size = p.shape[shard_dim]
-                full = full.narrow(shard_dim, rank * size, size)
+                if n_parts > 1:
+                    # local layout is the concat of this rank's shard of EACH
+                    # part — a contiguous narrow would give rank 0 all of the
+                    # first part (matches weights.py's per-part sharding)
+                    part_full = full.shape[shard_dim] // n_parts
+                    part_loc = size // n_parts
+                    full = torch.cat(
+                        [full.narrow(shard_dim, j * part_full + rank * part_loc, part_loc)
+                         for j in range(n_parts)],
+                        dim=shard_dim,
+                    )
+                else:
+                    full = full.narrow(shard_dim, rank * size, size)
             p.copy_(full.to(p.dtype))
 
     def _owner_module(self, param_name: str):

@@ -27,8 +27,10 @@ from .llama import LlamaDecoderLayer, LlamaForCausalLM
 
 
 class MixtralMoE(nn.Module):
-    # TP sharding of the stacked expert tensors (dim scaled by tp at full size)
-    shard_spec = {"w_gate_up": 1, "w_down": 2}
+    # TP sharding of the stacked expert tensors (dim scaled by tp at full
+    # size); w_gate_up's dim 1 is the FUSED [gate; up] stack — 2 parts, each
+    # sharded separately (matches weights.py's per-part expert sharding)
+    shard_spec = {"w_gate_up": (1, 2), "w_down": 2}
 
     def __init__(self, cfg: ModelArchConfig, ctx: ParallelContext, dtype):
         super().__init__()

@@ -233,3 +233,57 @@ def test_tp4_matches_tp1():
     # all ranks produced the identical decode (same model, same batch)
     tok_sets = {json.dumps(v[2]) for v in results.values()}
     assert len(tok_sets) == 1
+
+
+def _mixtral_tp_worker(rank: int, world_size: int, q):
+    """TP=2 Mixtral over gloo: experts TP-sharded (each expert's gate/up and
+    down split across ranks, one all-reduce per MoE layer) must match TP=1."""
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(PORT + 11)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.engine import GenRequest, LLMEngine
+        from kllms_amd.engine.sampling import SamplingParams
+        from kllms_amd.parallel.tp import ParallelContext
+
+        def build(tp, ctx):
+            return LLMEngine(
+                EngineConfig(model="mid-mixtral", tp_size=tp, max_kv_blocks=128,
+                             use_hip_graphs=False, device="cpu", seed=0),
+                parallel_ctx=ctx,
+            )
+
+        eng2 = build(world_size, ParallelContext(world_size=world_size, rank=rank))
+        eng1 = build(1, None)
+        ids = list(range(1, 37))
+        req = lambda: GenRequest(prompt_ids=ids, n=2,
+                                 sampling=SamplingParams(temperature=0.0, max_tokens=8))
+        g2 = eng2.generate([req()])[0]
+        g1 = eng1.generate([req()])[0]
+        q.put((rank,
+               [s.token_ids for s in g2.streams] == [s.token_ids for s in g1.streams],
+               [s.token_ids for s in g2.streams]))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_mixtral_tp2_matches_tp1():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_mixtral_tp_worker, args=(r, 2, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, equal, toks = q.get(timeout=240)
+        results[rank] = (equal, toks)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    for rank, (equal, toks) in results.items():
+        assert equal, f"rank {rank}: Mixtral TP2 greedy decode diverged from TP1"
+    assert results[0][1] == results[1][1]
