@@ -240,3 +240,48 @@ class TestChunkedPrefill:
         sched.shutdown()
         assert len(ok.streams) == 1
         assert eng.kv.allocator.num_free == free0
+
+
+class TestInteractionSoak:
+    def test_mixed_workload_soak(self):
+        """Concurrent soak across feature interactions: chunked prefill x
+        constrained decoding x fp8 KV x varying n — everything completes,
+        every constrained 'stop' stream parses, allocator balances."""
+        import json as _json
+        from concurrent.futures import wait
+        from pydantic import BaseModel
+
+        from kllms_amd.engine.constrained import JsonSchemaConstraint
+
+        class Rec(BaseModel):
+            tag: str
+            num: int
+
+        eng = LLMEngine(EngineConfig(
+            model="tiny-llama", max_kv_blocks=1024, use_hip_graphs=False,
+            device="cpu", seed=0, max_batch_size=32, prefill_chunk_tokens=8,
+            kv_cache_dtype="fp8_e4m3",
+        ))
+        free0 = eng.kv.allocator.num_free
+        constraint = JsonSchemaConstraint(Rec.model_json_schema(), eng.tokenizer)
+        sched = BatchScheduler(eng, admit_wait_s=0.02)
+        futs = []
+        for i in range(14):
+            long = i % 3 == 0
+            prompt = [(i * 7 + j) % 150 + 1 for j in range(30 if long else 5)]
+            futs.append(sched.submit(GenRequest(
+                prompt_ids=prompt, n=1 + i % 3,
+                sampling=SamplingParams(temperature=0.9, max_tokens=60, seed=i),
+                constraint=constraint if i % 2 == 0 else None,
+            )))
+        done, not_done = wait(futs, timeout=180)
+        assert not not_done
+        for i, f in enumerate(futs):
+            out = f.result()
+            assert len(out.streams) == 1 + i % 3
+            if i % 2 == 0:
+                for s in out.streams:
+                    if s.finish_reason == "stop":
+                        Rec.model_validate(_json.loads(s.text))
+        sched.shutdown()
+        assert eng.kv.allocator.num_free == free0
