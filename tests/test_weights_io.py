@@ -152,3 +152,82 @@ def test_safetensors_tp2_shards(tmp_path):
         got_down = model.layers[0].mlp.down_proj.weight
         half = cfg.intermediate_size // 2
         assert torch.equal(got_down, down_full[:, rank * half:(rank + 1) * half])
+
+
+def _make_hf_mixtral_checkpoint(tmp_path, cfg, seed=0):
+    from safetensors.torch import save_file
+
+    g = torch.Generator().manual_seed(seed)
+    H, D, I = cfg.hidden_size, cfg.head_dim_, cfg.intermediate_size
+    tensors = {
+        "model.embed_tokens.weight": torch.randn(cfg.vocab_size, H, generator=g),
+        "model.norm.weight": torch.randn(H, generator=g),
+        "lm_head.weight": torch.randn(cfg.vocab_size, H, generator=g),
+    }
+    for L in range(cfg.num_layers):
+        p = f"model.layers.{L}"
+        tensors[f"{p}.self_attn.q_proj.weight"] = torch.randn(cfg.num_heads * D, H, generator=g)
+        tensors[f"{p}.self_attn.k_proj.weight"] = torch.randn(cfg.num_kv_heads * D, H, generator=g)
+        tensors[f"{p}.self_attn.v_proj.weight"] = torch.randn(cfg.num_kv_heads * D, H, generator=g)
+        tensors[f"{p}.self_attn.o_proj.weight"] = torch.randn(H, cfg.num_heads * D, generator=g)
+        tensors[f"{p}.block_sparse_moe.gate.weight"] = torch.randn(cfg.num_experts, H, generator=g)
+        for e in range(cfg.num_experts):
+            q = f"{p}.block_sparse_moe.experts.{e}"
+            tensors[f"{q}.w1.weight"] = torch.randn(I, H, generator=g)   # gate
+            tensors[f"{q}.w2.weight"] = torch.randn(H, I, generator=g)   # down
+            tensors[f"{q}.w3.weight"] = torch.randn(I, H, generator=g)   # up
+        tensors[f"{p}.input_layernorm.weight"] = torch.randn(H, generator=g)
+        tensors[f"{p}.post_attention_layernorm.weight"] = torch.randn(H, generator=g)
+    save_file({k: v.contiguous() for k, v in tensors.items()}, str(tmp_path / "model.safetensors"))
+    return tensors
+
+
+def test_mixtral_expert_tp2_shards(tmp_path):
+    """TP=2 Mixtral: each rank's stacked w_gate_up holds [gate_shard; up_shard]
+    per expert (per-part sharding, matching the TP-invariant random init)."""
+    pytest.importorskip("safetensors")
+    from kllms_amd.models.mixtral import MixtralForCausalLM
+
+    cfg = MODEL_PRESETS["mid-mixtral"]
+    tensors = _make_hf_mixtral_checkpoint(tmp_path, cfg, seed=3)
+    I = cfg.intermediate_size
+    half = I // 2
+    for rank in (0, 1):
+        ctx = ParallelContext(world_size=2, rank=rank)
+        model = MixtralForCausalLM(cfg, ctx, dtype=torch.float32)
+        load_safetensors_weights(model, str(tmp_path), ctx)
+        moe = model.layers[0].mlp
+        for e in range(cfg.num_experts):
+            gate_full = tensors[f"model.layers.0.block_sparse_moe.experts.{e}.w1.weight"]
+            up_full = tensors[f"model.layers.0.block_sparse_moe.experts.{e}.w3.weight"]
+            down_full = tensors[f"model.layers.0.block_sparse_moe.experts.{e}.w2.weight"]
+            want_gu = torch.cat([
+                gate_full[rank * half:(rank + 1) * half],
+                up_full[rank * half:(rank + 1) * half],
+            ], dim=0)
+            assert torch.equal(moe.w_gate_up[e], want_gu), f"rank {rank} expert {e} gate_up"
+            assert torch.equal(moe.w_down[e], down_full[:, rank * half:(rank + 1) * half])
+
+
+def test_mixtral_random_init_matches_loader_layout(tmp_path):
+    """random_init_ under TP must produce the SAME local layout convention as
+    the loader: reassembling both ranks' w_gate_up recovers a tensor whose
+    gate half equals ranks' first parts (regression for the fused-stack
+    contiguous-narrow bug)."""
+    from kllms_amd.models.mixtral import MixtralForCausalLM
+
+    cfg = MODEL_PRESETS["mid-mixtral"]
+    m1 = MixtralForCausalLM(cfg, ParallelContext(), dtype=torch.float32)
+    m1.random_init_(seed=0)
+    full = m1.layers[0].mlp.w_gate_up  # [E, 2I, H] at TP=1
+    I = cfg.intermediate_size
+    half = I // 2
+    for rank in (0, 1):
+        m2 = MixtralForCausalLM(cfg, ParallelContext(world_size=2, rank=rank), dtype=torch.float32)
+        m2.random_init_(seed=0)
+        local = m2.layers[0].mlp.w_gate_up  # [E, I, H]
+        want = torch.cat([
+            full[:, rank * half:(rank + 1) * half],            # gate shard
+            full[:, I + rank * half:I + (rank + 1) * half],    # up shard
+        ], dim=1)
+        assert torch.equal(local, want), f"rank {rank} init layout mismatch"
