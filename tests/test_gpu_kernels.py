@@ -472,9 +472,13 @@ class TestFp8KVCacheGPU:
         from kllms_amd.engine.engine import GenRequest, LLMEngine
         from kllms_amd.engine.sampling import SamplingParams
 
-        eng = LLMEngine(EngineConfig(model="mid-llama", max_kv_blocks=512, use_hip_graphs=False,
+        # hipGraph capture over fp8 caches exercised too (decode graph holds
+        # fp8 cache pointers; replay must hit the fp8-templated kernels)
+        eng = LLMEngine(EngineConfig(model="mid-llama", max_kv_blocks=512, use_hip_graphs=True,
+                                     hip_graph_batch_sizes=[1, 2, 4], max_seq_len=512,
                                      seed=3, kv_cache_dtype="fp8_e4m3"))
         out = eng.generate([GenRequest(prompt_ids=list(range(1, 80)), n=2,
                                        sampling=SamplingParams(temperature=0.0, max_tokens=12))])[0]
+        assert eng._graph_runner is not None and eng._graph_runner._enabled
         assert out.streams[0].token_ids == out.streams[1].token_ids
         assert all(np.isfinite(lp) for lp in out.streams[0].logprobs)
