@@ -174,6 +174,30 @@ torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor B) {
   return D;
 }
 
+extern "C" void launch_one_shot_allreduce(const void* const* srcs, void* const* dsts,
+                                          int n_peers, long numel, int write_all,
+                                          hipStream_t stream);
+
+void one_shot_allreduce(std::vector<torch::Tensor> bufs) {
+  // In-place sum across up to 8 same-shaped bf16 buffers (single-GPU
+  // simulation of the one-shot xGMI all-reduce: every "peer" buffer ends
+  // holding the fp32-accumulated sum). Multi-GPU wiring (IPC-mapped peer
+  // pointers, write_all=0) is round-2 work — docs/ROADMAP.md item 1.
+  TORCH_CHECK(!bufs.empty() && bufs.size() <= 8, "1..8 peer buffers");
+  const auto numel = bufs[0].numel();
+  TORCH_CHECK(numel % 8 == 0, "numel must be a multiple of 8 (16B vector path)");
+  const void* srcs[8];
+  void* dsts[8];
+  for (size_t r = 0; r < bufs.size(); ++r) {
+    CHECK_BF16_CONTIG(bufs[r]);
+    TORCH_CHECK(bufs[r].numel() == numel, "peer buffers must match in size");
+    srcs[r] = bufs[r].data_ptr();
+    dsts[r] = bufs[r].data_ptr();
+  }
+  launch_one_shot_allreduce(srcs, dsts, (int)bufs.size(), (long)numel, /*write_all=*/1,
+                            cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm);
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
@@ -184,4 +208,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill", &attn_prefill);
   m.def("sample", &sample);
   m.def("mfma_selftest", &mfma_selftest);
+  m.def("one_shot_allreduce", &one_shot_allreduce);
 }

@@ -24,6 +24,7 @@ SOURCES = [
         "attn_prefill.hip",
         "sampling.hip",
         "mfma_selftest.hip",
+        "allreduce.hip",
     )
 ]
 

@@ -482,3 +482,28 @@ class TestFp8KVCacheGPU:
         assert eng._graph_runner is not None and eng._graph_runner._enabled
         assert out.streams[0].token_ids == out.streams[1].token_ids
         assert all(np.isfinite(lp) for lp in out.streams[0].logprobs)
+
+
+class TestOneShotAllreduce:
+    def test_allreduce_matches_fp32_sum(self):
+        """One-shot all-reduce (single-GPU peer simulation): all N buffers end
+        holding the fp32-accumulated sum of the inputs (round-2 groundwork
+        for the xGMI one-shot path)."""
+        from kllms_amd import ops as kops
+
+        torch.manual_seed(5)
+        for n_peers in (2, 4, 8):
+            for numel in (8, 4096, 5 * 4096, 120 * 4096):
+                bufs = [torch.randn(numel, dtype=torch.bfloat16, device=DEV) for _ in range(n_peers)]
+                want = torch.stack([b.float() for b in bufs]).sum(0).to(torch.bfloat16)
+                kops._hip_or_raise().one_shot_allreduce(bufs)
+                for r, b in enumerate(bufs):
+                    assert torch.equal(b, want), f"peers={n_peers} numel={numel} rank-buf {r}"
+
+    def test_allreduce_rejects_mismatch(self):
+        from kllms_amd import ops as kops
+
+        a = torch.randn(64, dtype=torch.bfloat16, device=DEV)
+        b = torch.randn(128, dtype=torch.bfloat16, device=DEV)
+        with pytest.raises(RuntimeError):
+            kops._hip_or_raise().one_shot_allreduce([a, b])
