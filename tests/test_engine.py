@@ -192,3 +192,39 @@ class TestFp8KVCache:
         a, b = out_bf16.streams[0].token_ids, out_fp8.streams[0].token_ids
         agree = sum(x == y for x, y in zip(a, b))
         assert agree >= len(a) - 2, f"fp8 cache diverged early: {a} vs {b}"
+
+
+class TestPenalties:
+    def test_apply_penalties_math(self):
+        """OpenAI penalty semantics: logit -= freq_pen * count + pres_pen * present."""
+        from kllms_amd.engine.engine import _Stream
+        from kllms_amd.engine.kvcache import SequenceKV
+
+        eng = LLMEngine(EngineConfig(model="tiny-llama", max_kv_blocks=64,
+                                     use_hip_graphs=False, device="cpu"))
+        V = eng.arch.vocab_size
+        seq = eng.kv.alloc_sequence(4)
+        s = _Stream(0, 0, seq, SamplingParams(frequency_penalty=0.5, presence_penalty=0.25), seed=0)
+        s.out.token_ids = [7, 7, 9]
+        logits = torch.zeros(1, V)
+        out = eng._apply_penalties(logits.clone(), [s])
+        assert out[0, 7].item() == pytest.approx(-(0.5 * 2 + 0.25))
+        assert out[0, 9].item() == pytest.approx(-(0.5 * 1 + 0.25))
+        assert out[0, 3].item() == 0.0
+        # no-penalty stream short-circuits untouched
+        s2 = _Stream(0, 0, seq, SamplingParams(), seed=0)
+        s2.out.token_ids = [7]
+        out2 = eng._apply_penalties(logits.clone(), [s2])
+        assert torch.equal(out2, logits)
+        eng.kv.free_sequence(seq)
+
+    def test_frequency_penalty_discourages_repeats(self):
+        """Greedy decode with a huge frequency penalty cannot emit the same
+        token twice in a row (its logit drops immediately)."""
+        eng = LLMEngine(EngineConfig(model="tiny-llama", max_kv_blocks=256,
+                                     use_hip_graphs=False, device="cpu", seed=0))
+        out = eng.generate([GenRequest(
+            prompt_ids=[5, 6, 7], n=1,
+            sampling=SamplingParams(temperature=0.0, max_tokens=12, frequency_penalty=100.0))])[0]
+        toks = out.streams[0].token_ids
+        assert len(toks) == len(set(toks)), f"repeated token under huge penalty: {toks}"
