@@ -228,3 +228,63 @@ class TestPenalties:
             sampling=SamplingParams(temperature=0.0, max_tokens=12, frequency_penalty=100.0))])[0]
         toks = out.streams[0].token_ids
         assert len(toks) == len(set(toks)), f"repeated token under huge penalty: {toks}"
+
+
+class TestStopStrings:
+    def test_stop_string_trims_text(self):
+        eng = LLMEngine(EngineConfig(model="tiny-llama", max_kv_blocks=256,
+                                     use_hip_graphs=False, device="cpu", seed=0))
+        base = eng.generate([GenRequest(
+            prompt_ids=[9, 9, 9], n=1,
+            sampling=SamplingParams(temperature=0.0, max_tokens=10))])[0]
+        text = base.streams[0].text
+        assert len(text) > 8
+        # pick a substring the greedy continuation definitely contains
+        stop = text[4:8]
+        out = eng.generate([GenRequest(
+            prompt_ids=[9, 9, 9], n=1,
+            sampling=SamplingParams(temperature=0.0, max_tokens=10, stop=stop))])[0]
+        s = out.streams[0]
+        assert s.finish_reason == "stop"
+        assert stop not in s.text
+        assert text.startswith(s.text)
+
+    def test_stop_list(self):
+        eng = LLMEngine(EngineConfig(model="tiny-llama", max_kv_blocks=256,
+                                     use_hip_graphs=False, device="cpu", seed=0))
+        base = eng.generate([GenRequest(
+            prompt_ids=[2, 4, 6], n=1,
+            sampling=SamplingParams(temperature=0.0, max_tokens=10))])[0]
+        stop2 = base.streams[0].text[3:6]
+        out = eng.generate([GenRequest(
+            prompt_ids=[2, 4, 6], n=1,
+            sampling=SamplingParams(temperature=0.0, max_tokens=10,
+                                    stop=["@@never@@", stop2]))])[0]
+        assert out.streams[0].finish_reason == "stop"
+        assert stop2 not in out.streams[0].text
+
+
+class TestServingMetadata:
+    def test_timings_and_create_many(self):
+        from kllms_amd.engine.api import LocalEngineClient
+
+        client = LocalEngineClient(model="tiny-llama", max_kv_blocks=512,
+                                   use_hip_graphs=False, device="cpu", seed=0)
+        r = client.chat_completions_create(
+            messages=[{"role": "user", "content": "hello"}], model="tiny-llama",
+            n=2, max_tokens=4, temperature=0.0, seed=1)
+        tm = r.timings
+        assert tm["completion_tokens"] == r.usage.completion_tokens
+        for key in ("prefill_ms", "decode_ms", "decode_steps", "decode_tokens_per_s_batch"):
+            assert key in tm, f"missing timing {key}: {tm}"
+        assert tm["prefill_ms"] > 0 and tm["decode_ms"] > 0
+
+        # create_many == individual create for seeded greedy requests
+        params = [dict(messages=[{"role": "user", "content": f"p{i}"}],
+                       model="tiny-llama", n=2, max_tokens=5, temperature=0.0, seed=10 + i)
+                  for i in range(3)]
+        batch = client.chat_completions_create_many([dict(p) for p in params])
+        singles = [client.chat_completions_create(**dict(p)) for p in params]
+        for b, s in zip(batch, singles):
+            assert [c.message.content for c in b.choices] == [c.message.content for c in s.choices]
+            assert b.usage.prompt_tokens == s.usage.prompt_tokens
