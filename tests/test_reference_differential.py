@@ -195,3 +195,59 @@ def test_lists_alignment_matches_reference(ref, lists):
     got, got_idx = lists_alignment([list(l) for l in lists], ours_sim, min_support_ratio=0.51)
     _deep_eq(got, want, "aligned")
     _deep_eq(got_idx, want_idx, "orig_idx")
+
+
+# targeted distributions for rarely-hit branches ----------------------------
+
+_centers = st.sampled_from([0.0, 1.0, -1.0, 10.0, 0.001, -1000.0, 1e6])
+_jitter = st.floats(min_value=-0.05, max_value=0.05, allow_nan=False)
+clustered_numbers = st.lists(
+    st.one_of(
+        st.tuples(_centers, _jitter).map(lambda t: t[0] * (1.0 + t[1])),
+        st.tuples(_centers, _jitter).map(lambda t: t[0] * (1.0 + t[1]) * 10.0),   # power-of-10 twins
+        st.tuples(_centers, _jitter).map(lambda t: -t[0] * (1.0 + t[1])),         # signless twins
+        st.none(),
+        st.integers(min_value=-5, max_value=5),
+    ),
+    min_size=2, max_size=7,
+)
+
+
+@settings(max_examples=200, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(clustered_numbers)
+def test_numeric_consensus_matches_reference(ref, values):
+    """Hybrid numeric clustering incl. tie-breaks by cross-cluster support,
+    signless and power-of-10 equivalence (ref :1127-1219)."""
+    try:
+        want = ref.consensus_values(list(values), ref.ConsensusSettings(), fake_embed)
+    except Exception:
+        return
+    got = consensus_values(list(values), ConsensusSettings(), fake_embed)
+    _deep_eq(got[0], want[0], "value")
+    _deep_eq(got[1], want[1], "confidence")
+
+
+@settings(max_examples=150, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(json_values, json_values, st.sampled_from(["jaccard", "hamming", "embeddings"]))
+def test_similarity_methods_match_reference(ref, a, b, method):
+    """All four string-similarity methods, incl. the >50-char embeddings gate
+    falling back to levenshtein below it (ref :797-824)."""
+    ours = generic_similarity(a, b, method, fake_embed)
+    theirs = ref.generic_similarity(a, b, method, fake_embed)
+    assert ours == pytest.approx(theirs, abs=1e-9), f"{method}: {a!r} vs {b!r}"
+
+
+@settings(max_examples=60, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(st.lists(st.text(alphabet="abcdefgh ", min_size=45, max_size=70), min_size=2, max_size=4))
+def test_long_string_embeddings_consensus_matches_reference(ref, values):
+    """string_consensus 'centroid' on long strings routes through the
+    embeddings medoid on both sides (same injected embedder)."""
+    s_ours = ConsensusSettings()          # embeddings + centroid defaults
+    s_ref = ref.ConsensusSettings()
+    try:
+        want = ref.consensus_values(list(values), s_ref, fake_embed)
+    except Exception:
+        return
+    got = consensus_values(list(values), s_ours, fake_embed)
+    _deep_eq(got[0], want[0], "value")
+    _deep_eq(got[1], want[1], "confidence")
