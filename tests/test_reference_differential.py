@@ -370,3 +370,57 @@ def test_consolidate_chat_completions_matches_reference(ref_consolidation, conte
         assert g.message.content == w.message.content, f"choice {g.index}"
     _deep_eq(got.likelihoods, want.likelihoods, "likelihoods")
     assert got.usage.model_dump() == want.usage.model_dump()
+
+
+@settings(max_examples=60, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(st.lists(content_strings, min_size=1, max_size=4))
+def test_async_consolidation_matches_reference(ref_consolidation, contents):
+    """The reference's ASYNC consolidation is hand-duplicated code (not a
+    bridge) and has genuinely drifted from its own sync path on bool/number
+    mixtures (e.g. contents 0 vs false: sync -> numeric consensus 0.0 at
+    confidence 0.5, async -> enum-like 0 at confidence 1.0). Ours is one
+    shared impl bridged to async, so it matches the reference's SYNC
+    behavior everywhere — asserted here; where the reference's async agrees
+    with its sync, ours must match it too (docs/PARITY.md)."""
+    import asyncio
+
+    from kllms_amd.consensus.aio import async_consolidate_chat_completions
+
+    async def aembed(texts):
+        return fake_embed(texts)
+
+    async def run_ref():
+        return await ref_consolidation.async_consolidate_chat_completions(
+            _mk_completion(list(contents)), aembed, None
+        )
+
+    async def run_ours():
+        return await async_consolidate_chat_completions(_mk_completion(list(contents)), aembed)
+
+    try:
+        want_async = asyncio.run(run_ref())
+        want_sync = ref_consolidation.consolidate_chat_completions(
+            _mk_completion(list(contents)), fake_embed, None
+        )
+    except Exception as e:
+        if "unhashable" in str(e):
+            return
+        with pytest.raises(type(e)):
+            asyncio.run(run_ours())
+        return
+    got = asyncio.run(run_ours())
+    # ours always equals the reference's SYNC behavior
+    assert len(got.choices) == len(want_sync.choices)
+    for g, w in zip(got.choices, want_sync.choices):
+        assert (g.index, g.finish_reason, g.message.content) == (w.index, w.finish_reason, w.message.content)
+    _deep_eq(got.likelihoods, want_sync.likelihoods, "likelihoods(sync)")
+    assert got.usage.model_dump() == want_sync.usage.model_dump()
+    # and equals the async result wherever the reference hasn't drifted
+    ref_agrees = all(
+        a.message.content == b.message.content
+        for a, b in zip(want_async.choices, want_sync.choices)
+    ) and want_async.likelihoods == want_sync.likelihoods
+    if ref_agrees:
+        for g, w in zip(got.choices, want_async.choices):
+            assert g.message.content == w.message.content
+        _deep_eq(got.likelihoods, want_async.likelihoods, "likelihoods(async)")
