@@ -77,16 +77,23 @@ def _consensus_over_contents(
 
             precompute_similarity_cache(contents, get_embeddings_from_text)
         if aligner == "key":
-            from .key_based_alignment import recursive_align as align_fn
+            # the reference's key-based aligner takes no embed fn / client
+            # (key matching is exact; ref key_based_alignment.py:350-359)
+            from .key_based_alignment import recursive_align
+
+            aligned_seq, _ = recursive_align(
+                contents,
+                consensus_settings.string_similarity_method,
+                consensus_settings.min_support_ratio,
+            )
         else:
-            align_fn = recursive_list_alignments
-        aligned_seq, _ = align_fn(
-            contents,
-            consensus_settings.string_similarity_method,
-            get_embeddings_from_text,
-            client,
-            consensus_settings.min_support_ratio,
-        )
+            aligned_seq, _ = recursive_list_alignments(
+                contents,
+                consensus_settings.string_similarity_method,
+                get_embeddings_from_text,
+                client,
+                consensus_settings.min_support_ratio,
+            )
         contents = [(d if isinstance(d, dict) else {}) for d in aligned_seq]
     return consensus_values(
         contents,

@@ -283,8 +283,6 @@ def _materialize_source_view(
 def recursive_align(
     values: Sequence[Any],
     string_similarity_method: str,
-    sync_get_openai_embeddings_from_text=None,
-    client: Any = None,
     min_support_ratio: float = 0.5,
     max_novelty_ratio: float = 0.25,
     current_path: str = "",
@@ -292,9 +290,11 @@ def recursive_align(
     min_uniqueness: Optional[float] = None,
     min_coverage: Optional[float] = None,
 ) -> Tuple[Sequence[Any], Dict[str, List[Optional[str]]]]:
-    """Key-based recursive alignment with the similarity aligner's API
-    (ref :350-474). ``string_similarity_method`` / embeddings / client args
-    are accepted for signature parity and unused (key matching is exact)."""
+    """Key-based recursive alignment (ref key_based_alignment.py:350-474).
+
+    Signature matches the reference's EXACTLY — note it differs from the
+    similarity aligner's (no embeddings fn / client params: key matching is
+    exact, ``string_similarity_method`` is accepted and unused)."""
     if not values:
         return list(values), {}
     if all(v is None for v in values):

@@ -102,7 +102,7 @@ def test_key_aligner_total_and_shape_preserving(values):
     per-path lists have one entry per source."""
     from kllms_amd.consensus.key_based_alignment import recursive_align
 
-    aligned, km = recursive_align(values, "levenshtein", no_embed, None, 0.5)
+    aligned, km = recursive_align(values, "levenshtein", 0.5)
     assert len(aligned) == len(values)
     for path, sources in km.items():
         assert len(sources) == len(values), (path, sources)
@@ -115,6 +115,6 @@ def test_key_aligner_consensus_pipeline(values):
     range-valid likelihood tree (the 'key' aligner end-to-end)."""
     from kllms_amd.consensus.key_based_alignment import recursive_align
 
-    aligned, _ = recursive_align(values, "levenshtein", no_embed, None, 0.5)
+    aligned, _ = recursive_align(values, "levenshtein", 0.5)
     consensus, conf = consensus_values(list(aligned), SETTINGS, no_embed)
     check_likelihood_tree(conf)

@@ -101,8 +101,12 @@ def _assert_known_ref_crash(e):
     else means the harness itself miscalled the reference (which would make
     every comparison silently vacuous)."""
     msg = f"{type(e).__name__}: {e}"
-    assert "unhashable" in str(e) or isinstance(e, RecursionError), \
-        f"unexpected reference crash treated as divergence: {msg}"
+    known = (
+        "unhashable" in str(e)                      # Counter on list/dict votes
+        or "has no len()" in str(e)                 # key-aligner None-mixed lists
+        or isinstance(e, RecursionError)
+    )
+    assert known, f"unexpected reference crash treated as divergence: {msg}"
 
 
 def fake_embed(texts):
@@ -424,3 +428,66 @@ def test_async_consolidation_matches_reference(ref_consolidation, contents):
         for g, w in zip(got.choices, want_async.choices):
             assert g.message.content == w.message.content
         _deep_eq(got.likelihoods, want_async.likelihoods, "likelihoods(async)")
+
+
+# ---------------------------------------------------------------------------
+# Key-based aligner differential (C42-C44, the dormant-but-shipped L1b)
+# ---------------------------------------------------------------------------
+
+@pytest.fixture(scope="module")
+def ref_key(ref):
+    for sub in ("key_selection", "fuzzy_key_selection", "key_based_alignment"):
+        name = f"_refk.{sub}"
+        if name not in sys.modules:
+            spec = importlib.util.spec_from_file_location(name, os.path.join(REF_UTILS, f"{sub}.py"))
+            mod = importlib.util.module_from_spec(spec)
+            sys.modules[name] = mod
+            spec.loader.exec_module(mod)
+    return sys.modules["_refk.key_based_alignment"]
+
+
+# record-shaped candidates: lists of dicts with scalar fields (what key-based
+# alignment exists for), plus arbitrary JSON to exercise its fallbacks
+_record = st.dictionaries(
+    st.sampled_from(["id", "name", "qty", "price", "tag"]),
+    st.one_of(st.integers(min_value=0, max_value=30), ascii_text, st.none(),
+              st.floats(allow_nan=False, allow_infinity=False, width=16)),
+    min_size=1, max_size=4,
+)
+_record_lists = st.lists(st.lists(_record, max_size=4), min_size=2, max_size=4)
+
+
+@settings(max_examples=100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(st.one_of(_record_lists, st.lists(json_values, min_size=2, max_size=4)))
+def test_key_based_recursive_align_matches_reference(ref_key, values):
+    from kllms_amd.consensus.key_based_alignment import recursive_align
+
+    try:
+        want, want_km = ref_key.recursive_align(list(values), "levenshtein", 0.5)
+    except Exception as e:
+        _assert_known_ref_crash(e)
+        return
+    got, got_km = recursive_align(list(values), "levenshtein", 0.5)
+    _deep_eq(list(got), list(want), "aligned")
+    _deep_eq(got_km, want_km, "key_mappings")
+
+
+@settings(max_examples=100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(_record_lists)
+def test_key_selection_matches_reference(ref_key, lists_of_records):
+    """The cascade's chosen alignment keys must match (C42/C43)."""
+    import _refk.key_selection as ref_sel  # type: ignore[import-not-found]
+
+    from kllms_amd.consensus import key_selection as our_sel
+
+    records = [r for lst in lists_of_records for r in lst]
+    try:
+        want = ref_sel.select_best_keys(list(records))
+    except Exception as e:
+        # ValueError("No extractions"/"No scalar candidate paths") must
+        # reproduce identically on our side (crash parity)
+        with pytest.raises(type(e)):
+            our_sel.select_best_keys(list(records))
+        return
+    got = our_sel.select_best_keys(list(records))
+    _deep_eq(got.model_dump(), want.model_dump(), "selected_keys")
