@@ -13,7 +13,7 @@ identical to the reference mirrors.
 from __future__ import annotations
 
 import asyncio
-from typing import Any, List, Optional, Type, Union
+from typing import Any, Optional, Type
 
 from pydantic import BaseModel
 

@@ -20,7 +20,7 @@ advance — all the heavy work is one-time per (schema, vocab) and cached.
 from __future__ import annotations
 
 import json
-from typing import Any, Dict, List, Optional, Sequence, Tuple
+from typing import Any, Dict, List, Optional, Tuple
 
 import numpy as np
 import torch

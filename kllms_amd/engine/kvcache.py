@@ -13,7 +13,7 @@ Layout per layer: k_cache/v_cache = [num_blocks, kv_heads, block_size, head_dim]
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 

@@ -27,7 +27,7 @@ from __future__ import annotations
 import queue
 import threading
 from concurrent.futures import Future
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 from .engine import GenRequest, LLMEngine, RequestOutput, _DecodeBatchState, _Stream
