@@ -66,15 +66,15 @@ class Opt(_Node):
 _DIGITS = set(b"0123456789")
 _HEX = set(b"0123456789abcdefABCDEF")
 _WS_SET = set(b" \t\n\r")
+# JSON string body chars: printable ASCII except '"' and '\\'. Non-ASCII
+# content is expressible via \uXXXX escapes, which keeps every constrained
+# output valid UTF-8 regardless of how the tokenizer splits bytes.
+_STR_CHAR = {b for b in range(0x20, 0x7F) if b not in (0x22, 0x5C)}
 
 
 def _lead_ws(node: _Node, ws: bool) -> _Node:
     """Optional JSON whitespace before a token (whitespace-tolerant mode)."""
     return Seq([Star(Cls(_WS_SET)), node]) if ws else node
-# JSON string body chars: printable ASCII except '"' and '\\'. Non-ASCII
-# content is expressible via \uXXXX escapes, which keeps every constrained
-# output valid UTF-8 regardless of how the tokenizer splits bytes.
-_STR_CHAR = {b for b in range(0x20, 0x7F) if b not in (0x22, 0x5C)}
 
 
 def _bounded(inner_factory, min_n: int, max_n: Optional[int]) -> _Node:
