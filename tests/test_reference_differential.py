@@ -492,3 +492,41 @@ def test_key_selection_matches_reference(ref_key, lists_of_records):
         return
     got = our_sel.select_best_keys(list(records))
     _deep_eq(got.model_dump(), want.model_dump(), "selected_keys")
+
+
+@settings(max_examples=100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(st.lists(json_values, min_size=1, max_size=5))
+def test_compute_similarity_scores_matches_reference(ref, values):
+    from kllms_amd.consensus.similarity import compute_similarity_scores
+
+    want = ref.compute_similarity_scores(
+        list(values), ref.ConsensusSettings(string_similarity_method="levenshtein"), fake_embed
+    )
+    got = compute_similarity_scores(
+        list(values), ConsensusSettings(string_similarity_method="levenshtein"), fake_embed
+    )
+    _deep_eq(list(got), list(want), "scores")
+
+
+@settings(max_examples=100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(st.lists(st.lists(st.integers(min_value=0, max_value=9), min_size=1, max_size=5),
+                min_size=2, max_size=4))
+def test_sort_by_original_majority_matches_reference(ref, originals):
+    """Condorcet column ordering (C41): identical aligned permutations feed
+    both implementations; identity-keyed original-position lookup must agree."""
+    import random
+
+    from kllms_amd.consensus import sort_by_original_majority
+
+    ms = sys.modules["_refk.majority_sorting"]
+    rng = random.Random(sum(len(o) for o in originals))
+    width = min(len(o) for o in originals)
+    perm = list(range(width))
+    rng.shuffle(perm)
+    aligned = [[row[j] for j in perm] for row in originals]
+    # identity lookup requires the SAME cell objects; ints are interned for
+    # this value range, so both sides see consistent identities
+    want = ms.sort_by_original_majority([list(r) for r in aligned], [list(o) for o in originals])
+    got = sort_by_original_majority([list(r) for r in aligned], [list(o) for o in originals])
+    _deep_eq(list(got[0]), list(want[0]), "sorted_lists")
+    _deep_eq(list(got[1]), list(want[1]), "orig_idx")
