@@ -250,3 +250,21 @@ class TestReferenceScenarios:
             pass
         assert owner.client._scheduler is not None  # borrower must not kill it
         owner.close()
+
+    def test_embeddings_namespace(self, client):
+        """OpenAI-shaped embeddings namespace (client.client.embeddings.create)."""
+        r = client.client.embeddings.create(input=["alpha", "beta"], model="text-embedding-3-small")
+        assert len(r.data) == 2
+        assert r.data[0].index == 0 and r.data[1].index == 1
+        assert len(r.data[0].embedding) == len(r.data[1].embedding) > 0
+        assert r.usage.prompt_tokens > 0
+
+    def test_async_get_embeddings(self):
+        ak = AsyncKLLMs(**TINY)
+
+        async def run():
+            return await ak.get_embeddings(["one", "one", "two"], "text-embedding-3-small", 2048, False)
+
+        embs = asyncio.run(run())
+        assert len(embs) == 3
+        assert embs[0] == embs[1] != embs[2]

@@ -530,3 +530,24 @@ def test_sort_by_original_majority_matches_reference(ref, originals):
     got = sort_by_original_majority([list(r) for r in aligned], [list(o) for o in originals])
     _deep_eq(list(got[0]), list(want[0]), "sorted_lists")
     _deep_eq(list(got[1]), list(want[1]), "orig_idx")
+
+
+@settings(max_examples=120, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(json_values)
+def test_intermediary_cleanup_matches_reference(ref, value):
+    from kllms_amd.consensus.values import intermediary_consensus_cleanup
+
+    _deep_eq(
+        intermediary_consensus_cleanup(value),
+        ref.intermediary_consensus_cleanup(value),
+        "cleanup",
+    )
+
+
+@settings(max_examples=120, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(st.lists(st.floats(min_value=0.0, max_value=1.0, allow_nan=False), max_size=10))
+def test_outlier_cutoff_matches_reference(ref, data):
+    from kllms_amd.consensus import low_cutoff_bound, remove_outliers
+
+    assert low_cutoff_bound(list(data)) == pytest.approx(ref.low_cutoff_bound(list(data)), abs=1e-12)
+    _deep_eq(remove_outliers(list(data)), ref.remove_outliers(list(data)), "outliers")
