@@ -551,3 +551,26 @@ def test_outlier_cutoff_matches_reference(ref, data):
 
     assert low_cutoff_bound(list(data)) == pytest.approx(ref.low_cutoff_bound(list(data)), abs=1e-12)
     _deep_eq(remove_outliers(list(data)), ref.remove_outliers(list(data)), "outliers")
+
+
+@settings(max_examples=200, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(ascii_text)
+def test_string_normalization_matches_reference(ref, s):
+    """normalize_string + sanitize_value: the voting/similarity equivalence
+    keys (transliteration itself is the shared stub; the lower/strip/regex
+    logic is independent code on each side)."""
+    from kllms_amd.consensus.similarity import normalize_string, sanitize_value
+
+    assert normalize_string(s) == ref.normalize_string(s)
+    assert sanitize_value(s) == ref.sanitize_value(s)
+
+
+@settings(max_examples=200, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(st.one_of(st.booleans(), st.integers(-10**6, 10**6),
+                 st.floats(allow_nan=False, allow_infinity=False)),
+       st.one_of(st.booleans(), st.integers(-10**6, 10**6),
+                 st.floats(allow_nan=False, allow_infinity=False)))
+def test_numerical_similarity_matches_reference(ref, a, b):
+    from kllms_amd.consensus.similarity import numerical_similarity
+
+    assert numerical_similarity(a, b) == pytest.approx(ref.numerical_similarity(a, b), abs=1e-12)
