@@ -37,6 +37,13 @@ from .config import EngineConfig
 from .sampling import SamplingParams
 
 
+class ContextLengthExceededError(ValueError):
+    """Prompt does not fit in the model's context window (OpenAI-compatible
+    behavior: the API raises context_length_exceeded rather than silently
+    truncating the prompt, which would cut the chat template's assistant
+    header and make the model continue the user text)."""
+
+
 class LocalEngineClient:
     """Engine handle with .chat/.beta/.embeddings namespaces."""
 
@@ -116,8 +123,7 @@ class LocalEngineClient:
         eng = self.engine
         prompt = eng.tokenizer.apply_chat_template(messages)
         prompt_ids = eng.tokenizer.encode(prompt)
-        max_prompt = self.config.max_seq_len - (sampling.max_tokens or self.config.default_max_new_tokens)
-        prompt_ids = prompt_ids[: max(1, max_prompt)]
+        self._fit_context(prompt_ids, sampling)
 
         req = GenRequest(prompt_ids=prompt_ids, n=n, sampling=sampling, constraint=constraint)
         if scheduled:
@@ -126,6 +132,20 @@ class LocalEngineClient:
             with self._engine_lock:
                 out = eng.generate([req])[0]
         return out, model, sampling
+
+    def _fit_context(self, prompt_ids: List[int], sampling: SamplingParams) -> None:
+        """Error on over-long prompts (never truncate — that would cut the
+        chat template's assistant header); clamp max_tokens to the remaining
+        window when the prompt itself fits."""
+        max_seq = self.config.max_seq_len
+        if len(prompt_ids) >= max_seq:
+            raise ContextLengthExceededError(
+                f"prompt is {len(prompt_ids)} tokens but the context window is "
+                f"{max_seq} tokens (context_length_exceeded)"
+            )
+        room = max_seq - len(prompt_ids)
+        want = sampling.max_tokens or self.config.default_max_new_tokens
+        sampling.max_tokens = min(want, room)
 
     def _build_constraint(self, response_format: Any, constrained: bool):
         from .constrained import JsonSchemaConstraint
@@ -220,8 +240,7 @@ class LocalEngineClient:
                 constraint = self._build_constraint(rf, constrained=False)
             prompt = eng.tokenizer.apply_chat_template(messages)
             prompt_ids = eng.tokenizer.encode(prompt)
-            max_prompt = eng.config.max_seq_len - (sampling.max_tokens or self.config.default_max_new_tokens)
-            prompt_ids = prompt_ids[: max(1, max_prompt)]
+            self._fit_context(prompt_ids, sampling)
             reqs.append(GenRequest(prompt_ids=prompt_ids, n=n, sampling=sampling, constraint=constraint))
             samplings.append(sampling)
         with self._engine_lock:

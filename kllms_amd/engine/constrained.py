@@ -364,7 +364,8 @@ class JsonSchemaConstraint:
 
     def __init__(self, schema: Dict[str, Any], tokenizer, whitespace: bool = False):
         self.schema = schema
-        key = (json.dumps(schema, sort_keys=True), id(type(tokenizer)), tokenizer.vocab_size, whitespace)
+        tok_key = getattr(tokenizer, "cache_key", None) or (id(type(tokenizer)), tokenizer.vocab_size)
+        key = (json.dumps(schema, sort_keys=True), tok_key, whitespace)
         cached = _TABLE_CACHE.get(key)  # type: ignore[arg-type]
         if cached is not None:
             self.__dict__.update(cached.__dict__)
@@ -415,7 +416,10 @@ class JsonSchemaConstraint:
         # dead-end guard: states with no allowed token at all -> EOS-only
         empty = ~allowed.any(axis=1) & ~accepting
         if empty.any() and self.eos_id is not None:
-            self._mask_t[np.nonzero(empty)[0], self.eos_id // 32] |= np.int32(1 << (self.eos_id % 32))
+            # build the bit in uint32 and view as int32: 1 << 31 overflows
+            # np.int32 directly (NumPy >= 1.24 raises OverflowError)
+            eos_bit = int(np.uint32(1 << (self.eos_id % 32)).view(np.int32))
+            self._mask_t[np.nonzero(empty)[0], self.eos_id // 32] |= eos_bit
         # terminal = accepting with no possible continuation (e.g. the closing
         # brace of the top-level object): generation stops there immediately.
         # Accepting-but-extendable states (e.g. "12" under an integer schema)

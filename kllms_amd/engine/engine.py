@@ -138,6 +138,8 @@ class LLMEngine:
         else:
             self.ctx = ParallelContext()
 
+        # monotonic per-request counter for unseeded-request RNG derivation
+        self._seed_counter = 0
         self.dtype = {"bfloat16": torch.bfloat16, "float16": torch.float16, "float32": torch.float32}[config.dtype]
         # CPU bf16 matmuls are slow and torch CPU attention paths prefer f32
         if self.device.type == "cpu" and self.dtype == torch.bfloat16:
@@ -333,7 +335,13 @@ class LLMEngine:
         prefill logits row."""
         new_streams: List[_Stream] = []
         for ri, req in enumerate(requests):
-            base_seed = req.sampling.seed if req.sampling.seed is not None else (self.config.seed * 1000003 + ri)
+            # unseeded requests draw from a per-engine monotonic counter, NOT
+            # the batch-local index: sampled output must not depend on which
+            # other requests were admitted together (the scheduler's
+            # result-transparent merging contract)
+            rid = self._seed_counter
+            self._seed_counter += 1
+            base_seed = req.sampling.seed if req.sampling.seed is not None else (self.config.seed * 1000003 + rid)
             for si in range(max(1, req.n)):
                 seq = self.kv.fork(parent_seqs[ri])
                 cstate = req.constraint.init_state() if req.constraint is not None else None

@@ -162,18 +162,25 @@ class BatchScheduler:
         chunk = eng.config.prefill_chunk_tokens or len(ids)
         end = min(p.next_pos + chunk, len(ids))
         final = end == len(ids)
+        popped = False
+        new_streams: List[_Stream] = []
         try:
             logits = eng.prefill_chunk(p.seq, ids, p.next_pos, end, want_logits=final)
             p.next_pos = end
             if not final:
                 return
             ctx.pending.pop(0)
-            new_streams: List[_Stream] = []
+            popped = True
             eng._fork_and_sample([t.request], [p.seq], logits, new_streams)
         except Exception as e:
-            ctx.pending.pop(0)
+            # pop exactly once: only if the try block didn't get there
+            if not popped:
+                ctx.pending.pop(0)
             if p.seq.blocks:
                 eng.kv.free_sequence(p.seq)
+            for st in new_streams:
+                if st.seq.blocks:
+                    eng.kv.free_sequence(st.seq)
             if not t.future.done():
                 t.future.set_exception(e)
             return

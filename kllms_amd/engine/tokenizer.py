@@ -57,6 +57,7 @@ class ByteTokenizer(BaseTokenizer):
         self.bos_id = self.N_BYTES
         self.eos_id = self.N_BYTES + 1
         self.pad_id = self.N_BYTES + 2
+        self.cache_key = ("byte", vocab_size)
 
     def encode(self, text: str, add_bos: bool = False) -> list[int]:
         ids = list(text.encode("utf-8", errors="replace"))
@@ -106,18 +107,72 @@ class ByteTokenizer(BaseTokenizer):
         return f"<tk{i}>".encode()
 
 
+def _bytelevel_unicode_to_byte() -> dict:
+    """Inverse of the GPT-2/Llama-3 ByteLevel byte->unicode alphabet: every
+    raw byte is represented in vocab strings by a printable unicode char
+    (0x20 -> 'Ġ', etc.). Standard table from the GPT-2 tokenizer."""
+    bs = list(range(ord("!"), ord("~") + 1)) + list(range(0xA1, 0xAD)) + list(range(0xAE, 0x100))
+    cs = bs[:]
+    n = 0
+    for b in range(256):
+        if b not in bs:
+            bs.append(b)
+            cs.append(256 + n)
+            n += 1
+    return {chr(c): b for b, c in zip(bs, cs)}
+
+
 class HFTokenizer(BaseTokenizer):
-    """Wraps a HuggingFace ``tokenizers`` fast tokenizer file."""
+    """Wraps a HuggingFace ``tokenizers`` fast tokenizer file.
+
+    ``token_bytes`` derives each token's TRUE byte string from the vocab
+    surface form — ByteLevel alphabet chars (Ġ = space, Ċ = newline, …)
+    mapped through the inverse GPT-2 table, SentencePiece ▁ markers mapped
+    to spaces and ``<0xNN>`` byte-fallback tokens to their raw byte — NOT
+    from per-token ``decode([i])``, which mangles byte-fallback and marker
+    tokens (reference boundary this feeds:
+    /root/reference/k_llms/resources/completions/completions.py:134 — the
+    server-side parse() must constrain over the real vocab)."""
 
     def __init__(self, tokenizer_json: str):
+        import json as _json
+
         from tokenizers import Tokenizer  # offline wheelhouse package
 
         self._tok = Tokenizer.from_file(tokenizer_json)
         self.vocab_size = self._tok.get_vocab_size()
+        self.cache_key = ("hf", os.path.abspath(tokenizer_json), os.path.getmtime(tokenizer_json))
         vocab = self._tok.get_vocab()
-        self.bos_id = vocab.get("<|begin_of_text|>") or vocab.get("<s>")
-        self.eos_id = vocab.get("<|eot_id|>") or vocab.get("<|end_of_text|>") or vocab.get("</s>")
-        self.pad_id = self.eos_id or 0
+        def _first_id(*names):
+            # `or`-chaining is wrong here: a valid token id of 0 is falsy
+            for nm in names:
+                if nm in vocab:
+                    return vocab[nm]
+            return None
+
+        self.bos_id = _first_id("<|begin_of_text|>", "<s>")
+        self.eos_id = _first_id("<|eot_id|>", "<|end_of_text|>", "</s>")
+        self.pad_id = self.eos_id if self.eos_id is not None else 0
+
+        # id -> vocab surface string, and the set of special/added ids
+        self._id_to_token = {i: t for t, i in vocab.items()}
+        with open(tokenizer_json, "r", encoding="utf-8") as f:
+            spec = _json.load(f)
+        self._special_ids = {a["id"] for a in spec.get("added_tokens", []) if a.get("special")}
+        self._u2b = _bytelevel_unicode_to_byte()
+        # scheme detection: ByteLevel decoder/pretokenizer => GPT-2 alphabet;
+        # otherwise SentencePiece-style (metaspace + <0xNN> byte fallback)
+        def _has_bytelevel(node) -> bool:
+            if isinstance(node, dict):
+                if node.get("type") == "ByteLevel":
+                    return True
+                return any(_has_bytelevel(v) for v in node.values())
+            if isinstance(node, list):
+                return any(_has_bytelevel(v) for v in node)
+            return False
+
+        self._byte_level = _has_bytelevel(spec.get("decoder")) or _has_bytelevel(spec.get("pre_tokenizer"))
+        self._tb_cache: dict = {}
 
     def encode(self, text: str, add_bos: bool = False) -> list[int]:
         ids = self._tok.encode(text, add_special_tokens=False).ids
@@ -129,8 +184,32 @@ class HFTokenizer(BaseTokenizer):
         return self._tok.decode(list(ids), skip_special_tokens=True)
 
     def token_bytes(self, i: int) -> Optional[bytes]:
-        s = self._tok.decode([i], skip_special_tokens=True)
-        return s.encode() if s else None
+        if i in self._tb_cache:
+            return self._tb_cache[i]
+        b = self._token_bytes_uncached(i)
+        self._tb_cache[i] = b
+        return b
+
+    def _token_bytes_uncached(self, i: int) -> Optional[bytes]:
+        if i in self._special_ids:
+            return None
+        s = self._id_to_token.get(i)
+        if s is None:
+            return None
+        if self._byte_level:
+            try:
+                return bytes(self._u2b[ch] for ch in s)
+            except KeyError:
+                # not in the ByteLevel alphabet (e.g. an added non-special
+                # token stored verbatim): take its UTF-8 bytes
+                return s.encode("utf-8")
+        # SentencePiece-style vocab
+        if len(s) == 6 and s.startswith("<0x") and s.endswith(">"):
+            try:
+                return bytes([int(s[3:5], 16)])
+            except ValueError:
+                pass
+        return s.replace("▁", " ").encode("utf-8")
 
 
 def load_tokenizer(model: str, weights_path: Optional[str], vocab_size: int) -> BaseTokenizer:

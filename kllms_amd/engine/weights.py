@@ -39,11 +39,14 @@ def load_safetensors_weights(model, weights_dir: str, ctx: ParallelContext) -> N
 
     # staging for fused tensors
     staged: Dict[str, dict] = {}
+    assigned: set = set()
+    expert_loads: Dict[str, int] = {}   # "layers.{L}.mlp.w_gate_up" -> #experts loaded
 
     def assign(our_name: str, tensor: torch.Tensor):
         p = params[our_name]
         assert p.shape == tensor.shape, (our_name, p.shape, tensor.shape)
         p.copy_(tensor.to(p.dtype))
+        assigned.add(our_name)
 
     for f in files:
         with safe_open(f, framework="pt") as sf:
@@ -96,7 +99,31 @@ def load_safetensors_weights(model, weights_dir: str, ctx: ParallelContext) -> N
                         if len(st) == 2:
                             model.load_expert_(int(layer), e, "gate_up", torch.cat([st["gate"], st["up"]], dim=0))
                             del staged[key]
+                            k = f"layers.{layer}.mlp.w_gate_up"
+                            expert_loads[k] = expert_loads.get(k, 0) + 1
                     else:  # w2
                         model.load_expert_(int(layer), e, "down", _shard(t, 1, rank, world))
+                        k = f"layers.{layer}.mlp.w_down"
+                        expert_loads[k] = expert_loads.get(k, 0) + 1
     if staged:
         raise RuntimeError(f"incomplete fused groups after load: {list(staged)[:4]}")
+
+    # tied-embedding checkpoints (e.g. Llama-3.2) ship no lm_head.weight
+    if cfg.tie_word_embeddings and "lm_head.weight" not in assigned:
+        params["lm_head.weight"].copy_(params["embed_tokens.weight"])
+        assigned.add("lm_head.weight")
+
+    # completeness: a parameter the checkpoint never touched keeps its random
+    # init and silently produces garbage logits — fail loudly instead.
+    missing = []
+    for name in params:
+        if name in assigned:
+            continue
+        if name in expert_loads:
+            if expert_loads[name] == cfg.num_experts:
+                continue
+            missing.append(f"{name} ({expert_loads[name]}/{cfg.num_experts} experts)")
+            continue
+        missing.append(name)
+    if missing:
+        raise RuntimeError(f"checkpoint left {len(missing)} parameters unassigned: {missing[:6]}")

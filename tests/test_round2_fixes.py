@@ -1,0 +1,334 @@
+"""Round-2 regression tests: advisor findings (scheduler double-pop, tied
+embeddings, context-length error, DFA eos-bit overflow, batch-independent
+seeds) and real-BPE-vocab fidelity (HFTokenizer.token_bytes derived from the
+vocab, constrained decoding over a genuine ByteLevel BPE tokenizer.json)."""
+
+import json
+import os
+import shutil
+
+import pytest
+import torch
+
+from kllms_amd.engine.config import MODEL_PRESETS, EngineConfig
+from kllms_amd.engine.engine import GenRequest, LLMEngine
+from kllms_amd.engine.sampling import SamplingParams
+from kllms_amd.engine.tokenizer import ByteTokenizer, HFTokenizer
+
+DATA = os.path.join(os.path.dirname(__file__), "data")
+BPE_JSON = os.path.join(DATA, "bpe_tokenizer.json")
+
+
+# ---------------------------------------------------------------------------
+# HFTokenizer.token_bytes on a real ByteLevel BPE vocab
+# ---------------------------------------------------------------------------
+class TestHFTokenizerBytes:
+    @pytest.fixture(scope="class")
+    def tok(self):
+        return HFTokenizer(BPE_JSON)
+
+    def test_vocab_is_byte_level(self, tok):
+        assert tok._byte_level
+        assert any(t.startswith("Ġ") for t in tok._id_to_token.values())  # Ġ markers
+
+    @pytest.mark.parametrize("s", [
+        'Hello world {"key": "value", "n": 42}',
+        "café naïve données",      # multi-byte UTF-8 split across merges
+        "東京 and Ελλάδα",
+        "  leading and   inner   spaces  ",
+        "tabs\tand\nnewlines\r\n",
+    ])
+    def test_token_bytes_concat_equals_utf8(self, tok, s):
+        ids = tok.encode(s)
+        got = b"".join(tok.token_bytes(i) or b"" for i in ids)
+        assert got == s.encode("utf-8")
+
+    def test_every_nonspecial_token_has_bytes(self, tok):
+        for i in range(tok.vocab_size):
+            b = tok.token_bytes(i)
+            if i in tok._special_ids:
+                assert b is None
+            else:
+                assert isinstance(b, bytes) and len(b) >= 1
+
+    def test_space_marker_token_maps_to_space(self, tok):
+        # find a Ġ-prefixed merge and check its bytes start with b' '
+        for i, t in tok._id_to_token.items():
+            if t.startswith("Ġ") and len(t) > 1:
+                assert tok.token_bytes(i).startswith(b" ")
+                return
+        pytest.skip("no multi-char space-marker token in fixture vocab")
+
+    def test_specials_resolved_with_id_zero(self, tok):
+        # <|begin_of_text|> is id 0 in the fixture: `or`-chaining would lose it
+        assert tok.bos_id == 0
+        assert tok.eos_id is not None
+
+    def test_sentencepiece_style_branch(self, tok):
+        # exercise the SP branch directly: metaspace marker + byte fallback
+        import copy
+        sp = copy.copy(tok)
+        sp._byte_level = False
+        sp._tb_cache = {}
+        sp._id_to_token = {10: "▁Hello", 11: "<0x41>", 12: "plain"}
+        sp._special_ids = set()
+        assert sp.token_bytes(10) == b" Hello"
+        assert sp.token_bytes(11) == b"A"
+        assert sp.token_bytes(12) == b"plain"
+
+
+# ---------------------------------------------------------------------------
+# Constrained decoding over the real BPE vocab
+# ---------------------------------------------------------------------------
+class TestConstrainedRealVocab:
+    @pytest.fixture(scope="class")
+    def tok(self):
+        return HFTokenizer(BPE_JSON)
+
+    def _walk(self, c, tok, doc):
+        st = c.init_state()
+        for t in tok.encode(doc):
+            mask = c.allowed_mask(st)
+            assert (int(mask[t // 32].item()) >> (t % 32)) & 1, (
+                f"token {tok._id_to_token.get(t)!r} masked out at state {st}")
+            nxt = int(c.next_state[st, t])
+            assert nxt != 0xFFFF
+            st = nxt
+        return st
+
+    def test_compact_doc_walks_to_accepting(self, tok):
+        from kllms_amd.engine.constrained import JsonSchemaConstraint
+        schema = {"type": "object",
+                  "properties": {"name": {"type": "string"}, "age": {"type": "integer"}},
+                  "required": ["name", "age"]}
+        c = JsonSchemaConstraint(schema, tok)
+        st = self._walk(c, tok, '{"name":"Alice","age":30}')
+        assert bool(c.accepting[st])
+
+    def test_whitespace_doc_walks_to_accepting(self, tok):
+        from kllms_amd.engine.constrained import JsonSchemaConstraint
+        schema = {"type": "object",
+                  "properties": {"name": {"type": "string"}, "age": {"type": "integer"}},
+                  "required": ["name", "age"]}
+        c = JsonSchemaConstraint(schema, tok, whitespace=True)
+        st = self._walk(c, tok, '{"name": "Alice", "age": 30}')
+        assert bool(c.accepting[st])
+
+    def test_masks_differ_from_byte_tokenizer(self, tok):
+        """The DFA x vocab product must be computed per vocab: a multi-byte
+        merge token like '":"' is legal mid-object for the BPE vocab and
+        doesn't exist for the byte tokenizer."""
+        from kllms_amd.engine.constrained import JsonSchemaConstraint
+        schema = {"type": "object", "properties": {"q": {"type": "string"}}, "required": ["q"]}
+        c = JsonSchemaConstraint(schema, tok)
+        st = c.init_state()
+        # at least one allowed token from the start must be multi-byte
+        mask = c.allowed_mask(st)
+        multi = [i for i in range(tok.vocab_size)
+                 if (int(mask[i // 32].item()) >> (i % 32)) & 1
+                 and tok.token_bytes(i) and len(tok.token_bytes(i)) > 1]
+        assert multi, "no multi-byte merge admitted — DFA not using true token bytes"
+
+    def test_eos_bit31_no_overflow(self):
+        """eos_id % 32 == 31 with an empty DFA state used to raise
+        OverflowError (np.int32(1 << 31)) at schema-compile time."""
+        from kllms_amd.engine.constrained import JsonSchemaConstraint
+
+        tok = ByteTokenizer(512)
+        tok.eos_id = 287  # 287 % 32 == 31; still > N_BYTES so byte ids intact
+        tok.cache_key = ("byte-eos31", 512)
+        # enum of a single value makes most states non-accepting; dead-end
+        # states exist wherever no byte continues the literal
+        schema = {"type": "object", "properties": {"v": {"enum": ["x"]}}, "required": ["v"]}
+        c = JsonSchemaConstraint(schema, tok)  # must not raise
+        assert c.eos_id == 287
+
+
+# ---------------------------------------------------------------------------
+# Advisor fixes: engine / scheduler / api / weights
+# ---------------------------------------------------------------------------
+def _tiny_cfg(**kw):
+    base = dict(model="tiny-llama", max_kv_blocks=256, use_hip_graphs=False,
+                default_max_new_tokens=8, device="cpu", seed=0)
+    base.update(kw)
+    return EngineConfig(**base)
+
+
+class TestSeedCounterIndependence:
+    def test_unseeded_output_independent_of_batchmates(self):
+        """An unseeded request's samples must not depend on which other
+        requests were admitted in the same batch (result-transparent merging).
+        Fresh engines so _seed_counter starts equal."""
+        sp = lambda: SamplingParams(temperature=1.0, max_tokens=10)
+        e1 = LLMEngine(_tiny_cfg())
+        solo = e1.generate([GenRequest(prompt_ids=[1, 2, 3], n=2, sampling=sp())])[0]
+        e2 = LLMEngine(_tiny_cfg())
+        pair = e2.generate([
+            GenRequest(prompt_ids=[1, 2, 3], n=2, sampling=sp()),
+            GenRequest(prompt_ids=[9, 8, 7], n=1, sampling=sp()),
+        ])[0]
+        assert [s.token_ids for s in solo.streams] == [s.token_ids for s in pair.streams]
+
+    def test_sequential_unseeded_requests_draw_fresh_streams(self):
+        eng = LLMEngine(_tiny_cfg())
+        sp = lambda: SamplingParams(temperature=1.0, max_tokens=12)
+        a = eng.generate([GenRequest(prompt_ids=list(range(1, 20)), n=1, sampling=sp())])[0]
+        b = eng.generate([GenRequest(prompt_ids=list(range(1, 20)), n=1, sampling=sp())])[0]
+        # same prompt, no seed: the second call must NOT replay the first call's RNG
+        assert a.streams[0].token_ids != b.streams[0].token_ids
+
+
+class TestSchedulerChunkedFailure:
+    def test_fork_failure_does_not_drop_next_pending(self, monkeypatch):
+        """If _fork_and_sample raises on the final chunk, the NEXT pending
+        request must survive (the old code popped ctx.pending twice)."""
+        from kllms_amd.engine.scheduler import BatchScheduler
+
+        eng = LLMEngine(_tiny_cfg(prefill_chunk_tokens=8, max_kv_blocks=512))
+        sched = BatchScheduler(eng)
+
+        orig = eng._fork_and_sample
+        fails = {"left": 1}
+
+        def flaky(requests, parent_seqs, logits, streams):
+            if fails["left"] > 0:
+                fails["left"] -= 1
+                raise RuntimeError("injected fork failure")
+            return orig(requests, parent_seqs, logits, streams)
+
+        monkeypatch.setattr(eng, "_fork_and_sample", flaky)
+        free0 = eng.kv.allocator.num_free
+        sp = SamplingParams(temperature=0.0, max_tokens=4, seed=0)
+        f1 = sched.submit(GenRequest(prompt_ids=list(range(1, 30)), n=2, sampling=sp))
+        f2 = sched.submit(GenRequest(prompt_ids=list(range(1, 25)), n=1, sampling=sp))
+        # first fails with the injected error; second must still complete
+        with pytest.raises(RuntimeError, match="injected"):
+            f1.result(timeout=30)
+        out2 = f2.result(timeout=30)
+        assert len(out2.streams) == 1 and len(out2.streams[0].token_ids) > 0
+        sched.shutdown()
+        assert eng.kv.allocator.num_free == free0  # no leaked KV blocks
+
+
+class TestContextLength:
+    def test_overlong_prompt_raises(self):
+        from kllms_amd.engine.api import ContextLengthExceededError, LocalEngineClient
+
+        client = LocalEngineClient(model="tiny-llama", max_kv_blocks=256, use_hip_graphs=False,
+                                   device="cpu", max_seq_len=64, default_max_new_tokens=8)
+        with pytest.raises(ContextLengthExceededError, match="context_length_exceeded"):
+            client.chat.completions.create(
+                messages=[{"role": "user", "content": "x" * 500}], max_tokens=4)
+
+    def test_max_tokens_clamped_not_prompt_truncated(self):
+        from kllms_amd.engine.api import LocalEngineClient
+
+        client = LocalEngineClient(model="tiny-llama", max_kv_blocks=256, use_hip_graphs=False,
+                                   device="cpu", max_seq_len=256, default_max_new_tokens=8)
+        msgs = [{"role": "user", "content": "y" * 40}]
+        prompt = client.engine.tokenizer.apply_chat_template(msgs)
+        plen = len(client.engine.tokenizer.encode(prompt))
+        assert plen < 256
+        res = client.chat.completions.create(messages=msgs, max_tokens=10_000, temperature=0.0)
+        # full prompt charged (not truncated); completion clamped to the window
+        assert res.usage.prompt_tokens == plen
+        assert res.usage.completion_tokens <= 256 - plen
+
+
+class TestWeightsCompleteness:
+    def _write_ckpt(self, tmp_path, cfg, drop=(), tie=False):
+        from tests.test_weights_io import _make_hf_llama_checkpoint
+        from safetensors.torch import save_file
+
+        tensors = _make_hf_llama_checkpoint(tmp_path, cfg)
+        if drop or tie:
+            for k in drop:
+                tensors.pop(k)
+            if tie:
+                tensors.pop("lm_head.weight", None)
+            save_file({k: v.contiguous() for k, v in tensors.items()},
+                      str(tmp_path / "model.safetensors"))
+        return tensors
+
+    def test_missing_param_raises(self, tmp_path):
+        pytest.importorskip("safetensors")
+        from kllms_amd.engine.weights import load_safetensors_weights
+        from kllms_amd.models.llama import LlamaForCausalLM
+        from kllms_amd.parallel.tp import ParallelContext
+
+        cfg = MODEL_PRESETS["tiny-llama"]
+        self._write_ckpt(tmp_path, cfg, drop=["model.layers.1.mlp.down_proj.weight"])
+        model = LlamaForCausalLM(cfg, ParallelContext(), dtype=torch.float32)
+        with pytest.raises(RuntimeError, match="unassigned"):
+            load_safetensors_weights(model, str(tmp_path), ParallelContext())
+
+    def test_tied_embeddings_copied(self, tmp_path):
+        pytest.importorskip("safetensors")
+        from kllms_amd.engine.weights import load_safetensors_weights
+        from kllms_amd.models.llama import LlamaForCausalLM
+        from kllms_amd.parallel.tp import ParallelContext
+
+        cfg = MODEL_PRESETS["tiny-llama"].model_copy(update={"tie_word_embeddings": True})
+        self._write_ckpt(tmp_path, cfg, tie=True)
+        model = LlamaForCausalLM(cfg, ParallelContext(), dtype=torch.float32)
+        load_safetensors_weights(model, str(tmp_path), ParallelContext())
+        assert torch.equal(model.lm_head.weight, model.embed_tokens.weight)
+
+    def test_untied_missing_lm_head_raises(self, tmp_path):
+        pytest.importorskip("safetensors")
+        from kllms_amd.engine.weights import load_safetensors_weights
+        from kllms_amd.models.llama import LlamaForCausalLM
+        from kllms_amd.parallel.tp import ParallelContext
+
+        cfg = MODEL_PRESETS["tiny-llama"]
+        self._write_ckpt(tmp_path, cfg, tie=True)  # drops lm_head, tie NOT set
+        model = LlamaForCausalLM(cfg, ParallelContext(), dtype=torch.float32)
+        with pytest.raises(RuntimeError, match="unassigned"):
+            load_safetensors_weights(model, str(tmp_path), ParallelContext())
+
+
+# ---------------------------------------------------------------------------
+# End-to-end: engine serving a real-BPE-vocab model dir (HF-style layout)
+# ---------------------------------------------------------------------------
+class TestRealVocabEndToEnd:
+    @pytest.fixture(scope="class")
+    def model_dir(self, tmp_path_factory):
+        pytest.importorskip("safetensors")
+        from tests.test_weights_io import _make_hf_llama_checkpoint
+
+        d = tmp_path_factory.mktemp("hfmodel")
+        arch = MODEL_PRESETS["tiny-llama"].model_copy(update={"vocab_size": 571})
+        hf_cfg = {
+            "model_type": "llama", "vocab_size": 571, "hidden_size": arch.hidden_size,
+            "intermediate_size": arch.intermediate_size, "num_hidden_layers": arch.num_layers,
+            "num_attention_heads": arch.num_heads, "num_key_value_heads": arch.num_kv_heads,
+            "rope_theta": arch.rope_theta, "rms_norm_eps": arch.rms_norm_eps,
+            "max_position_embeddings": 512, "tie_word_embeddings": False,
+        }
+        with open(d / "config.json", "w") as f:
+            json.dump(hf_cfg, f)
+        shutil.copy(BPE_JSON, d / "tokenizer.json")
+        _make_hf_llama_checkpoint(d, arch)
+        return str(d)
+
+    def test_engine_loads_dir_with_hf_tokenizer(self, model_dir):
+        cfg = EngineConfig(model=model_dir, max_kv_blocks=256, use_hip_graphs=False,
+                           device="cpu", default_max_new_tokens=8, max_seq_len=256)
+        eng = LLMEngine(cfg)
+        assert isinstance(eng.tokenizer, HFTokenizer)
+        assert eng.model.lm_head.weight.shape[0] == 571
+        out = eng.generate([GenRequest(
+            prompt_ids=eng.tokenizer.encode("The capital of France is"),
+            n=2, sampling=SamplingParams(temperature=0.0, max_tokens=6))])[0]
+        assert len(out.streams) == 2
+        assert out.streams[0].token_ids == out.streams[1].token_ids  # greedy
+
+    def test_public_client_over_real_vocab(self, model_dir):
+        from kllms_amd import KLLMs
+
+        kllms = KLLMs(model=model_dir, max_kv_blocks=256, use_hip_graphs=False,
+                      device="cpu", default_max_new_tokens=6, max_seq_len=256)
+        res = kllms.chat.completions.create(
+            model="local", messages=[{"role": "user", "content": "What is 2+3?"}], n=3, temperature=0.8)
+        assert len(res.choices) == 4  # consensus + 3 originals
+        assert res.choices[0].index == 0
