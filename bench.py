@@ -4,20 +4,35 @@
 Measures the BASELINE.json headline metric — end-to-end consensus requests/sec
 (and per-request latency) at n=5 on Llama-3-8B TP=1 — on synthetic prompts and
 random-init weights (no network). Each timed step serves a fixed batch of
-consensus requests end to end: shared prefill, n fanned decode streams per
-request (HIP kernels, paged KV, hipGraph decode), then alignment + consensus
-consolidation into the final KLLMsChatCompletion.
+consensus requests END TO END **through the public client API**
+(``AsyncKLLMs().chat.completions.create`` — the literal L4 contract,
+reference k_llms/client.py:31-72): concurrent create() calls merge in the
+continuous-batching scheduler (shared prefill, n fanned decode streams per
+request, HIP kernels, paged KV, hipGraph decode), and each call returns a
+consolidated KLLMsChatCompletion with likelihoods — consensus inside the
+timed region.
+
+Configs (BASELINE.json):
+  --config llama8b   (default) chat.completions.create n=5, Llama-3-8B
+  --config parse16   chat.completions.parse, nested Pydantic schema,
+                     n=16 + likelihoods (BASELINE config 3)
+  --config mixtral8  Mixtral-8x7B parse() n=8 (BASELINE config 5)
+  --config llama70b  Llama-3-70B create n=5 (BASELINE config 4; TP=8 with
+                     --parallel tp under an 8-rank launch)
 
 Multi-GPU (--gpus N, launched by torch.distributed.run with one rank per GPU):
-data-parallel serving — each rank runs an independent TP=1 engine on its GPU
-and serves its own request stream (weak scaling; the reported value is the
-whole-job aggregate across ranks).
+default is data-parallel weak scaling — each rank runs an independent TP=1
+engine serving its own request stream; the reported value is the whole-job
+aggregate. ``--parallel tp`` instead shards ONE engine across all ranks
+(column/row-parallel layers + RCCL/xGMI all-reduce): rank 0 runs the client
+and scheduler, other ranks follow in lockstep (parallel/serve.py).
 
 Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
 Rank 0 prints exactly one JSON result line.
 """
 
 import argparse
+import asyncio
 import json
 import os
 import sys
@@ -27,7 +42,13 @@ import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-REFERENCE_BASELINE_REQ_S = 1.2  # consensus n=3 throughput vs OpenAI API
+REFERENCE_BASELINE = {
+    # BASELINE.md rows (remote OpenAI API numbers, context bar)
+    "llama8b": 1.2,    # consensus n=3 throughput (req/s)
+    "llama70b": 1.2,
+    "parse16": 0.7,    # structured output throughput
+    "mixtral8": 0.7,
+}
 
 
 def log(msg):
@@ -94,18 +115,61 @@ def make_prompt(rank: int, step: int, i: int, prompt_len_tokens: int) -> str:
     return (seedtxt + body)[:prompt_len_tokens]
 
 
+def make_schema():
+    """Nested Pydantic schema for parse() configs (mirrors the reference's
+    Company/Department test shape, README_TESTS.md:63-73)."""
+    from typing import List
+
+    from pydantic import BaseModel
+
+    class Employee(BaseModel):
+        name: str
+        role: str
+
+    class Department(BaseModel):
+        name: str
+        headcount: int
+        employees: List[Employee]
+
+    class Company(BaseModel):
+        company: str
+        founded: int
+        departments: List[Department]
+
+    return Company
+
+
+CONFIG_DEFAULTS = {
+    # model, n, mode, batch, max_new
+    "llama8b": dict(model="llama-3-8b", n=5, mode="create", batch=24, max_new=64),
+    "parse16": dict(model="llama-3-8b", n=16, mode="parse", batch=4, max_new=200),
+    "mixtral8": dict(model="mixtral-8x7b", n=8, mode="parse", batch=4, max_new=200),
+    "llama70b": dict(model="llama-3-70b", n=5, mode="create", batch=4, max_new=64),
+}
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
-    ap.add_argument("--model", type=str, default="llama-3-8b")
-    ap.add_argument("--n", type=int, default=5)
-    ap.add_argument("--batch", type=int, default=24, help="consensus requests per step per GPU")
+    ap.add_argument("--config", type=str, default="llama8b", choices=list(CONFIG_DEFAULTS))
+    ap.add_argument("--model", type=str, default=None)
+    ap.add_argument("--n", type=int, default=None)
+    ap.add_argument("--batch", type=int, default=None, help="consensus requests per step per engine")
     ap.add_argument("--prompt-len", type=int, default=512, help="approx prompt tokens")
-    ap.add_argument("--max-new", type=int, default=64, help="decode tokens per stream")
+    ap.add_argument("--max-new", type=int, default=None, help="decode tokens per stream")
     ap.add_argument("--no-graphs", action="store_true")
+    ap.add_argument("--parallel", type=str, default="dp", choices=["dp", "tp"],
+                    help="dp: one engine per rank (weak scaling); tp: ONE engine sharded over all ranks")
     args = ap.parse_args()
+
+    cfgd = CONFIG_DEFAULTS[args.config]
+    model = args.model or cfgd["model"]
+    n = args.n if args.n is not None else cfgd["n"]
+    batch = args.batch if args.batch is not None else cfgd["batch"]
+    max_new = args.max_new if args.max_new is not None else cfgd["max_new"]
+    mode = cfgd["mode"]
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -122,117 +186,152 @@ def main():
             torch.cuda.set_device(local_rank)
 
     device = f"cuda:{local_rank}" if on_gpu else "cpu"
+    tp_mode = args.parallel == "tp" and world_size > 1
 
     if on_gpu:
         _enable_tunableop(local_rank)
 
-    from kllms_amd import KLLMs
-    from kllms_amd.consensus.consolidation import consolidate_chat_completions
+    from kllms_amd import AsyncKLLMs
 
-    log(f"[bench] building engine: {args.model} on {device}")
+    log(f"[bench] building engine: {model} on {device} ({'tp' + str(world_size) if tp_mode else 'tp1'})")
     t_build = time.perf_counter()
     client_kwargs = dict(
-        model=args.model,
+        model=model,
         device=device,
         use_hip_graphs=(on_gpu and not args.no_graphs),
-        seed=rank,
-        default_max_new_tokens=args.max_new,
-        max_seq_len=max(1024, args.prompt_len + args.max_new + 64),
+        seed=0 if tp_mode else rank,
+        default_max_new_tokens=max_new,
+        max_seq_len=max(1024, args.prompt_len + max_new + 64),
     )
+    if tp_mode:
+        client_kwargs["tp_size"] = world_size
     if not on_gpu:
-        client_kwargs["max_kv_blocks"] = 2048
-    k = KLLMs(**client_kwargs)
+        client_kwargs["max_kv_blocks"] = 4096
+
+    follower = None
+    if tp_mode and rank != 0:
+        # non-zero TP ranks run the lockstep follower loop; rank 0 drives the
+        # public client below and broadcasts every engine action.
+        from kllms_amd.engine.api import LocalEngineClient
+        from kllms_amd.parallel.serve import TPFollower
+
+        eng_client = LocalEngineClient(**client_kwargs)
+        follower = TPFollower(eng_client.engine)
+        log(f"[bench] rank {rank}: follower ready in {time.perf_counter() - t_build:.1f}s")
+        follower.run()  # blocks until rank 0 sends stop
+        if dist is not None:
+            dist.barrier()
+            dist.destroy_process_group()
+        return
+
+    k = AsyncKLLMs(**client_kwargs)
     eng_client = k.client
     _ = eng_client.engine  # materialize weights now
+    if tp_mode:
+        from kllms_amd.parallel.serve import TPCoordinator
+
+        eng_client.scheduler.coordinator = TPCoordinator(eng_client.engine)
     log(f"[bench] engine ready in {time.perf_counter() - t_build:.1f}s")
 
-    def embeddings_wrapper(texts):
-        return k.get_embeddings(texts, "text-embedding-3-small", 2048, False)
-
+    schema = make_schema() if mode == "parse" else None
     quality_scores: list = []
+    schema_valid = [0, 0]
 
-    def run_step(step_idx: int) -> None:
-        call_params_list = [
-            {
-                "messages": [{"role": "user", "content": make_prompt(rank, step_idx, i, args.prompt_len)}],
-                "model": args.model,
-                "n": args.n,
-                "temperature": 0.8,
-                "max_tokens": args.max_new,
-                "seed": 10_000 * rank + 100 * step_idx + i,
-            }
-            for i in range(args.batch)
-        ]
-        completions = eng_client.chat_completions_create_many(call_params_list)
-        t_gen = time.perf_counter()
-        for comp in completions:
-            result = consolidate_chat_completions(comp, embeddings_wrapper, client=eng_client)
-            quality_scores.append(_mean_likelihood(result.likelihoods))
-        if os.environ.get("KLLMS_BENCH_VERBOSE"):
-            tm = getattr(eng_client.engine, "last_timings", {})
-            log(f"[bench] step {step_idx}: engine={tm}, consensus={1000 * (time.perf_counter() - t_gen):.1f}ms")
+    async def one_request(step_idx: int, i: int):
+        msgs = [{"role": "user", "content": make_prompt(rank, step_idx, i, args.prompt_len)}]
+        seed = 10_000 * rank + 100 * step_idx + i
+        if mode == "parse":
+            res = await k.chat.completions.parse(
+                messages=msgs, model=model, response_format=schema,
+                n=n, temperature=0.9, max_tokens=max_new, seed=seed,
+            )
+            for ch in res.choices[1:]:
+                schema_valid[1] += 1
+                if ch.message.parsed is not None:
+                    schema_valid[0] += 1
+        else:
+            res = await k.chat.completions.create(
+                messages=msgs, model=model,
+                n=n, temperature=0.8, max_tokens=max_new, seed=seed,
+            )
+        quality_scores.append(_mean_likelihood(res.likelihoods))
+        return res
+
+    async def run_step(step_idx: int):
+        await asyncio.gather(*(one_request(step_idx, i) for i in range(batch)))
 
     def barrier_sync():
-        if dist is not None:
+        if dist is not None and not tp_mode:
             dist.barrier()
         if on_gpu:
             torch.cuda.synchronize()
 
     for w in range(args.warmup):
         log(f"[bench] warmup {w + 1}/{args.warmup}")
-        run_step(-1 - w)
+        asyncio.run(run_step(-1 - w))
 
     barrier_sync()
     t0 = time.perf_counter()
     for s in range(args.steps):
-        run_step(s)
+        asyncio.run(run_step(s))
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
-    # MAX over ranks
-    if dist is not None:
+    # MAX over ranks (DP mode; in TP mode only rank 0 times the job)
+    if dist is not None and not tp_mode:
         t = torch.tensor([elapsed], dtype=torch.float64,
                          device=device if dist.get_backend() == "nccl" else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    n_ranks = world_size
-    total_requests = args.steps * args.batch * n_ranks
+    n_engines = 1 if tp_mode else world_size
+    total_requests = args.steps * batch * n_engines
     value = total_requests / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
     if rank == 0:
+        config = {
+            "model": model,
+            "n": n,
+            "mode": mode,
+            "global_batch": batch * n_engines,
+            "seq_len": args.prompt_len,
+            "max_new_tokens": max_new,
+            "parallelism": f"tp{world_size}" if tp_mode else f"dp{world_size} tp1",
+            "api_path": "AsyncKLLMs.chat.completions." + ("parse" if mode == "parse" else "create"),
+            "completions_per_s": round(value * n, 3),
+            # amortized (throughput) latency and the user-perceived batch
+            # wall time (a sync batch's requests all complete together)
+            "consensus_latency_s_per_request": round(elapsed / (args.steps * batch), 4),
+            "batch_wall_s": round(ms_per_step / 1000.0, 4),
+            "mean_consensus_quality": round(sum(quality_scores) / len(quality_scores), 4) if quality_scores else None,
+        }
+        if mode == "parse":
+            config["schema_valid_fraction"] = round(schema_valid[0] / max(1, schema_valid[1]), 4)
         result = {
             "metric": "end_to_end_consensus_requests_per_s",
             "value": round(value, 4),
             "unit": "req/s",
-            "n_gpus": n_ranks,
+            "n_gpus": world_size,
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": round(ms_per_step, 2),
             "higher_is_better": True,
-            "scaling": "weak",
-            "vs_baseline": round(value / REFERENCE_BASELINE_REQ_S, 3),
+            "scaling": "weak" if not tp_mode else "strong",
+            "vs_baseline": round(value / REFERENCE_BASELINE[args.config], 3),
             "dtype": "bf16" if on_gpu else "float32",
             "data": "synthetic prompts, random-init weights",
-            "config": {
-                "model": args.model,
-                "n": args.n,
-                "global_batch": args.batch * n_ranks,
-                "seq_len": args.prompt_len,
-                "max_new_tokens": args.max_new,
-                "parallelism": f"dp{n_ranks} tp1",
-                "completions_per_s": round(value * args.n, 3),
-                # amortized (throughput) latency and the user-perceived batch
-                # wall time (a sync batch's requests all complete together)
-                "consensus_latency_s_per_request": round(elapsed / (args.steps * args.batch), 4),
-                "batch_wall_s": round(ms_per_step / 1000.0, 4),
-                "mean_consensus_quality": round(sum(quality_scores) / len(quality_scores), 4) if quality_scores else None,
-            },
+            "config": config,
         }
         print(json.dumps(result), flush=True)
 
+    if tp_mode:
+        sched = eng_client.scheduler
+        coord = sched.coordinator
+        sched.shutdown()      # join the worker thread first …
+        coord.stop()          # … then release the followers
     if dist is not None:
+        dist.barrier()
         dist.destroy_process_group()
 
 

@@ -364,6 +364,7 @@ class JsonSchemaConstraint:
 
     def __init__(self, schema: Dict[str, Any], tokenizer, whitespace: bool = False):
         self.schema = schema
+        self.whitespace = whitespace
         tok_key = getattr(tokenizer, "cache_key", None) or (id(type(tokenizer)), tokenizer.vocab_size)
         key = (json.dumps(schema, sort_keys=True), tok_key, whitespace)
         cached = _TABLE_CACHE.get(key)  # type: ignore[arg-type]

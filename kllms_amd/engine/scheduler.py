@@ -67,6 +67,9 @@ class BatchScheduler:
         # serializes engine access against direct (non-scheduled) callers
         self.engine_lock = engine_lock or threading.Lock()
         self._queue: "queue.Queue[_Ticket]" = queue.Queue()
+        # TP>1 serving: rank 0 attaches a parallel.serve.TPCoordinator that
+        # broadcasts each engine-touching phase to the follower ranks
+        self.coordinator = None
         self._thread: Optional[threading.Thread] = None
         self._lock = threading.Lock()
         self._stop = threading.Event()
@@ -122,12 +125,19 @@ class BatchScheduler:
                 # pending chunked prefills reserve their future stream slots
                 reserved = sum(max(1, p.ticket.request.n) for p in ctx.pending)
                 tickets = self._drain(len(ctx.active) + reserved, block=idle)
+                coord = self.coordinator
                 with self.engine_lock:
                     if tickets:
+                        if coord is not None:
+                            coord.admit([t.request for t in tickets])
                         self._admit(ctx, tickets)
                     if ctx.pending:
+                        if coord is not None:
+                            coord.advance_prefill()
                         self._advance_prefill(ctx)
                     if ctx.active:
+                        if coord is not None:
+                            coord.step()
                         self._step(ctx)
 
     def _admit(self, ctx: "_WorkerContext", tickets: List[_Ticket]) -> None:
