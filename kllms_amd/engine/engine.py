@@ -138,6 +138,14 @@ class LLMEngine:
         else:
             self.ctx = ParallelContext()
 
+        # size-switched custom collective: one-shot xGMI all-reduce for small
+        # (decode-step) tensors, RCCL ring above threshold
+        if self.ctx.world_size > 1 and self.ctx.custom_ar is None:
+            from ..parallel.collective import maybe_init_custom_allreduce
+
+            dev_for_ar = self.device if self.device.type == "cuda" else torch.device("cpu")
+            maybe_init_custom_allreduce(self.ctx, dev_for_ar)
+
         # monotonic per-request counter for unseeded-request RNG derivation
         self._seed_counter = 0
         self.dtype = {"bfloat16": torch.bfloat16, "float16": torch.float16, "float32": torch.float32}[config.dtype]

@@ -1,24 +1,224 @@
-// One-shot all-reduce kernel for small (latency-bound) decode-step tensors.
+// One-shot all-reduce for small (latency-bound) decode-step tensors, over
+// hipIpc-mapped peer buffers on the xGMI mesh.
 //
-// Round-2 groundwork for SURVEY §5.8 / §2.3: on an 8-GPU MI355X node the
-// xGMI fabric is fully connected (7 p2p links per GPU), so for the tens-of-KB
-// tensors a TP decode step reduces ([n_streams, hidden] bf16), the
-// latency-optimal schedule is ONE SHOT: every rank reads all peers' buffers
-// directly over xGMI and writes the sum — no ring hops, no chunk pipeline.
+// on an 8-GPU MI355X node the xGMI fabric is fully
+// connected (7 p2p links per GPU), so for the tens-of-KB tensors a TP decode
+// step reduces ([n_streams, hidden] bf16) the latency-optimal schedule is
+// ONE SHOT: every rank reads all peers' staged buffers directly over xGMI
+// and writes its own output — no ring hops, no chunk pipeline. Ring (RCCL)
+// takes over above a size threshold (python: parallel/collective.py).
 //
-// This kernel is the device side of that design and is testable on a single
-// GPU: the peer pointers are just N buffers (same device here; IPC-mapped
-// peer memory in the multi-GPU wiring, which needs an 8-GPU box to
-// validate — see docs/ROADMAP.md item 1). Reduction is fp32-accumulated
-// bf16, 8-element (16 B) vectorized per lane, grid-strided.
+// Synchronization (two phases, per-block epoch flags in UNCACHED memory):
+//   1. every rank: release-fence (system scope) -> store epoch into each
+//      peer's start[block][me] -> spin on OWN start[block][r] for all r
+//      (poll local memory, relaxed; one acquire fence after) — guarantees
+//      every rank's staged data is globally visible before anyone reads it.
+//   2. after the reduce: same handshake on end[block][*] — guarantees no
+//      rank re-stages its buffer while a peer is still reading it.
+// Epochs come from a DEVICE-resident per-block counter, so the kernel is
+// hipGraph-replayable (no per-launch salt argument — frozen under replay;
+// cdna_hip_programming.md §6 G16 "Re-initialise every call").
+// Spins are BOUNDED: a dead peer trips s_trap instead of hanging the GPU.
 //
-// Reference parity note: the reference has no collectives at all (its
-// backend is a remote API); this replaces nothing and accelerates the native
-// TP path only.
+// Reference parity note: the reference has no collectives (its backend is a
+// remote API); this accelerates the native TP path only.
 
 #include "common.h"
 
+#include <cstdio>
+#include <cstring>
+
 #define AR_MAX_PEERS 8
+#define AR_MAX_BLOCKS 64
+#define AR_SPIN_LIMIT 100000000u  // ~ seconds at s_sleep(2) per spin
+
+struct ArSignals {
+  unsigned start[AR_MAX_BLOCKS][AR_MAX_PEERS];
+  unsigned end[AR_MAX_BLOCKS][AR_MAX_PEERS];
+  unsigned epoch[AR_MAX_BLOCKS];  // only the owning rank touches
+};
+
+struct ArPeers {
+  const bf16_t* data[AR_MAX_PEERS];  // every rank's staging buffer
+  ArSignals* sig[AR_MAX_PEERS];      // every rank's signal page (uncached)
+};
+
+__device__ __forceinline__ void ar_spin_eq(volatile unsigned* w, unsigned e) {
+  unsigned spins = 0;
+  while (__hip_atomic_load((unsigned*)w, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM) != e) {
+    __builtin_amdgcn_s_sleep(2);
+    if (++spins > AR_SPIN_LIMIT) __builtin_trap();  // dead peer: abort, don't hang the box
+  }
+}
+
+__global__ void ipc_allreduce_kernel(ArPeers p, bf16_t* __restrict__ dst, int me,
+                                     int world, long n8) {
+  const int b = blockIdx.x;
+  __shared__ unsigned s_epoch;
+  if (threadIdx.x == 0) {
+    const unsigned e = p.sig[me]->epoch[b] + 1;
+    p.sig[me]->epoch[b] = e;
+    s_epoch = e;
+  }
+  __syncthreads();
+  const unsigned e = s_epoch;
+
+  // ---- phase 1: publish "my staged data is ready", wait for all peers ----
+  if (threadIdx.x == 0) {
+    // system-scope release: staged bytes (written by the stream-ordered copy
+    // before this kernel) leave L2 before any peer can see the flag. The asm
+    // wait restates the post-wbl2 vmcnt the compiler may drop (§6 G16 p12).
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __syncthreads();
+  if (threadIdx.x < (unsigned)world) {
+    const int r = threadIdx.x;
+    __hip_atomic_store(&p.sig[r]->start[b][me], e, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+    ar_spin_eq(&p.sig[me]->start[b][r], e);
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+  __syncthreads();
+
+  // ---- reduce: grid-stride over 16-B groups, fp32 accumulate ----
+  const long i = (long)b * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long g = i; g < n8; g += stride) {
+    float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    for (int r = 0; r < world; ++r) {
+      bf16x8_vec v = *reinterpret_cast<const bf16x8_vec*>(p.data[r] + g * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += bf16_to_f32(v[j]);
+    }
+    bf16x8_vec out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = f32_to_bf16(acc[j]);
+    *reinterpret_cast<bf16x8_vec*>(dst + g * 8) = out;
+  }
+
+  // ---- phase 2: done reading peers; nobody may re-stage until all done ----
+  __syncthreads();
+  if (threadIdx.x < (unsigned)world) {
+    const int r = threadIdx.x;
+    __hip_atomic_store(&p.sig[r]->end[b][me], e, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+    ar_spin_eq(&p.sig[me]->end[b][r], e);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host-side IPC context
+// ---------------------------------------------------------------------------
+
+struct IpcArCtx {
+  int rank = 0;
+  int world = 1;
+  size_t max_bytes = 0;
+  void* data[AR_MAX_PEERS] = {};
+  ArSignals* sig[AR_MAX_PEERS] = {};
+  bool opened_data[AR_MAX_PEERS] = {};
+  bool opened_sig[AR_MAX_PEERS] = {};
+};
+
+#define AR_HIP_CHECK(expr)                                                     \
+  do {                                                                         \
+    hipError_t e_ = (expr);                                                    \
+    if (e_ != hipSuccess) {                                                    \
+      fprintf(stderr, "ipc_allreduce: %s failed: %s\n", #expr,                 \
+              hipGetErrorString(e_));                                          \
+      return nullptr;                                                          \
+    }                                                                          \
+  } while (0)
+
+#define AR_HIP_CHECK_INT(expr)                                                 \
+  do {                                                                         \
+    hipError_t e_ = (expr);                                                    \
+    if (e_ != hipSuccess) {                                                    \
+      fprintf(stderr, "ipc_allreduce: %s failed: %s\n", #expr,                 \
+              hipGetErrorString(e_));                                          \
+      return -1;                                                               \
+    }                                                                          \
+  } while (0)
+
+// Creates the local staging + signal allocations and writes their IPC
+// handles (2 x 64 bytes) into handles_out. Returns an opaque ctx pointer.
+extern "C" void* ipc_ar_create(int rank, int world, size_t max_bytes,
+                               unsigned char* handles_out) {
+  if (world < 1 || world > AR_MAX_PEERS) return nullptr;
+  IpcArCtx* ctx = new IpcArCtx();
+  ctx->rank = rank;
+  ctx->world = world;
+  ctx->max_bytes = max_bytes;
+  AR_HIP_CHECK(hipMalloc(&ctx->data[rank], max_bytes));
+  // signal flags live in UNCACHED (fine-grained) memory so cross-device
+  // polling sees stores without cache games
+  void* sigmem = nullptr;
+  AR_HIP_CHECK(hipExtMallocWithFlags(&sigmem, sizeof(ArSignals), hipDeviceMallocUncached));
+  AR_HIP_CHECK(hipMemset(sigmem, 0, sizeof(ArSignals)));
+  ctx->sig[rank] = reinterpret_cast<ArSignals*>(sigmem);
+  hipIpcMemHandle_t hd, hs;
+  AR_HIP_CHECK(hipIpcGetMemHandle(&hd, ctx->data[rank]));
+  AR_HIP_CHECK(hipIpcGetMemHandle(&hs, sigmem));
+  std::memcpy(handles_out, &hd, sizeof(hd));
+  std::memcpy(handles_out + sizeof(hd), &hs, sizeof(hs));
+  return ctx;
+}
+
+// all_handles: world * 128 bytes (each rank's [data_handle | sig_handle]).
+extern "C" int ipc_ar_connect(void* ctx_, const unsigned char* all_handles) {
+  IpcArCtx* ctx = reinterpret_cast<IpcArCtx*>(ctx_);
+  for (int r = 0; r < ctx->world; ++r) {
+    if (r == ctx->rank) continue;
+    hipIpcMemHandle_t hd, hs;
+    std::memcpy(&hd, all_handles + r * 128, sizeof(hd));
+    std::memcpy(&hs, all_handles + r * 128 + 64, sizeof(hs));
+    void* pd = nullptr;
+    void* ps = nullptr;
+    AR_HIP_CHECK_INT(hipIpcOpenMemHandle(&pd, hd, hipIpcMemLazyEnablePeerAccess));
+    AR_HIP_CHECK_INT(hipIpcOpenMemHandle(&ps, hs, hipIpcMemLazyEnablePeerAccess));
+    ctx->data[r] = pd;
+    ctx->sig[r] = reinterpret_cast<ArSignals*>(ps);
+    ctx->opened_data[r] = ctx->opened_sig[r] = true;
+  }
+  return 0;
+}
+
+extern "C" int ipc_ar_run(void* ctx_, const void* inp, void* out, long numel,
+                          hipStream_t stream) {
+  IpcArCtx* ctx = reinterpret_cast<IpcArCtx*>(ctx_);
+  const size_t bytes = (size_t)numel * sizeof(bf16_t);
+  if (bytes > ctx->max_bytes || numel % 8 != 0) return -2;
+  AR_HIP_CHECK_INT(hipMemcpyAsync(ctx->data[ctx->rank], inp, bytes,
+                                  hipMemcpyDeviceToDevice, stream));
+  ArPeers p;
+  for (int r = 0; r < AR_MAX_PEERS; ++r) {
+    p.data[r] = reinterpret_cast<const bf16_t*>(ctx->data[r < ctx->world ? r : 0]);
+    p.sig[r] = ctx->sig[r < ctx->world ? r : 0];
+  }
+  const long n8 = numel / 8;
+  int blocks = (int)((n8 + 255) / 256);
+  if (blocks > AR_MAX_BLOCKS) blocks = AR_MAX_BLOCKS;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(ipc_allreduce_kernel, dim3(blocks), dim3(256), 0, stream,
+                     p, reinterpret_cast<bf16_t*>(out), ctx->rank, ctx->world, n8);
+  return 0;
+}
+
+extern "C" void ipc_ar_destroy(void* ctx_) {
+  IpcArCtx* ctx = reinterpret_cast<IpcArCtx*>(ctx_);
+  if (ctx == nullptr) return;
+  for (int r = 0; r < ctx->world; ++r) {
+    if (ctx->opened_data[r]) (void)hipIpcCloseMemHandle(ctx->data[r]);
+    if (ctx->opened_sig[r]) (void)hipIpcCloseMemHandle(ctx->sig[r]);
+  }
+  (void)hipFree(ctx->data[ctx->rank]);
+  (void)hipFree(ctx->sig[ctx->rank]);
+  delete ctx;
+}
+
+// ---------------------------------------------------------------------------
+// Single-GPU simulation kernel (kept: numerics tests + rank-0-writes-all)
+// ---------------------------------------------------------------------------
 
 struct PeerPtrs {
   const bf16_t* src[AR_MAX_PEERS];
@@ -36,14 +236,13 @@ __global__ void one_shot_allreduce_kernel(PeerPtrs p, int n_peers, long n8,
       bf16x8_vec v = *reinterpret_cast<const bf16x8_vec*>(p.src[r] + g * 8);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        bf16_t b = __builtin_bit_cast(bf16_t, (short)v[j]);
-        acc[j] += __bfloat162float(b);
+        acc[j] += bf16_to_f32(v[j]);
       }
     }
     bf16x8_vec out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      out[j] = __builtin_bit_cast(short, __float2bfloat16(acc[j]));
+      out[j] = f32_to_bf16(acc[j]);
     }
     if (write_all) {
       // single-GPU simulation / rank-0-writes-peers variant

@@ -198,7 +198,45 @@ void one_shot_allreduce(std::vector<torch::Tensor> bufs) {
                             cur_stream());
 }
 
+// --- hipIpc one-shot all-reduce (allreduce.hip) ------------------------------
+extern "C" void* ipc_ar_create(int rank, int world, size_t max_bytes, unsigned char* handles_out);
+extern "C" int ipc_ar_connect(void* ctx, const unsigned char* all_handles);
+extern "C" int ipc_ar_run(void* ctx, const void* inp, void* out, long numel, hipStream_t stream);
+extern "C" void ipc_ar_destroy(void* ctx);
+
+std::pair<int64_t, py::bytes> ipc_allreduce_create(int rank, int world, int64_t max_bytes) {
+  unsigned char handles[128];
+  void* ctx = ipc_ar_create(rank, world, (size_t)max_bytes, handles);
+  TORCH_CHECK(ctx != nullptr, "ipc_ar_create failed (see stderr)");
+  return {reinterpret_cast<int64_t>(ctx), py::bytes(reinterpret_cast<char*>(handles), 128)};
+}
+
+void ipc_allreduce_connect(int64_t ctx, py::bytes all_handles) {
+  std::string h = all_handles;
+  TORCH_CHECK(h.size() % 128 == 0, "handle blob must be world*128 bytes");
+  int rc = ipc_ar_connect(reinterpret_cast<void*>(ctx),
+                          reinterpret_cast<const unsigned char*>(h.data()));
+  TORCH_CHECK(rc == 0, "ipc_ar_connect failed (see stderr)");
+}
+
+void ipc_allreduce_run(int64_t ctx, torch::Tensor inp, torch::Tensor out) {
+  CHECK_BF16_CONTIG(inp);
+  CHECK_BF16_CONTIG(out);
+  TORCH_CHECK(inp.numel() == out.numel(), "in/out size mismatch");
+  int rc = ipc_ar_run(reinterpret_cast<void*>(ctx), inp.data_ptr(), out.data_ptr(),
+                      (long)inp.numel(), cur_stream());
+  TORCH_CHECK(rc == 0, "ipc_ar_run failed rc=", rc);
+}
+
+void ipc_allreduce_destroy(int64_t ctx) {
+  ipc_ar_destroy(reinterpret_cast<void*>(ctx));
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("ipc_allreduce_create", &ipc_allreduce_create);
+  m.def("ipc_allreduce_connect", &ipc_allreduce_connect);
+  m.def("ipc_allreduce_run", &ipc_allreduce_run);
+  m.def("ipc_allreduce_destroy", &ipc_allreduce_destroy);
   m.def("rmsnorm", &rmsnorm);
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
   m.def("silu_mul", &silu_mul);

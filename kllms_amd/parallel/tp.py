@@ -31,6 +31,8 @@ class ParallelContext:
     world_size: int = 1
     rank: int = 0
     group: Optional[object] = None  # dist.ProcessGroup
+    # size-switched custom collective (parallel/collective.py); None = RCCL only
+    custom_ar: Optional[object] = None
 
     @classmethod
     def from_env_or_single(cls) -> "ParallelContext":
@@ -40,6 +42,9 @@ class ParallelContext:
 
     def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
         if self.world_size > 1:
+            ca = self.custom_ar
+            if ca is not None and ca.should_use(t):
+                return ca.all_reduce_(t)
             dist.all_reduce(t, group=self.group)
         return t
 

@@ -1,0 +1,171 @@
+"""hipIpc one-shot all-reduce on real hardware (VERDICT round-1 item 1).
+
+A gpurun box has ONE MI355X, so multi-rank validation runs as 2 processes
+SHARING cuda:0: hipIpc handle exchange, the two-phase epoch handshake and the
+reduce kernel are identical to the 8-GPU case — only the transport under the
+peer reads differs (local HBM here, xGMI links on a node). gloo carries the
+handle exchange; RCCL is not used (two ranks cannot share one device under
+NCCL semantics).
+
+Covers: exact sums vs a host reference, repeated calls (device-resident
+epoch counters), hipGraph capture+replay of the collective, and a full TP=2
+engine forward (custom AR under the TP layers) matching the TP=1 engine's
+greedy decode.
+"""
+
+import multiprocessing as mp
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+PORT = 29821
+
+
+def _ar_worker(rank: int, world: int, q):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(PORT)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.cuda.set_device(0)
+        from kllms_amd.parallel.collective import CustomAllReduce
+
+        car = CustomAllReduce(rank, world, torch.device("cuda:0"))
+
+        results = {}
+        # --- exact sums across sizes, repeated (epochs advance) -------------
+        for trial in range(3):
+            for numel in (8, 4096, 5 * 4096, 65536, 262144):
+                g = torch.Generator(device="cpu").manual_seed(1000 + trial * 10 + numel % 7)
+                local_cpu = [torch.randn(numel, generator=g) for _ in range(world)]
+                want = sum(x.float() for x in (t.bfloat16() for t in local_cpu))
+                t = local_cpu[rank].bfloat16().cuda()
+                car.all_reduce_(t)
+                torch.cuda.synchronize()
+                got = t.float().cpu()
+                err = (got - want.bfloat16().float()).abs().max().item()
+                results[f"sum_{trial}_{numel}"] = err
+        # --- hipGraph capture + replay (decode-step usage) -------------------
+        buf = torch.zeros(4096, dtype=torch.bfloat16, device="cuda")
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            car.all_reduce_(buf)  # warmup on the side stream
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        dist.barrier()  # both ranks warm before capture
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            car.all_reduce_(buf)
+        dist.barrier()  # both ranks captured before first replay
+        for trial in range(3):
+            fill = float(rank + 1) * (trial + 1)
+            buf.fill_(fill)
+            graph.replay()
+            torch.cuda.synchronize()
+            dist.barrier()  # replays are collective: both ranks per trial
+            want = sum(float(r + 1) * (trial + 1) for r in range(world))
+            err = (buf.float() - want).abs().max().item()
+            results[f"graph_{trial}"] = err
+        car.close()
+        q.put((rank, results))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_ipc_allreduce_two_ranks_one_gpu():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_ar_worker, args=(r, 2, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, res = q.get(timeout=480)
+        results[rank] = res
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    for rank in (0, 1):
+        for k, err in results[rank].items():
+            # bf16 sum of 2 bf16 values, fp32-accumulated: one rounding step
+            assert err <= 0.05, (rank, k, err)
+    # graph path must be exact (integer fills)
+    for rank in (0, 1):
+        for k, err in results[rank].items():
+            if k.startswith("graph"):
+                assert err == 0.0, (rank, k, err)
+
+
+def _tp_engine_worker(rank: int, world: int, q):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(PORT + 1)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.cuda.set_device(0)
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.engine import GenRequest, LLMEngine
+        from kllms_amd.engine.sampling import SamplingParams
+        from kllms_amd.parallel.tp import ParallelContext
+
+        ctx = ParallelContext(world_size=world, rank=rank)
+        # both ranks on cuda:0; the TP layers' gloo all-reduce is replaced by
+        # the custom IPC path for small tensors, which is what decode uses
+        from kllms_amd.parallel.collective import maybe_init_custom_allreduce
+
+        maybe_init_custom_allreduce(ctx, torch.device("cuda:0"))
+        assert ctx.custom_ar is not None, "custom AR failed to initialize"
+
+        eng = LLMEngine(
+            EngineConfig(model="mid-llama", tp_size=world, max_kv_blocks=512,
+                         use_hip_graphs=True, device="cuda:0", seed=0,
+                         default_max_new_tokens=12),
+            parallel_ctx=ctx,
+        )
+        ids = list(range(1, 40))
+        out = eng.generate([GenRequest(prompt_ids=ids, n=2,
+                                       sampling=SamplingParams(temperature=0.0, max_tokens=12))])[0]
+        calls = ctx.custom_ar.calls
+        ctx.custom_ar.close()
+        q.put((rank, out.streams[0].token_ids, calls))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_tp2_engine_one_gpu_custom_ar_matches_tp1():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tp_engine_worker, args=(r, 2, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, tokens, calls = q.get(timeout=480)
+        results[rank] = (tokens, calls)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+
+    assert results[0][0] == results[1][0], "TP ranks diverged"
+    assert results[0][1] > 0, "custom all-reduce was never used"
+
+    # TP=1 single-process reference on the same GPU
+    from kllms_amd.engine.config import EngineConfig
+    from kllms_amd.engine.engine import GenRequest, LLMEngine
+    from kllms_amd.engine.sampling import SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="mid-llama", tp_size=1, max_kv_blocks=512,
+                                 use_hip_graphs=True, device="cuda:0", seed=0,
+                                 default_max_new_tokens=12))
+    out = eng.generate([GenRequest(prompt_ids=list(range(1, 40)), n=2,
+                                   sampling=SamplingParams(temperature=0.0, max_tokens=12))])[0]
+    assert results[0][0] == out.streams[0].token_ids, (
+        "TP=2 (custom one-shot AR) greedy decode != TP=1")
