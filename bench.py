@@ -266,16 +266,19 @@ def main():
         if on_gpu:
             torch.cuda.synchronize()
 
-    for w in range(args.warmup):
-        log(f"[bench] warmup {w + 1}/{args.warmup}")
-        asyncio.run(run_step(-1 - w))
+    async def timed_run():
+        # ONE event loop (and one to_thread pool) for the whole bench
+        for w in range(args.warmup):
+            log(f"[bench] warmup {w + 1}/{args.warmup}")
+            await run_step(-1 - w)
+        barrier_sync()
+        t0 = time.perf_counter()
+        for s in range(args.steps):
+            await run_step(s)
+        barrier_sync()
+        return time.perf_counter() - t0
 
-    barrier_sync()
-    t0 = time.perf_counter()
-    for s in range(args.steps):
-        asyncio.run(run_step(s))
-    barrier_sync()
-    elapsed = time.perf_counter() - t0
+    elapsed = asyncio.run(timed_run())
 
     # MAX over ranks (DP mode; in TP mode only rank 0 times the job)
     if dist is not None and not tp_mode:
