@@ -8,7 +8,12 @@
 //           (per-wave copies kill the LDS same-bin atomic serialization that
 //           made a single shared histogram ~50x slower)
 //   pass C: admission threshold from the histogram suffix (top-p by mass,
-//           top-k by count; exact up to one bin width)
+//           top-k by count), then up to 3 REFINEMENT sweeps that re-histogram
+//           only the boundary bin at 256x finer resolution each time —
+//           threshold resolution 20/256^4 ~ 5e-9 in scaled-logit space, i.e.
+//           exact top-k/top-p up to fp32 rounding of the logits themselves
+//           (the old single-pass cut admitted everything
+//           within one 0.078-wide bin of the true boundary)
 //   pass D: Gumbel-max draw over admitted tokens — counter-based RNG
 //           hash(seed, step, token): deterministic, stream-ordered, and
 //           identical regardless of batch composition.
@@ -122,22 +127,96 @@ extern "C" __global__ void __launch_bounds__(NT) sample_kernel(
       hist_cnt[tid] = c;
     }
     __syncthreads();
+    // boundary-interval state (sx space: 0 at the max, negative below);
+    // need_* is the admission still OWED by the boundary interval after
+    // everything strictly above it is admitted
+    __shared__ float sh_lo, sh_hi, sh_need_mass;
+    __shared__ unsigned int sh_need_cnt;
+    __shared__ int sh_state;  // 0 = threshold at sh_lo, 1 = refine, 2 = admit all
     if (tid == 0) {
       const float need_mass = top_p < 1.0f ? top_p * gsum_scaled : INFINITY;
-      const int need_cnt = (top_k > 0 && top_k < V) ? top_k : 0x7fffffff;
+      const unsigned int need_cnt =
+          (top_k > 0 && top_k < V) ? (unsigned)top_k : 0xffffffffu;
       float acc_mass = 0.0f;
       unsigned int acc_cnt = 0;
-      int cut_bin = SBINS - 1;
+      int cut_bin = -1;
       for (int bn = 0; bn < SBINS; ++bn) {
         acc_mass += hist_mass[bn];
         acc_cnt += hist_cnt[bn];
-        if (acc_mass >= need_mass || acc_cnt >= (unsigned)need_cnt) { cut_bin = bn; break; }
+        if (acc_mass >= need_mass || acc_cnt >= need_cnt) { cut_bin = bn; break; }
       }
-      // admit tokens whose scaled logit is within (cut_bin+1) bins of the max
-      sh_thresh = (smax - (cut_bin + 1) * (SRANGE / SBINS)) * temp;
+      if (cut_bin < 0) {
+        sh_state = 2;  // window never met the need: admit everything
+      } else {
+        const float binw = SRANGE / SBINS;
+        const float before_mass = acc_mass - hist_mass[cut_bin];
+        const unsigned int before_cnt = acc_cnt - hist_cnt[cut_bin];
+        sh_lo = -(cut_bin + 1) * binw;
+        sh_hi = -cut_bin * binw;
+        sh_need_mass = need_mass - before_mass;
+        sh_need_cnt = (need_cnt == 0xffffffffu) ? 0xffffffffu : need_cnt - before_cnt;
+        // the whole bin is exactly owed -> no refinement needed
+        sh_state = (hist_cnt[cut_bin] == sh_need_cnt) ? 0 : 1;
+      }
     }
     __syncthreads();
-    x_thresh = sh_thresh;
+
+    // refinement: re-histogram ONLY the boundary interval at 256x finer
+    // resolution; each sweep divides the threshold uncertainty by SBINS
+    for (int it = 0; it < 3; ++it) {
+      if (sh_state != 1) break;
+      const float lo = sh_lo, hi = sh_hi;
+      const float inv_w = SBINS / (hi - lo);
+      for (int i = tid; i < NW * SBINS; i += NT) {
+        hist_mass[i] = 0.0f;
+        hist_cnt[i] = 0;
+      }
+      __syncthreads();
+      for (int i = tid; i < V; i += NT) {
+        if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
+        const float sx = row[i] * inv_t - smax;
+        if (sx < lo || sx >= hi) continue;
+        int bn = (int)((hi - sx) * inv_w);
+        bn = min(max(bn, 0), SBINS - 1);
+        atomicAdd(&my_mass[bn], __expf(sx));
+        atomicAdd(&my_cnt[bn], 1u);
+      }
+      __syncthreads();
+      if (tid < SBINS) {
+        float m = 0.0f;
+        unsigned int c = 0;
+#pragma unroll 4
+        for (int w = 0; w < NW; ++w) {
+          m += hist_mass[w * SBINS + tid];
+          c += hist_cnt[w * SBINS + tid];
+        }
+        hist_mass[tid] = m;
+        hist_cnt[tid] = c;
+      }
+      __syncthreads();
+      if (tid == 0) {
+        float acc_mass = 0.0f;
+        unsigned int acc_cnt = 0;
+        int cut_bin = SBINS - 1;  // need is guaranteed met inside [lo, hi)
+        for (int bn = 0; bn < SBINS; ++bn) {
+          acc_mass += hist_mass[bn];
+          acc_cnt += hist_cnt[bn];
+          if (acc_mass >= sh_need_mass || acc_cnt >= sh_need_cnt) { cut_bin = bn; break; }
+        }
+        const float subw = (hi - lo) / SBINS;
+        sh_need_mass -= acc_mass - hist_mass[cut_bin];
+        if (sh_need_cnt != 0xffffffffu) sh_need_cnt -= acc_cnt - hist_cnt[cut_bin];
+        sh_lo = hi - (cut_bin + 1) * subw;
+        sh_hi = hi - cut_bin * subw;
+        sh_state = (hist_cnt[cut_bin] == sh_need_cnt) ? 0 : 1;
+      }
+      __syncthreads();
+    }
+    if (sh_state != 2) {
+      if (tid == 0) sh_thresh = (smax + sh_lo) * temp;
+      __syncthreads();
+      x_thresh = sh_thresh;
+    }
   }
 
   // ---- pass D: Gumbel-max over admitted tokens ----------------------------

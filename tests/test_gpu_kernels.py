@@ -178,14 +178,53 @@ class TestSampling:
         assert all(int(x) in allowed for x in toks.cpu())
 
     def test_top_k_confines(self):
+        # EXACT top-k (refined histogram cut): the drawn token must be a
+        # member of the true top-k, no bin-boundary slack
         B, V = 16, 8192
         logits, t, tp, tk, seeds, steps = self._tensors(B, V, top_k=5)
         toks, _ = ops.sample(logits, t, tp, tk, seeds, steps)
         for b in range(B):
-            # histogram truncation is bin-approximate: chosen token must be
-            # within the top-k-ish set (allow 4x slack for bin boundaries)
-            topk = set(logits[b].topk(20).indices.cpu().tolist())
+            topk = set(logits[b].topk(5).indices.cpu().tolist())
             assert int(toks[b]) in topk
+
+    def test_top_k_exact_with_crowded_boundary(self):
+        # many near-identical logits straddling the k-th value: the refinement
+        # sweeps must separate them (the coarse 0.078-wide bin cannot)
+        B, V, K = 8, 8192, 7
+        g = torch.Generator(device="cpu").manual_seed(5)
+        base = torch.randn(B, V, generator=g) * 0.01  # everything within ~1 bin
+        logits = base.to(DEV)
+        t = torch.full((B,), 0.9, device=DEV)
+        tp = torch.ones(B, device=DEV)
+        tk = torch.full((B,), K, dtype=torch.int32, device=DEV)
+        seeds = torch.arange(B, dtype=torch.int64, device=DEV) + 3
+        for s in range(6):
+            steps = torch.full((B,), s, dtype=torch.int64, device=DEV)
+            toks, _ = ops.sample(logits, t, tp, tk, seeds, steps)
+            for b in range(B):
+                topk = set(logits[b].topk(K).indices.cpu().tolist())
+                assert int(toks[b]) in topk, (b, s, int(toks[b]))
+
+    def test_top_p_exact_nucleus(self):
+        # nucleus membership: drawn token must be inside the smallest
+        # prefix (by descending prob) whose mass reaches top_p
+        B, V, P = 8, 8192, 0.7
+        g = torch.Generator(device="cpu").manual_seed(9)
+        logits = (torch.randn(B, V, generator=g) * 2.0).to(DEV)
+        t = torch.ones(B, device=DEV)
+        tp = torch.full((B,), P, device=DEV)
+        tk = torch.zeros(B, dtype=torch.int32, device=DEV)
+        seeds = torch.arange(B, dtype=torch.int64, device=DEV) + 17
+        probs = torch.softmax(logits.float(), dim=-1)
+        sp, si = probs.sort(dim=-1, descending=True)
+        cum = sp.cumsum(-1)
+        for s in range(6):
+            steps = torch.full((B,), s, dtype=torch.int64, device=DEV)
+            toks, _ = ops.sample(logits, t, tp, tk, seeds, steps)
+            for b in range(B):
+                ncut = int((cum[b] < P).sum().item()) + 1  # smallest prefix >= P
+                nucleus = set(si[b, :ncut].cpu().tolist())
+                assert int(toks[b]) in nucleus, (b, s, int(toks[b]), ncut)
 
     def test_distribution_tracks_softmax(self):
         # one peaked row sampled across many steps: empirical freq of the top
