@@ -62,6 +62,8 @@ class MixtralMoE(nn.Module):
 
         if T <= 8 * self.E:
             out = self._forward_dense(x, topw, topi)
+        elif x.is_cuda:
+            out = self._forward_grouped_hip(x, topw, topi)
         else:
             out = self._forward_grouped(x, topw, topi)
         return self.ctx.all_reduce(out)
@@ -81,6 +83,54 @@ class MixtralMoE(nn.Module):
         w = torch.zeros(self.E, T, device=x.device, dtype=torch.float32)
         w.scatter_(0, topi.t().long(), topw.t().float())
         return (ye.float() * w.unsqueeze(-1)).sum(dim=0).to(x.dtype)
+
+    def _forward_grouped_hip(self, x: torch.Tensor, topw: torch.Tensor, topi: torch.Tensor) -> torch.Tensor:
+        """Prefill-shape path on GPU: TWO grouped MFMA kernel launches
+        (ops/hip/moe.hip) over tokens sorted by expert — gate/up GEMM with
+        fused silu*mul, then the down GEMM — plus one index_add to scatter
+        the routing-weighted rows back to token order. Replaces the
+        per-expert loop (<=2 GEMM launches per layer)."""
+        from .. import ops
+
+        BM = 128
+        T, H = x.shape
+        dev = x.device
+        flat_exp = topi.reshape(-1).to(torch.int32)               # [T*k]
+        flat_tok = torch.arange(T, device=dev, dtype=torch.int32).repeat_interleave(self.k)
+        flat_w = topw.reshape(-1)
+        order = torch.argsort(flat_exp, stable=True)
+        sorted_exp = flat_exp[order].long()
+        sorted_tok = flat_tok[order]
+        sorted_w = flat_w[order]
+
+        counts = torch.bincount(sorted_exp, minlength=self.E)
+        padded = (counts + BM - 1) // BM * BM                      # 0 stays 0
+        pad_off = torch.zeros(self.E + 1, device=dev, dtype=torch.long)
+        torch.cumsum(padded, 0, out=pad_off[1:])
+        starts = torch.zeros(self.E, device=dev, dtype=torch.long)
+        torch.cumsum(counts[:-1], 0, out=starts[1:])
+        N = sorted_exp.numel()
+        dest = (pad_off[sorted_exp] + torch.arange(N, device=dev) - starts[sorted_exp])
+
+        S = int(pad_off[-1].item())                                # prefill-only host sync
+        sorted_ids = torch.full((S,), -1, device=dev, dtype=torch.int32)
+        sorted_ids[dest] = sorted_tok
+        w_padded = torch.zeros(S, device=dev, dtype=flat_w.dtype)
+        w_padded[dest] = sorted_w
+
+        if getattr(self, "_zeros_page", None) is None or self._zeros_page.numel() < H:
+            self._zeros_page = torch.zeros(H, device=dev, dtype=x.dtype)
+        act = torch.empty(S, self.I, device=dev, dtype=x.dtype)
+        y = torch.empty(S, H, device=dev, dtype=x.dtype)
+        pad_off32 = pad_off.to(torch.int32)
+        ops.moe_gateup(act, x, self.w_gate_up, sorted_ids, pad_off32, self._zeros_page)
+        ops.moe_down(y, act, self.w_down, pad_off32)
+
+        valid = sorted_ids >= 0
+        out = torch.zeros(T, H, device=dev, dtype=torch.float32)
+        out.index_add_(0, sorted_ids[valid].long(),
+                       y[valid].float() * w_padded[valid].unsqueeze(1).float())
+        return out.to(x.dtype)
 
     def _forward_grouped(self, x: torch.Tensor, topw: torch.Tensor, topi: torch.Tensor) -> torch.Tensor:
         """Prefill-shape path: tokens bucketed per expert, one GEMM group per

@@ -159,3 +159,27 @@ def sample(
         )
         return tokens, logprobs
     return torch_ref.sample(logits, temperatures, top_ps, top_ks, seeds, steps, mask)
+
+
+def moe_gateup(
+    act: torch.Tensor,
+    x: torch.Tensor,
+    w_gate_up: torch.Tensor,
+    sorted_ids: torch.Tensor,
+    pad_offsets: torch.Tensor,
+    zeros: torch.Tensor,
+) -> None:
+    """Grouped MoE gate/up GEMM + fused silu*mul into `act` (sorted-by-expert
+    row space). GPU-only (Mixtral prefill hot path); the CPU path keeps the
+    per-expert torch loop in models/mixtral.py."""
+    _hip_or_raise().moe_gateup(act, x, w_gate_up, sorted_ids, pad_offsets, zeros)
+
+
+def moe_down(
+    y: torch.Tensor,
+    act: torch.Tensor,
+    w_down: torch.Tensor,
+    pad_offsets: torch.Tensor,
+) -> None:
+    """Grouped MoE down GEMM in sorted-row space (see moe_gateup)."""
+    _hip_or_raise().moe_down(y, act, w_down, pad_offsets)

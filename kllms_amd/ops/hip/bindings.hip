@@ -198,6 +198,44 @@ void one_shot_allreduce(std::vector<torch::Tensor> bufs) {
                             cur_stream());
 }
 
+// --- grouped MoE expert GEMMs (moe.hip) --------------------------------------
+extern "C" void launch_moe_gateup(void* act, const void* x, const void* w,
+                                  const int* sorted_ids, const int* pad_offsets,
+                                  const void* zeros, int E, int IN, int K,
+                                  hipStream_t stream);
+extern "C" void launch_moe_down(void* y, const void* act, const void* w,
+                                const int* pad_offsets, int E, int H, int IN,
+                                hipStream_t stream);
+
+void moe_gateup(torch::Tensor act, torch::Tensor x, torch::Tensor w,
+                torch::Tensor sorted_ids, torch::Tensor pad_offsets, torch::Tensor zeros) {
+  CHECK_BF16_CONTIG(act); CHECK_BF16_CONTIG(x); CHECK_BF16_CONTIG(w); CHECK_BF16_CONTIG(zeros);
+  TORCH_CHECK(sorted_ids.scalar_type() == at::kInt && pad_offsets.scalar_type() == at::kInt,
+              "routing tensors must be int32");
+  const int E = w.size(0);
+  const int IN = w.size(1) / 2;
+  const int K = w.size(2);
+  TORCH_CHECK(IN % 64 == 0 && K % 32 == 0, "IN % 64, K % 32 required");
+  TORCH_CHECK(act.size(1) == IN && x.size(1) == K && zeros.numel() >= K);
+  TORCH_CHECK(act.size(0) % 128 == 0, "sorted rows must be padded to 128");
+  TORCH_CHECK(pad_offsets.numel() == E + 1);
+  launch_moe_gateup(act.data_ptr(), x.data_ptr(), w.data_ptr(),
+                    sorted_ids.data_ptr<int>(), pad_offsets.data_ptr<int>(),
+                    zeros.data_ptr(), E, IN, K, cur_stream());
+}
+
+void moe_down(torch::Tensor y, torch::Tensor act, torch::Tensor w, torch::Tensor pad_offsets) {
+  CHECK_BF16_CONTIG(y); CHECK_BF16_CONTIG(act); CHECK_BF16_CONTIG(w);
+  const int E = w.size(0);
+  const int H = w.size(1);
+  const int IN = w.size(2);
+  TORCH_CHECK(H % 64 == 0 && IN % 32 == 0, "H % 64, IN % 32 required");
+  TORCH_CHECK(y.size(1) == H && act.size(1) == IN);
+  TORCH_CHECK(pad_offsets.numel() == E + 1);
+  launch_moe_down(y.data_ptr(), act.data_ptr(), w.data_ptr(),
+                  pad_offsets.data_ptr<int>(), E, H, IN, cur_stream());
+}
+
 // --- hipIpc one-shot all-reduce (allreduce.hip) ------------------------------
 extern "C" void* ipc_ar_create(int rank, int world, size_t max_bytes, unsigned char* handles_out);
 extern "C" int ipc_ar_connect(void* ctx, const unsigned char* all_handles);
@@ -233,6 +271,8 @@ void ipc_allreduce_destroy(int64_t ctx) {
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("moe_gateup", &moe_gateup);
+  m.def("moe_down", &moe_down);
   m.def("ipc_allreduce_create", &ipc_allreduce_create);
   m.def("ipc_allreduce_connect", &ipc_allreduce_connect);
   m.def("ipc_allreduce_run", &ipc_allreduce_run);

@@ -25,6 +25,7 @@ SOURCES = [
         "sampling.hip",
         "mfma_selftest.hip",
         "allreduce.hip",
+        "moe.hip",
     )
 ]
 
