@@ -236,8 +236,15 @@ def _get_value_by_path(obj: Any, path: Optional[str]) -> Any:
     for token in path.split("."):
         if token == "":
             continue
-        if token.lstrip("-").isdigit():
+        # int-parse attempt, exactly as the reference: tokens int() accepts
+        # (including unicode decimal digits) are list indices; tokens it
+        # rejects (e.g. superscript '¹', which str.isdigit() wrongly admits)
+        # fall through to dict lookup
+        try:
             idx = int(token)
+        except ValueError:
+            idx = None
+        if idx is not None:
             if isinstance(cur, list) and 0 <= idx < len(cur):
                 cur = cur[idx]
                 continue
