@@ -117,24 +117,28 @@ def make_prompt(rank: int, step: int, i: int, prompt_len_tokens: int) -> str:
 
 def make_schema():
     """Nested Pydantic schema for parse() configs (mirrors the reference's
-    Company/Department test shape, README_TESTS.md:63-73)."""
+    Company/Department test shape, README_TESTS.md:63-73). Fields carry
+    bounds (maxLength / maxItems / integer ranges) so the byte DFA forces
+    structurally complete objects in a bounded token budget — on random-init
+    weights an unbounded field would run to the max_tokens cap instead of
+    closing (the round-1 schema_valid caveat)."""
     from typing import List
 
-    from pydantic import BaseModel
+    from pydantic import BaseModel, Field
 
     class Employee(BaseModel):
-        name: str
-        role: str
+        name: str = Field(max_length=16)
+        role: str = Field(max_length=12)
 
     class Department(BaseModel):
-        name: str
-        headcount: int
-        employees: List[Employee]
+        name: str = Field(max_length=16)
+        headcount: int = Field(ge=0, le=9999)
+        employees: List[Employee] = Field(max_length=2)
 
     class Company(BaseModel):
-        company: str
-        founded: int
-        departments: List[Department]
+        company: str = Field(max_length=20)
+        founded: int = Field(ge=1500, le=2100)
+        departments: List[Department] = Field(max_length=2)
 
     return Company
 

@@ -175,7 +175,10 @@ extern "C" __global__ void __launch_bounds__(NT) sample_kernel(
       for (int i = tid; i < V; i += NT) {
         if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
         const float sx = row[i] * inv_t - smax;
-        if (sx < lo || sx >= hi) continue;
+        // interval convention is (lo, hi] — the coarse pass's truncating
+        // `(int)(-sx*inv)` puts a bin's TOP edge inside the bin (sx = 0,
+        // the max token, is in bin 0), so the refinement must too
+        if (sx <= lo || sx > hi) continue;
         int bn = (int)((hi - sx) * inv_w);
         bn = min(max(bn, 0), SBINS - 1);
         atomicAdd(&my_mass[bn], __expf(sx));
