@@ -262,7 +262,14 @@ def main():
         return res
 
     async def run_step(step_idx: int):
+        import time as _t
+        t0 = _t.perf_counter()
         await asyncio.gather(*(one_request(step_idx, i) for i in range(batch)))
+        if os.environ.get("KLLMS_BENCH_VERBOSE"):
+            sched = getattr(eng_client, "_scheduler", None)
+            log(f"[bench] step {step_idx}: wall={_t.perf_counter()-t0:.3f}s "
+                f"sched={sched.stats if sched else None} "
+                f"eng={getattr(eng_client.engine, 'last_timings', {})}")
 
     def barrier_sync():
         if dist is not None and not tp_mode:

@@ -108,6 +108,12 @@ class EngineConfig(BaseModel):
     # constrained decoding: allow optional JSON whitespace between tokens
     # (default emits compact JSON — smaller DFA, fewer wasted tokens)
     constrained_whitespace: bool = False
+    # cross-request prefix caching: full prompt blocks are published to a
+    # digest-keyed cache; later prompts sharing the prefix skip re-prefilling
+    # it (LRU + eviction under allocator pressure)
+    enable_prefix_caching: bool = True
+    prefix_cache_fraction: float = 0.5  # max fraction of KV blocks held
+
     # scheduler: prefill prompts longer than this in slices interleaved with
     # decode steps, so a long prompt doesn't stall running streams
     # (None = whole-prompt prefill at admission)

@@ -163,6 +163,13 @@ class LLMEngine:
 
         self.model = self._build_model()
         self.kv = self._build_kv_cache()
+        # cross-request shared-prefix KV reuse (kvcache.PrefixCache)
+        self.prefix_cache = None
+        if config.enable_prefix_caching:
+            from .kvcache import PrefixCache
+
+            max_pfx = max(8, int(self.kv.num_blocks * config.prefix_cache_fraction))
+            self.prefix_cache = PrefixCache(self.kv, max_pfx)
         self._graph_runner = None
         if config.use_hip_graphs and self.device.type == "cuda":
             from .graph_runner import DecodeGraphRunner
@@ -301,35 +308,86 @@ class LLMEngine:
         batching scheduler (which interleaves this with decode steps)."""
         dev = self.device
 
-        # ---- shared prefill: one packed varlen batch over all prompts -------
+        # ---- prefix-cache match + sequence allocation ----------------------
+        # a request whose prompt head is cached prefills only its TAIL (the
+        # tail attends to the cached KV through the paged cache); misses go
+        # through the packed varlen prefill as before
+        matched: List[int] = []
+        for req in requests:
+            ids = req.prompt_ids
+            assert len(ids) > 0, "empty prompt"
+            pb = self.prefix_cache.match(ids) if self.prefix_cache is not None else []
+            seq = self._alloc_with_prefix(ids, pb)
+            parent_seqs.append(seq)
+            matched.append(len(pb) * self.kv.block_size)
+
+        # ---- shared prefill: one packed varlen batch over the MISSES --------
         all_ids: List[int] = []
         all_pos: List[int] = []
         all_slots: List[int] = []
         cu = [0]
-        for req in requests:
+        pack_rows: List[int] = []
+        for i, req in enumerate(requests):
+            if matched[i] > 0:
+                continue
             ids = req.prompt_ids
-            assert len(ids) > 0, "empty prompt"
-            seq = self.kv.alloc_sequence(len(ids))
-            parent_seqs.append(seq)
+            seq = parent_seqs[i]
             all_ids.extend(ids)
             all_pos.extend(range(len(ids)))
             all_slots.extend(self.kv.prefill_slot_mapping(seq))
             cu.append(cu[-1] + len(ids))
+            pack_rows.append(i)
 
-        batch = ForwardBatch(
-            mode="prefill",
-            positions=torch.tensor(all_pos, dtype=torch.long, device=dev),
-            slot_mapping=torch.tensor(all_slots, dtype=torch.long, device=dev),
-            kv_caches=self.kv.layer_caches(),
-            cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
-        )
-        input_ids = torch.tensor(all_ids, dtype=torch.long, device=dev)
-        prefill_logits = self.model.forward_prefill(input_ids, batch)  # [n_req, V]
+        logits_by_req: List[Optional[torch.Tensor]] = [None] * len(requests)
+        if pack_rows:
+            batch = ForwardBatch(
+                mode="prefill",
+                positions=torch.tensor(all_pos, dtype=torch.long, device=dev),
+                slot_mapping=torch.tensor(all_slots, dtype=torch.long, device=dev),
+                kv_caches=self.kv.layer_caches(),
+                cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+            )
+            input_ids = torch.tensor(all_ids, dtype=torch.long, device=dev)
+            packed_logits = self.model.forward_prefill(input_ids, batch)  # [n_miss, V]
+            for row, i in enumerate(pack_rows):
+                logits_by_req[i] = packed_logits[row]
+        for i, req in enumerate(requests):
+            if matched[i] > 0:
+                logits_by_req[i] = self.prefill_chunk(
+                    parent_seqs[i], req.prompt_ids, matched[i], len(req.prompt_ids),
+                    want_logits=True,
+                )[0]
+        prefill_logits = torch.stack([l for l in logits_by_req])  # [n_req, V]
+
+        if self.prefix_cache is not None:
+            for req, seq in zip(requests, parent_seqs):
+                self.prefix_cache.register(req.prompt_ids, seq)
 
         # ---- fork + first-token sampling ------------------------------------
         outputs = [RequestOutput(prompt_tokens=len(r.prompt_ids)) for r in requests]
         self._fork_and_sample(requests, parent_seqs, prefill_logits, streams)
         return outputs
+
+    def _alloc_with_prefix(self, ids: List[int], prefix_blocks: List[int]) -> SequenceKV:
+        """Sequence backed by cached prefix blocks (increfed) + freshly
+        allocated tail blocks."""
+        if not prefix_blocks:
+            return self.kv.alloc_sequence(len(ids))
+        bs = self.kv.block_size
+        n_total = (len(ids) + bs - 1) // bs
+        seq = SequenceKV(blocks=[], num_tokens=len(ids))
+        try:
+            for b in prefix_blocks:
+                self.kv.allocator.incref(b)
+                seq.blocks.append(b)
+            for _ in range(n_total - len(prefix_blocks)):
+                seq.blocks.append(self.kv.allocator.alloc())
+        except Exception:
+            for b in seq.blocks:
+                self.kv.allocator.free(b)
+            seq.blocks = []
+            raise
+        return seq
 
     def _fork_and_sample(
         self,

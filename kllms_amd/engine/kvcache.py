@@ -12,8 +12,11 @@ Layout per layer: k_cache/v_cache = [num_blocks, kv_heads, block_size, head_dim]
 
 from __future__ import annotations
 
+import hashlib
+from array import array
+from collections import OrderedDict
 from dataclasses import dataclass, field
-from typing import List, Tuple
+from typing import Callable, List, Optional, Tuple
 
 import torch
 
@@ -23,12 +26,17 @@ class BlockAllocator:
         self.num_blocks = num_blocks
         self._free: list[int] = list(range(num_blocks - 1, -1, -1))
         self._refcount = [0] * num_blocks
+        # called when the pool runs dry before raising — the prefix cache
+        # hooks this to drop its cached (refcounted) blocks under pressure
+        self.on_pressure: Optional[Callable[[], None]] = None
 
     @property
     def num_free(self) -> int:
         return len(self._free)
 
     def alloc(self) -> int:
+        if not self._free and self.on_pressure is not None:
+            self.on_pressure()
         if not self._free:
             raise RuntimeError("KV cache out of blocks")
         b = self._free.pop()
@@ -154,3 +162,89 @@ class PagedKVCache:
             self.allocator.free(b)
         seq.blocks = []
         seq.num_tokens = 0
+
+
+class PrefixCache:
+    """Shared-prefix KV reuse across requests.
+
+    Consensus batches share the chat-template header, and real serving
+    usually shares a long system prompt; this maps a chain digest of each
+    FULL prompt block (BLAKE2b over (parent digest, block token ids) — chain
+    keying makes a block's identity include its whole prefix) to a cache
+    block id. Cached blocks hold one extra refcount owned by the cache; a
+    matching request increfs and SKIPS prefilling those tokens (its tail
+    attends to the cached KV through the paged cache, engine.prefill_chunk).
+    Full blocks are never written after prefill (decode appends go to new /
+    CoW-copied tail blocks), so sharing is safe by the same refcount
+    machinery the n-way fork uses.
+
+    Eviction: LRU above ``max_blocks``, and everything on allocator
+    pressure (BlockAllocator.on_pressure) — a cached block whose only ref is
+    the cache returns to the free pool immediately."""
+
+    def __init__(self, kv: "PagedKVCache", max_blocks: int):
+        self.kv = kv
+        self.max_blocks = max_blocks
+        self._map: "OrderedDict[bytes, int]" = OrderedDict()
+        self.hits = 0
+        self.misses = 0
+        self.tokens_saved = 0
+        kv.allocator.on_pressure = self.evict_all
+
+    def _digests(self, prompt_ids: List[int], n_blocks: int) -> List[bytes]:
+        bs = self.kv.block_size
+        out: List[bytes] = []
+        h = b""
+        for k in range(n_blocks):
+            m = hashlib.blake2b(h, digest_size=16)
+            m.update(array("q", prompt_ids[k * bs: (k + 1) * bs]).tobytes())
+            h = m.digest()
+            out.append(h)
+        return out
+
+    def match(self, prompt_ids: List[int]) -> List[int]:
+        """Longest cached chain of full prompt blocks (always leaves >= 1
+        prompt token un-matched so prefill still produces logits). Returns
+        block ids WITHOUT increfing — the caller owns that step."""
+        bs = self.kv.block_size
+        limit = (len(prompt_ids) - 1) // bs
+        if limit <= 0:
+            return []
+        blocks: List[int] = []
+        for d in self._digests(prompt_ids, limit):
+            b = self._map.get(d)
+            if b is None:
+                break
+            self._map.move_to_end(d)
+            blocks.append(b)
+        if blocks:
+            self.hits += 1
+            self.tokens_saved += len(blocks) * bs
+        else:
+            self.misses += 1
+        return blocks
+
+    def register(self, prompt_ids: List[int], seq: SequenceKV) -> None:
+        """Publish a prefilled prompt's full blocks."""
+        n_full = min(len(seq.blocks), len(prompt_ids) // self.kv.block_size)
+        if n_full <= 0:
+            return
+        for k, d in enumerate(self._digests(prompt_ids, n_full)):
+            if d in self._map:
+                self._map.move_to_end(d)
+                continue
+            self.kv.allocator.incref(seq.blocks[k])
+            self._map[d] = seq.blocks[k]
+        while len(self._map) > self.max_blocks:
+            _, b = self._map.popitem(last=False)
+            self.kv.allocator.free(b)
+
+    def evict_all(self) -> None:
+        while self._map:
+            _, b = self._map.popitem(last=False)
+            self.kv.allocator.free(b)
+
+    @property
+    def stats(self) -> dict:
+        return {"entries": len(self._map), "hits": self.hits,
+                "misses": self.misses, "tokens_saved": self.tokens_saved}

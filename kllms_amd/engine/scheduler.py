@@ -76,6 +76,23 @@ class BatchScheduler:
         # observability
         self.steps = 0
         self.admitted_batches = 0
+        # cumulative phase timers (seconds) — where the worker thread spends
+        # its time; read via .stats for bench diagnostics
+        self.t_drain = 0.0
+        self.t_admit = 0.0
+        self.t_prefill = 0.0
+        self.t_step = 0.0
+
+    @property
+    def stats(self) -> dict:
+        return {
+            "steps": self.steps,
+            "admitted_batches": self.admitted_batches,
+            "drain_s": round(self.t_drain, 4),
+            "admit_s": round(self.t_admit, 4),
+            "chunked_prefill_s": round(self.t_prefill, 4),
+            "decode_step_s": round(self.t_step, 4),
+        }
 
     # --- public -----------------------------------------------------------
     def submit(self, request: GenRequest) -> Future:
@@ -130,6 +147,8 @@ class BatchScheduler:
         return tickets
 
     def _loop(self) -> None:
+        import time as _time
+
         import torch
 
         ctx = _WorkerContext()
@@ -138,21 +157,31 @@ class BatchScheduler:
                 idle = not ctx.active and not ctx.pending
                 # pending chunked prefills reserve their future stream slots
                 reserved = sum(max(1, p.ticket.request.n) for p in ctx.pending)
+                t0 = _time.perf_counter()
                 tickets = self._drain(len(ctx.active) + reserved, block=idle)
+                t1 = _time.perf_counter()
+                self.t_drain += t1 - t0
                 coord = self.coordinator
                 with self.engine_lock:
                     if tickets:
                         if coord is not None:
                             coord.admit([t.request for t in tickets])
                         self._admit(ctx, tickets)
+                        t2 = _time.perf_counter()
+                        self.t_admit += t2 - t1
+                        t1 = t2
                     if ctx.pending:
                         if coord is not None:
                             coord.advance_prefill()
                         self._advance_prefill(ctx)
+                        t2 = _time.perf_counter()
+                        self.t_prefill += t2 - t1
+                        t1 = t2
                     if ctx.active:
                         if coord is not None:
                             coord.step()
                         self._step(ctx)
+                        self.t_step += _time.perf_counter() - t1
 
     def _admit(self, ctx: "_WorkerContext", tickets: List[_Ticket]) -> None:
         chunk = self.engine.config.prefill_chunk_tokens
@@ -166,14 +195,18 @@ class BatchScheduler:
 
     def _start_chunked(self, ctx: "_WorkerContext", t: _Ticket) -> None:
         eng = self.engine
+        ids = t.request.prompt_ids
         try:
-            seq = eng.kv.alloc_sequence(len(t.request.prompt_ids))
+            pb = eng.prefix_cache.match(ids) if eng.prefix_cache is not None else []
+            seq = eng._alloc_with_prefix(ids, pb)
         except Exception as e:
             t.future.set_exception(e)
             return
-        t.output = RequestOutput(prompt_tokens=len(t.request.prompt_ids))
+        t.output = RequestOutput(prompt_tokens=len(ids))
         t.remaining = max(1, t.request.n)
-        ctx.pending.append(_PendingPrefill(ticket=t, seq=seq))
+        # cached prefix blocks are already filled: start chunking after them
+        ctx.pending.append(_PendingPrefill(ticket=t, seq=seq,
+                                           next_pos=len(pb) * eng.kv.block_size))
         self.admitted_batches += 1
 
     def _advance_prefill(self, ctx: "_WorkerContext") -> None:
@@ -195,6 +228,8 @@ class BatchScheduler:
                 return
             ctx.pending.pop(0)
             popped = True
+            if eng.prefix_cache is not None:
+                eng.prefix_cache.register(ids, p.seq)
             eng._fork_and_sample([t.request], [p.seq], logits, new_streams)
         except Exception as e:
             # pop exactly once: only if the try block didn't get there

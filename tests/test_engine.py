@@ -81,6 +81,8 @@ class TestPagedDecodeNumerics:
 class TestKVFork:
     def test_fork_refcounts_and_eager_tail_copy(self, engine):
         kv: PagedKVCache = engine.kv
+        if engine.prefix_cache is not None:
+            engine.prefix_cache.evict_all()
         free0 = kv.allocator.num_free
         parent = kv.alloc_sequence(kv.block_size + 3)  # 2 blocks, second partial
         child1 = kv.fork(parent)
@@ -98,6 +100,8 @@ class TestKVFork:
         kv.free_sequence(parent)
         kv.free_sequence(child1)
         kv.free_sequence(child2)
+        if engine.prefix_cache is not None:
+            engine.prefix_cache.evict_all()
         assert kv.allocator.num_free == free0
 
     def test_fork_at_block_boundary_shares_all(self, engine):
@@ -114,8 +118,12 @@ class TestKVFork:
         kv.free_sequence(child)
 
     def test_no_block_leak_after_generate(self, engine):
+        if engine.prefix_cache is not None:
+            engine.prefix_cache.evict_all()
         free0 = engine.kv.allocator.num_free
         engine.generate([GenRequest(prompt_ids=list(range(1, 40)), n=5, sampling=greedy(9))])
+        if engine.prefix_cache is not None:
+            engine.prefix_cache.evict_all()
         assert engine.kv.allocator.num_free == free0
 
 
@@ -148,11 +156,15 @@ class TestRobustness:
         eng = LLMEngine(EngineConfig(
             model="tiny-llama", max_kv_blocks=8, use_hip_graphs=False, device="cpu", seed=0,
         ))
+        if eng.prefix_cache is not None:
+            eng.prefix_cache.evict_all()
         free0 = eng.kv.allocator.num_free
         with pytest.raises(RuntimeError):
             # 8 blocks * 16 = 128 slots; this needs far more
             eng.generate([GenRequest(prompt_ids=list(range(1, 100)), n=8,
                                      sampling=greedy(64))])
+        if eng.prefix_cache is not None:
+            eng.prefix_cache.evict_all()
         assert eng.kv.allocator.num_free == free0, "blocks leaked after failure"
         # engine still serves afterwards
         out = eng.generate([GenRequest(prompt_ids=[1, 2, 3], n=1, sampling=greedy(4))])[0]

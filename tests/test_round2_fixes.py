@@ -207,6 +207,8 @@ class TestSchedulerChunkedFailure:
         out2 = f2.result(timeout=30)
         assert len(out2.streams) == 1 and len(out2.streams[0].token_ids) > 0
         sched.shutdown()
+        if eng.prefix_cache is not None:
+            eng.prefix_cache.evict_all()
         assert eng.kv.allocator.num_free == free0  # no leaked KV blocks
 
 

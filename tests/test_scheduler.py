@@ -177,10 +177,14 @@ class TestChunkedPrefill:
             model="tiny-llama", max_kv_blocks=256, use_hip_graphs=False,
             device="cpu", seed=0, prefill_chunk_tokens=4,
         ))
+        if eng.prefix_cache is not None:
+            eng.prefix_cache.evict_all()
         free0 = eng.kv.allocator.num_free
         sched = BatchScheduler(eng)
         sched.submit(greedy_req([7] * 21, 6, n=3)).result(timeout=120)
         sched.shutdown()
+        if eng.prefix_cache is not None:
+            eng.prefix_cache.evict_all()
         assert eng.kv.allocator.num_free == free0
 
     def test_async_client_over_chunked_scheduler(self):
@@ -230,6 +234,8 @@ class TestChunkedPrefill:
             model="tiny-llama", max_kv_blocks=8, use_hip_graphs=False,
             device="cpu", seed=0, prefill_chunk_tokens=4, max_seq_len=4096,
         ))
+        if eng.prefix_cache is not None:
+            eng.prefix_cache.evict_all()
         free0 = eng.kv.allocator.num_free
         sched = BatchScheduler(eng)
         f = sched.submit(greedy_req([5] * 400, 4))  # needs 25 blocks, only 8 exist
@@ -239,6 +245,8 @@ class TestChunkedPrefill:
         ok = sched.submit(greedy_req([1, 2, 3], 4)).result(timeout=60)
         sched.shutdown()
         assert len(ok.streams) == 1
+        if eng.prefix_cache is not None:
+            eng.prefix_cache.evict_all()
         assert eng.kv.allocator.num_free == free0
 
 
@@ -262,6 +270,8 @@ class TestInteractionSoak:
             device="cpu", seed=0, max_batch_size=32, prefill_chunk_tokens=8,
             kv_cache_dtype="fp8_e4m3",
         ))
+        if eng.prefix_cache is not None:
+            eng.prefix_cache.evict_all()
         free0 = eng.kv.allocator.num_free
         constraint = JsonSchemaConstraint(Rec.model_json_schema(), eng.tokenizer)
         sched = BatchScheduler(eng, admit_wait_s=0.02)
@@ -284,4 +294,6 @@ class TestInteractionSoak:
                     if s.finish_reason == "stop":
                         Rec.model_validate(_json.loads(s.text))
         sched.shutdown()
+        if eng.prefix_cache is not None:
+            eng.prefix_cache.evict_all()
         assert eng.kv.allocator.num_free == free0
