@@ -64,6 +64,8 @@ def main():
     on_gpu = torch.cuda.is_available()
     if not on_gpu:
         args.model = "tiny-llama"
+        global SYSTEM
+        SYSTEM = SYSTEM[:260]  # tiny preset's 512-token context window
     off = run(False, args.model, args.steps, args.batch, args.n, on_gpu)
     on = run(True, args.model, args.steps, args.batch, args.n, on_gpu)
     speedup = round(on["req_per_s"] / off["req_per_s"], 3)
