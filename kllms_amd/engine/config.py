@@ -113,6 +113,9 @@ class EngineConfig(BaseModel):
     # it (LRU + eviction under allocator pressure)
     enable_prefix_caching: bool = True
     prefix_cache_fraction: float = 0.5  # max fraction of KV blocks held
+    # ignore hits shorter than this: a short cached head saves less prefill
+    # than the tail's decode-mode forward costs over the packed kernel
+    prefix_cache_min_tokens: int = 128
 
     # scheduler: prefill prompts longer than this in slices interleaved with
     # decode steps, so a long prompt doesn't stall running streams

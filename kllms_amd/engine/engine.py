@@ -313,10 +313,15 @@ class LLMEngine:
         # tail attends to the cached KV through the paged cache); misses go
         # through the packed varlen prefill as before
         matched: List[int] = []
+        min_hit = self.config.prefix_cache_min_tokens
         for req in requests:
             ids = req.prompt_ids
             assert len(ids) > 0, "empty prompt"
             pb = self.prefix_cache.match(ids) if self.prefix_cache is not None else []
+            if pb and len(pb) * self.kv.block_size < min_hit:
+                # a short hit saves less than the tail's decode-mode-prefill
+                # overhead costs — treat as a miss (packed prefill is faster)
+                pb = []
             seq = self._alloc_with_prefix(ids, pb)
             parent_seqs.append(seq)
             matched.append(len(pb) * self.kv.block_size)
@@ -351,12 +356,17 @@ class LLMEngine:
             packed_logits = self.model.forward_prefill(input_ids, batch)  # [n_miss, V]
             for row, i in enumerate(pack_rows):
                 logits_by_req[i] = packed_logits[row]
-        for i, req in enumerate(requests):
-            if matched[i] > 0:
-                logits_by_req[i] = self.prefill_chunk(
-                    parent_seqs[i], req.prompt_ids, matched[i], len(req.prompt_ids),
-                    want_logits=True,
-                )[0]
+        hit_idx = [i for i in range(len(requests)) if matched[i] > 0]
+        if hit_idx:
+            # ALL cached-prefix tails go through ONE decode-mode forward
+            # (rows from different sequences, per-row block tables)
+            tail_logits = self._prefill_tails(
+                [parent_seqs[i] for i in hit_idx],
+                [requests[i].prompt_ids for i in hit_idx],
+                [matched[i] for i in hit_idx],
+            )
+            for row, i in enumerate(hit_idx):
+                logits_by_req[i] = tail_logits[row]
         prefill_logits = torch.stack([l for l in logits_by_req])  # [n_req, V]
 
         if self.prefix_cache is not None:
@@ -367,6 +377,43 @@ class LLMEngine:
         outputs = [RequestOutput(prompt_tokens=len(r.prompt_ids)) for r in requests]
         self._fork_and_sample(requests, parent_seqs, prefill_logits, streams)
         return outputs
+
+    def _prefill_tails(self, seqs: List[SequenceKV], ids_list: List[List[int]],
+                       starts: List[int]) -> torch.Tensor:
+        """One decode-mode forward over the concatenated TAILS of several
+        prefix-cache-hit prompts (row i attends to its own sequence's KV,
+        cached head included). Returns last-token logits per sequence
+        [len(seqs), V]."""
+        dev = self.device
+        all_ids: List[int] = []
+        all_pos: List[int] = []
+        all_slots: List[int] = []
+        ctx_lens: List[int] = []
+        bt_blocks: List[List[int]] = []
+        last_rows: List[int] = []
+        for seq, ids, start in zip(seqs, ids_list, starts):
+            slots_all = self.kv.prefill_slot_mapping(seq)
+            for p in range(start, len(ids)):
+                all_ids.append(ids[p])
+                all_pos.append(p)
+                all_slots.append(slots_all[p])
+                ctx_lens.append(p + 1)
+                bt_blocks.append(seq.blocks)
+            last_rows.append(len(all_ids) - 1)
+        max_nb = max(len(b) for b in bt_blocks)
+        bt = torch.zeros((len(bt_blocks), max_nb), dtype=torch.int32)
+        for r, blocks in enumerate(bt_blocks):
+            bt[r, : len(blocks)] = torch.tensor(blocks, dtype=torch.int32)
+        batch = ForwardBatch(
+            mode="decode",
+            positions=torch.tensor(all_pos, dtype=torch.long, device=dev),
+            slot_mapping=torch.tensor(all_slots, dtype=torch.long, device=dev),
+            kv_caches=self.kv.layer_caches(),
+            block_tables=bt.to(dev),
+            context_lens=torch.tensor(ctx_lens, dtype=torch.int32, device=dev),
+        )
+        hidden = self.model.forward_hidden(torch.tensor(all_ids, dtype=torch.long, device=dev), batch)
+        return self.model.compute_logits(hidden[torch.tensor(last_rows, device=dev)])
 
     def _alloc_with_prefix(self, ids: List[int], prefix_blocks: List[int]) -> SequenceKV:
         """Sequence backed by cached prefix blocks (increfed) + freshly

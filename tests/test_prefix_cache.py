@@ -15,7 +15,8 @@ from kllms_amd.engine.sampling import SamplingParams
 
 def _cfg(**kw):
     base = dict(model="tiny-llama", max_kv_blocks=256, use_hip_graphs=False,
-                default_max_new_tokens=8, device="cpu", seed=0)
+                default_max_new_tokens=8, device="cpu", seed=0,
+                prefix_cache_min_tokens=16)
     base.update(kw)
     return EngineConfig(**base)
 
