@@ -261,15 +261,22 @@ def main():
         quality_scores.append(_mean_likelihood(res.likelihoods))
         return res
 
+    prev_stats = {}
+
     async def run_step(step_idx: int):
         import time as _t
         t0 = _t.perf_counter()
         await asyncio.gather(*(one_request(step_idx, i) for i in range(batch)))
         if os.environ.get("KLLMS_BENCH_VERBOSE"):
             sched = getattr(eng_client, "_scheduler", None)
+            cur = dict(sched.stats) if sched else {}
+            delta = {k: round(v - prev_stats.get(k, 0), 4) if isinstance(v, float)
+                     else v - prev_stats.get(k, 0) for k, v in cur.items()}
+            prev_stats.clear()
+            prev_stats.update(cur)
+            pc = getattr(eng_client.engine, "prefix_cache", None)
             log(f"[bench] step {step_idx}: wall={_t.perf_counter()-t0:.3f}s "
-                f"sched={sched.stats if sched else None} "
-                f"eng={getattr(eng_client.engine, 'last_timings', {})}")
+                f"sched_delta={delta} prefix={pc.stats if pc else None}")
 
     def barrier_sync():
         if dist is not None and not tp_mode:

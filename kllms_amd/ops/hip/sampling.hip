@@ -216,7 +216,11 @@ extern "C" __global__ void __launch_bounds__(NT) sample_kernel(
       __syncthreads();
     }
     if (sh_state != 2) {
-      if (tid == 0) sh_thresh = (smax + sh_lo) * temp;
+      // threshold stays in SX space: pass D recomputes each token's sx with
+      // the IDENTICAL f32 expression and compares sx > lo, so the decision
+      // is bit-consistent with the refinement that chose lo (converting
+      // back to x-space loses the ~1e-9 refined margin to fp32 rounding)
+      if (tid == 0) sh_thresh = sh_lo;
       __syncthreads();
       x_thresh = sh_thresh;
     }
@@ -230,7 +234,9 @@ extern "C" __global__ void __launch_bounds__(NT) sample_kernel(
   for (int i = tid; i < V; i += NT) {
     if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
     const float x = row[i];
-    if (x < x_thresh) continue;
+    // sx-space admission, exclusive lower edge (matches the (lo, hi] bin
+    // convention of passes B/C); x_thresh = -inf admits everything
+    if (x * inv_t - smax <= x_thresh) continue;
     const float u = u64_to_uniform(splitmix64(base ^ (uint64_t)i));
     const float gumbel = -__logf(-__logf(u));
     const float key = x * inv_t + gumbel;

@@ -20,7 +20,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch  # noqa: E402
 
 SYSTEM = ("You are a meticulous data-extraction assistant. Follow the output schema exactly. "
-          "Rules: " + " ".join(f"rule {i}: always check field {i} twice." for i in range(120)))
+          "Rules: " + " ".join(f"rule {i}: always check field {i} twice." for i in range(60)))
 
 
 def run(enable_cache: bool, model: str, steps: int, batch: int, n: int, on_gpu: bool):
@@ -28,7 +28,7 @@ def run(enable_cache: bool, model: str, steps: int, batch: int, n: int, on_gpu: 
 
     kw = dict(model=model, device="cuda:0" if on_gpu else "cpu",
               use_hip_graphs=on_gpu, seed=0, default_max_new_tokens=32,
-              max_seq_len=4096, enable_prefix_caching=enable_cache)
+              max_seq_len=8192, enable_prefix_caching=enable_cache)
     if not on_gpu:
         kw["max_kv_blocks"] = 4096
     k = KLLMs(**kw)
