@@ -72,10 +72,13 @@ def _consensus_over_contents(
     if len(contents) >= 2:
         if consensus_settings.string_similarity_method == "embeddings" and get_embeddings_from_text is not None:
             # M5: one batched on-device embedding pass + one cosine GEMM
-            # replaces every per-pair embedding call in the alignment hot loop
-            from .accel import precompute_similarity_cache
+            # replaces every per-pair embedding call in the alignment hot
+            # loop; short-string pairs batch through the Myers Levenshtein
+            # kernel (the reference's <=50-char fallback path)
+            from .accel import precompute_levenshtein_cache, precompute_similarity_cache
 
-            precompute_similarity_cache(contents, get_embeddings_from_text)
+            precompute_similarity_cache(contents, get_embeddings_from_text, client=client)
+            precompute_levenshtein_cache(contents, client=client)
         if aligner == "key":
             # the reference's key-based aligner takes no embed fn / client
             # (key matching is exact; ref key_based_alignment.py:350-359)

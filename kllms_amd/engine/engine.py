@@ -718,13 +718,22 @@ class LLMEngine:
         """Local embedding path: token-embedding mean-pool, L2-normalized.
         The whole batch runs as ONE flat gather + segment-mean on device.
         Returns (vectors, total_tokens)."""
+        unit, total = self.embed_dev(texts)
+        out = unit.cpu().tolist()
+        return out, total
+
+    def embed_dev(self, texts: List[str]) -> Tuple[torch.Tensor, int]:
+        """Device-resident variant of embed(): returns the [N, H] float32
+        unit-vector tensor ON the engine device (zero rows for empty texts),
+        so the consensus accel can run its cosine GEMM without a host round
+        trip (consolidation similarity math on-device)."""
         emb = self.model.embed_tokens.weight
         H = emb.shape[1]
         ids_per_text = [self.tokenizer.encode(t) for t in texts]
         total_tokens = sum(len(ids) for ids in ids_per_text)
         flat = [i for ids in ids_per_text for i in ids]
         if not flat:
-            return [[0.0] * H for _ in texts], 0
+            return torch.zeros(len(texts), H, device=self.device), 0
         flat_t = torch.tensor(flat, dtype=torch.long, device=self.device)
         seg = torch.tensor(
             [s for s, ids in enumerate(ids_per_text) for _ in ids],
@@ -739,8 +748,7 @@ class LLMEngine:
         means = sums / counts
         norms = means.norm(dim=1, keepdim=True).clamp_min(1e-12)
         unit = torch.where(norms > 1e-11, means / norms, means)
-        out = unit.cpu().tolist()
-        for i, ids in enumerate(ids_per_text):
-            if not ids:
-                out[i] = [0.0] * H
-        return out, total_tokens
+        empties = [i for i, ids in enumerate(ids_per_text) if not ids]
+        if empties:
+            unit[torch.tensor(empties, device=self.device)] = 0.0
+        return unit, total_tokens

@@ -183,3 +183,15 @@ def moe_down(
 ) -> None:
     """Grouped MoE down GEMM in sorted-row space (see moe_gateup)."""
     _hip_or_raise().moe_down(y, act, w_down, pad_offsets)
+
+
+def levenshtein_pairs(
+    chars: torch.Tensor,
+    lens: torch.Tensor,
+    pair_i: torch.Tensor,
+    pair_j: torch.Tensor,
+) -> torch.Tensor:
+    """Batched Myers bit-parallel edit distance over packed [N, 64] uint8
+    normalized strings; returns int32 distances per (i, j) pair. GPU-only —
+    the CPU consensus path keeps the pure-python DP (utils/text.py)."""
+    return _hip_or_raise().levenshtein_pairs(chars, lens, pair_i, pair_j)

@@ -236,6 +236,25 @@ void moe_down(torch::Tensor y, torch::Tensor act, torch::Tensor w, torch::Tensor
                   pad_offsets.data_ptr<int>(), E, H, IN, cur_stream());
 }
 
+// --- batched Levenshtein (levenshtein.hip) -----------------------------------
+extern "C" void launch_levenshtein(int* out, const unsigned char* chars, const int* lens,
+                                   const int* pi, const int* pj, long n_pairs,
+                                   hipStream_t stream);
+
+torch::Tensor levenshtein_pairs(torch::Tensor chars, torch::Tensor lens,
+                                torch::Tensor pair_i, torch::Tensor pair_j) {
+  TORCH_CHECK(chars.is_cuda() && chars.scalar_type() == at::kByte && chars.size(1) == 64 &&
+              chars.is_contiguous(), "chars must be contiguous uint8 [N, 64] on GPU");
+  TORCH_CHECK(lens.scalar_type() == at::kInt && pair_i.scalar_type() == at::kInt &&
+              pair_j.scalar_type() == at::kInt);
+  const long P = pair_i.numel();
+  auto out = torch::empty({P}, torch::dtype(torch::kInt).device(chars.device()));
+  launch_levenshtein(out.data_ptr<int>(),
+                     chars.data_ptr<unsigned char>(), lens.data_ptr<int>(),
+                     pair_i.data_ptr<int>(), pair_j.data_ptr<int>(), P, cur_stream());
+  return out;
+}
+
 // --- hipIpc one-shot all-reduce (allreduce.hip) ------------------------------
 extern "C" void* ipc_ar_create(int rank, int world, size_t max_bytes, unsigned char* handles_out);
 extern "C" int ipc_ar_connect(void* ctx, const unsigned char* all_handles);
@@ -271,6 +290,7 @@ void ipc_allreduce_destroy(int64_t ctx) {
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("levenshtein_pairs", &levenshtein_pairs);
   m.def("moe_gateup", &moe_gateup);
   m.def("moe_down", &moe_down);
   m.def("ipc_allreduce_create", &ipc_allreduce_create);

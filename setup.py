@@ -26,6 +26,7 @@ SOURCES = [
         "mfma_selftest.hip",
         "allreduce.hip",
         "moe.hip",
+        "levenshtein.hip",
     )
 ]
 

@@ -615,3 +615,48 @@ class TestMoEGroupedGEMM:
                                        sampling=SamplingParams(temperature=0.0, max_tokens=4))])[0]
         assert out.streams[0].token_ids == out.streams[1].token_ids
         assert len(out.streams[0].token_ids) == 4
+
+
+class TestLevenshteinKernel:
+    def test_matches_python_dp(self):
+        import random
+        import string as _string
+
+        from kllms_amd.utils.text import levenshtein_distance
+
+        rng = random.Random(11)
+        alpha = _string.ascii_lowercase + _string.digits
+        strs = [""] + ["".join(rng.choice(alpha) for _ in range(rng.randint(1, 50)))
+                       for _ in range(40)]
+        N = len(strs)
+        chars = torch.zeros(N, 64, dtype=torch.uint8)
+        lens = torch.zeros(N, dtype=torch.int32)
+        for i, s in enumerate(strs):
+            b = s.encode()
+            chars[i, :len(b)] = torch.tensor(list(b), dtype=torch.uint8)
+            lens[i] = len(b)
+        ii, jj = torch.triu_indices(N, N, offset=1)
+        out = ops.levenshtein_pairs(chars.to(DEV), lens.to(DEV),
+                                    ii.to(torch.int32).to(DEV), jj.to(torch.int32).to(DEV))
+        out = out.cpu().tolist()
+        for k in range(len(out)):
+            want = levenshtein_distance(strs[int(ii[k])], strs[int(jj[k])])
+            assert out[k] == want, (strs[int(ii[k])], strs[int(jj[k])], out[k], want)
+
+    def test_accel_precompute_consistent_with_cpu_path(self):
+        from kllms_amd.consensus import similarity as simmod
+        from kllms_amd.consensus.accel import precompute_levenshtein_cache
+        from kllms_amd.engine.api import LocalEngineClient
+
+        client = LocalEngineClient(model="tiny-llama", max_kv_blocks=128,
+                                   use_hip_graphs=False, device=DEV)
+        _ = client.engine
+        contents = [{"city": "Paris", "note": "short text A"},
+                    {"city": "paris!", "note": "short text B"},
+                    {"city": "Berlin", "note": "short text A"}]
+        n = precompute_levenshtein_cache(contents, client=client)
+        assert n > 0
+        # cached values equal the CPU formula
+        got = simmod.string_similarity("Paris", "paris!", "embeddings", None)
+        want = simmod.levenshtein_similarity("Paris", "paris!")
+        assert got == pytest.approx(want, abs=1e-9)
