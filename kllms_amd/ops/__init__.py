@@ -148,6 +148,7 @@ def sample(
     seeds: torch.Tensor,
     steps: torch.Tensor,
     mask: Optional[torch.Tensor] = None,
+    dbg: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     if logits.is_cuda and not _force_torch():
         B = logits.shape[0]
@@ -156,6 +157,7 @@ def sample(
         _hip_or_raise().sample(
             tokens, logprobs, logits, temperatures, top_ps, top_ks, seeds, steps,
             mask if mask is not None else torch.empty(0, dtype=torch.int32, device=logits.device),
+            dbg if dbg is not None else torch.empty(0, dtype=torch.float32, device=logits.device),
         )
         return tokens, logprobs
     return torch_ref.sample(logits, temperatures, top_ps, top_ks, seeds, steps, mask)

@@ -16,7 +16,7 @@ extern "C" __global__ void store_kv_fp8_kernel(const bf16_t*, const bf16_t*, uns
 extern "C" void launch_attn_decode_partial(float*, const bf16_t*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int, int, int, hipStream_t);
 extern "C" __global__ void attn_decode_reduce_kernel(bf16_t*, const float*, const int*, int, int, int, int);
 extern "C" __global__ void attn_prefill_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, const int*, float, int, int, int, int);  // grouped: [G], [G], [G*4]
-extern "C" __global__ void sample_kernel(int64_t*, float*, const float*, const float*, const float*, const int*, const int64_t*, const int64_t*, const uint32_t*, int);
+extern "C" __global__ void sample_kernel(int64_t*, float*, const float*, const float*, const float*, const int*, const int64_t*, const int64_t*, const uint32_t*, int, float*);
 extern "C" __global__ void mfma_selftest_kernel(float*, const bf16_t*, const bf16_t*);
 
 #define CHECK_BF16_CONTIG(t) \
@@ -151,7 +151,8 @@ void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k, torch::Te
 
 void sample(torch::Tensor tokens, torch::Tensor logprobs, torch::Tensor logits,
             torch::Tensor temperatures, torch::Tensor top_ps, torch::Tensor top_ks,
-            torch::Tensor seeds, torch::Tensor steps, torch::Tensor mask) {
+            torch::Tensor seeds, torch::Tensor steps, torch::Tensor mask,
+            torch::Tensor dbg) {
   TORCH_CHECK(logits.is_cuda() && logits.is_contiguous() && logits.scalar_type() == at::kFloat);
   const int B = logits.size(0), V = logits.size(1);
   const uint32_t* mptr = mask.numel() > 0
@@ -161,7 +162,8 @@ void sample(torch::Tensor tokens, torch::Tensor logprobs, torch::Tensor logits,
                      tokens.data_ptr<int64_t>(), logprobs.data_ptr<float>(),
                      logits.data_ptr<float>(), temperatures.data_ptr<float>(),
                      top_ps.data_ptr<float>(), top_ks.data_ptr<int>(),
-                     seeds.data_ptr<int64_t>(), steps.data_ptr<int64_t>(), mptr, V);
+                     seeds.data_ptr<int64_t>(), steps.data_ptr<int64_t>(), mptr, V,
+                     dbg.numel() ? dbg.data_ptr<float>() : nullptr);
 }
 
 torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor B) {

@@ -22,6 +22,19 @@
 
 #include "common.h"
 
+// sx(x) must be BIT-IDENTICAL across pass B (histogram), pass C's
+// refinement rescans and pass D (admission). hipcc contracts even
+// __fmul_rn/__fsub_rn pairs into an fma, and may do so at one site but not
+// another — the recomputed sx of the MAX token then comes out as the
+// product's rounding residual (~1e-10, either sign) instead of 0 and falls
+// outside the (lo, 0] boundary interval (observed: the max missing from the
+// refinement histogram on half the rows -> k+1 admitted). ONE explicit fma
+// is a single IR op the compiler cannot split or re-associate, so every
+// pass computes the same bits.
+__device__ __forceinline__ float sx_of(float x, float inv_t, float smax) {
+  return __builtin_fmaf(x, inv_t, -smax);
+}
+
 #define NT 1024
 #define NW (NT / WAVE)     // 16 waves
 #define SBINS 256
@@ -37,7 +50,7 @@ extern "C" __global__ void __launch_bounds__(NT) sample_kernel(
     const int64_t* __restrict__ seeds,
     const int64_t* __restrict__ steps,
     const uint32_t* __restrict__ mask,    // [B, ceil(V/32)] or nullptr
-    int V) {
+    int V, float* __restrict__ dbg) {   // dbg: [B, 16] or nullptr (diagnostics)
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -91,7 +104,7 @@ extern "C" __global__ void __launch_bounds__(NT) sample_kernel(
     const float x = row[i];
     lsum_unscaled += __expf(x - gmax);
     if (truncate) {
-      const float sx = x * inv_t - smax;
+      const float sx = sx_of(x, inv_t, smax);
       const float w = __expf(sx);
       lsum_scaled += w;
       int bin = (int)(-sx * (SBINS / SRANGE));
@@ -129,8 +142,13 @@ extern "C" __global__ void __launch_bounds__(NT) sample_kernel(
     __syncthreads();
     // boundary-interval state (sx space: 0 at the max, negative below);
     // need_* is the admission still OWED by the boundary interval after
-    // everything strictly above it is admitted
+    // everything strictly above it is admitted. MEMBERSHIP in the boundary
+    // interval is tested by re-evaluating the PREVIOUS level's exact binning
+    // expression (sel_*) — never by interval comparisons, which disagree
+    // with the truncating multiply by an ulp at bin edges (and at sx ~ 0).
     __shared__ float sh_lo, sh_hi, sh_need_mass;
+    __shared__ float sh_sel_hi, sh_sel_invw;
+    __shared__ int sh_sel_cut, sh_sel_coarse;
     __shared__ unsigned int sh_need_cnt;
     __shared__ int sh_state;  // 0 = threshold at sh_lo, 1 = refine, 2 = admit all
     if (tid == 0) {
@@ -155,8 +173,16 @@ extern "C" __global__ void __launch_bounds__(NT) sample_kernel(
         sh_hi = -cut_bin * binw;
         sh_need_mass = need_mass - before_mass;
         sh_need_cnt = (need_cnt == 0xffffffffu) ? 0xffffffffu : need_cnt - before_cnt;
+        sh_sel_coarse = 1;
+        sh_sel_cut = cut_bin;
         // the whole bin is exactly owed -> no refinement needed
         sh_state = (hist_cnt[cut_bin] == sh_need_cnt) ? 0 : 1;
+        if (dbg) {
+          dbg[b * 16 + 0] = (float)cut_bin;
+          dbg[b * 16 + 1] = (float)hist_cnt[cut_bin];
+          dbg[b * 16 + 2] = (float)sh_need_cnt;
+          dbg[b * 16 + 3] = (float)sh_state;
+        }
       }
     }
     __syncthreads();
@@ -172,13 +198,22 @@ extern "C" __global__ void __launch_bounds__(NT) sample_kernel(
         hist_cnt[i] = 0;
       }
       __syncthreads();
+      const int sel_coarse = sh_sel_coarse;
+      const int sel_cut = sh_sel_cut;
+      const float sel_hi = sh_sel_hi;
+      const float sel_invw = sh_sel_invw;
       for (int i = tid; i < V; i += NT) {
         if (mrow && !((mrow[i >> 5] >> (i & 31)) & 1)) continue;
-        const float sx = row[i] * inv_t - smax;
-        // interval convention is (lo, hi] — the coarse pass's truncating
-        // `(int)(-sx*inv)` puts a bin's TOP edge inside the bin (sx = 0,
-        // the max token, is in bin 0), so the refinement must too
-        if (sx <= lo || sx > hi) continue;
+        const float sx = sx_of(row[i], inv_t, smax);
+        // membership = the EXACT bin the previous level counted this token
+        // into equals its boundary bin (bit-identical re-evaluation)
+        int pb;
+        if (sel_coarse) {
+          pb = min((int)(-sx * (SBINS / SRANGE)), SBINS - 1);
+        } else {
+          pb = min(max((int)((sel_hi - sx) * sel_invw), 0), SBINS - 1);
+        }
+        if (pb != sel_cut) continue;
         int bn = (int)((hi - sx) * inv_w);
         bn = min(max(bn, 0), SBINS - 1);
         atomicAdd(&my_mass[bn], __expf(sx));
@@ -211,7 +246,23 @@ extern "C" __global__ void __launch_bounds__(NT) sample_kernel(
         if (sh_need_cnt != 0xffffffffu) sh_need_cnt -= acc_cnt - hist_cnt[cut_bin];
         sh_lo = hi - (cut_bin + 1) * subw;
         sh_hi = hi - cut_bin * subw;
+        sh_sel_coarse = 0;
+        sh_sel_cut = cut_bin;
+        sh_sel_hi = hi;
+        sh_sel_invw = inv_w;
         sh_state = (hist_cnt[cut_bin] == sh_need_cnt) ? 0 : 1;
+        if (dbg) {
+          dbg[b * 16 + 4 + it * 4 + 0] = (float)cut_bin;
+          dbg[b * 16 + 4 + it * 4 + 1] = (float)hist_cnt[cut_bin];
+          dbg[b * 16 + 4 + it * 4 + 2] = (float)sh_need_cnt;
+          dbg[b * 16 + 4 + it * 4 + 3] = sh_lo;
+          if (it == 0) {
+            dbg[b * 16 + 12] = (float)hist_cnt[0];
+            dbg[b * 16 + 13] = sx_of(row[argmax_tok], inv_t, smax);
+            dbg[b * 16 + 14] = smax;
+            dbg[b * 16 + 15] = __fmul_rn(row[argmax_tok], inv_t);
+          }
+        }
       }
       __syncthreads();
     }
@@ -236,7 +287,7 @@ extern "C" __global__ void __launch_bounds__(NT) sample_kernel(
     const float x = row[i];
     // sx-space admission, exclusive lower edge (matches the (lo, hi] bin
     // convention of passes B/C); x_thresh = -inf admits everything
-    if (x * inv_t - smax <= x_thresh) continue;
+    if (sx_of(x, inv_t, smax) <= x_thresh) continue;
     const float u = u64_to_uniform(splitmix64(base ^ (uint64_t)i));
     const float gumbel = -__logf(-__logf(u));
     const float key = x * inv_t + gumbel;
