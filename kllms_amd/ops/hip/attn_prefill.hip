@@ -158,13 +158,26 @@ extern "C" __global__ void __launch_bounds__(64 * NWAVES) attn_prefill_kernel(
         auto sw = __builtin_amdgcn_permlane32_swap(__float_as_uint(tmax), __float_as_uint(tmax), false, false);
         tmax = fmaxf(__uint_as_float(sw[0]), __uint_as_float(sw[1]));
       }
-      const float m_new = fmaxf(m_run, tmax);
-      const float alpha = (m_run == -INFINITY) ? 0.0f : __expf(m_run - m_new);
+      // defer-max (T13): rows are lane-local, so the defer decision is
+      // per-lane — keep the old running max while the tile max grew < THR
+      // (P then bounded by e^THR, fine for f32 p / bf16 pack). Decision
+      // happens BEFORE this tile's P is exponentiated (the safe order).
+      const float DEFER_THR = 8.0f;
+      float m_new, alpha;
+      if (tmax <= m_run + DEFER_THR) {       // first tile: m_run=-inf fails this
+        m_new = m_run;
+        alpha = 1.0f;
+      } else {
+        m_new = fmaxf(m_run, tmax);
+        alpha = (m_run == -INFINITY) ? 0.0f : __expf(m_run - m_new);
+      }
       m_run = m_new;
+      if (__builtin_amdgcn_ballot_w64(alpha != 1.0f)) {
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt)
+        for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) o_acc[dt][r] *= alpha;
+          for (int r = 0; r < 16; ++r) o_acc[dt][r] *= alpha;
+      }
 
       float p[16];
       float psum = 0.0f;
