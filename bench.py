@@ -137,7 +137,7 @@ def make_schema():
 
     class Company(BaseModel):
         company: str = Field(max_length=20)
-        founded: int = Field(ge=1500, le=2100)
+        founded: int = Field(ge=0, le=9999)  # range the digit-count DFA bound can honor
         departments: List[Department] = Field(max_length=2)
 
     return Company
@@ -146,8 +146,8 @@ def make_schema():
 CONFIG_DEFAULTS = {
     # model, n, mode, batch, max_new
     "llama8b": dict(model="llama-3-8b", n=5, mode="create", batch=24, max_new=64),
-    "parse16": dict(model="llama-3-8b", n=16, mode="parse", batch=4, max_new=200),
-    "mixtral8": dict(model="mixtral-8x7b", n=8, mode="parse", batch=4, max_new=200),
+    "parse16": dict(model="llama-3-8b", n=16, mode="parse", batch=4, max_new=352),
+    "mixtral8": dict(model="mixtral-8x7b", n=8, mode="parse", batch=8, max_new=352),
     "llama70b": dict(model="llama-3-70b", n=5, mode="create", batch=4, max_new=64),
 }
 
