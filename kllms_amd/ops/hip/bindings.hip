@@ -217,7 +217,7 @@ void moe_gateup(torch::Tensor act, torch::Tensor x, torch::Tensor w,
   const int E = w.size(0);
   const int IN = w.size(1) / 2;
   const int K = w.size(2);
-  TORCH_CHECK(IN % 64 == 0 && K % 32 == 0, "IN % 64, K % 32 required");
+  TORCH_CHECK(IN % 64 == 0 && K % 64 == 0, "IN % 64, K % 64 required");
   TORCH_CHECK(act.size(1) == IN && x.size(1) == K && zeros.numel() >= K);
   TORCH_CHECK(act.size(0) % 128 == 0, "sorted rows must be padded to 128");
   TORCH_CHECK(pad_offsets.numel() == E + 1);
@@ -231,7 +231,7 @@ void moe_down(torch::Tensor y, torch::Tensor act, torch::Tensor w, torch::Tensor
   const int E = w.size(0);
   const int H = w.size(1);
   const int IN = w.size(2);
-  TORCH_CHECK(H % 64 == 0 && IN % 32 == 0, "H % 64, IN % 32 required");
+  TORCH_CHECK(H % 64 == 0 && IN % 64 == 0, "H % 64, IN % 64 required");
   TORCH_CHECK(y.size(1) == H && act.size(1) == IN);
   TORCH_CHECK(pad_offsets.numel() == E + 1);
   launch_moe_down(y.data_ptr(), act.data_ptr(), w.data_ptr(),
