@@ -1,3 +1,11 @@
+"""Diagnostic: empirically probe the fused sampler's ADMITTED set.
+
+Draws many steps at fixed seeds and collects the distinct tokens drawn per
+row — with the Gumbel-max draw, the drawn-set over hundreds of steps equals
+the admitted set. Used to verify exact top-k on hardware (found the
+fp-contract sx inconsistency fixed in sampling.hip); the kernel's optional
+16-float debug buffer prints the refinement walk per row.
+"""
 import sys, os
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
