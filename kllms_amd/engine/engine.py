@@ -317,11 +317,8 @@ class LLMEngine:
         for req in requests:
             ids = req.prompt_ids
             assert len(ids) > 0, "empty prompt"
-            pb = self.prefix_cache.match(ids) if self.prefix_cache is not None else []
-            if pb and len(pb) * self.kv.block_size < min_hit:
-                # a short hit saves less than the tail's decode-mode-prefill
-                # overhead costs — treat as a miss (packed prefill is faster)
-                pb = []
+            pb = (self.prefix_cache.match(ids, min_hit)
+                  if self.prefix_cache is not None else [])
             seq = self._alloc_with_prefix(ids, pb)
             parent_seqs.append(seq)
             matched.append(len(pb) * self.kv.block_size)

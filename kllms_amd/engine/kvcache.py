@@ -202,10 +202,13 @@ class PrefixCache:
             out.append(h)
         return out
 
-    def match(self, prompt_ids: List[int]) -> List[int]:
+    def match(self, prompt_ids: List[int], min_tokens: int = 0) -> List[int]:
         """Longest cached chain of full prompt blocks (always leaves >= 1
-        prompt token un-matched so prefill still produces logits). Returns
-        block ids WITHOUT increfing — the caller owns that step."""
+        prompt token un-matched so prefill still produces logits). A chain
+        shorter than ``min_tokens`` counts as a miss and returns [] — a short
+        cached head saves less prefill than the decode-mode tail forward
+        costs over the packed kernel. Returns block ids WITHOUT increfing —
+        the caller owns that step."""
         bs = self.kv.block_size
         limit = (len(prompt_ids) - 1) // bs
         if limit <= 0:
@@ -217,6 +220,8 @@ class PrefixCache:
                 break
             self._map.move_to_end(d)
             blocks.append(b)
+        if blocks and len(blocks) * bs < min_tokens:
+            blocks = []
         if blocks:
             self.hits += 1
             self.tokens_saved += len(blocks) * bs

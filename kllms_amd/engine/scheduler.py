@@ -197,9 +197,8 @@ class BatchScheduler:
         eng = self.engine
         ids = t.request.prompt_ids
         try:
-            pb = eng.prefix_cache.match(ids) if eng.prefix_cache is not None else []
-            if pb and len(pb) * eng.kv.block_size < eng.config.prefix_cache_min_tokens:
-                pb = []
+            pb = (eng.prefix_cache.match(ids, eng.config.prefix_cache_min_tokens)
+                  if eng.prefix_cache is not None else [])
             seq = eng._alloc_with_prefix(ids, pb)
         except Exception as e:
             t.future.set_exception(e)

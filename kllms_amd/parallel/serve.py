@@ -25,6 +25,13 @@ Design (one decision maker, deterministic replay):
 
 The control plane is gloo (host memory) so tiny action records never touch
 the GPU or interleave with RCCL compute collectives.
+
+CONSTRAINT: at TP>1 every engine-touching call must go through the
+scheduler (the async client's default path). A direct rank-0-only
+``generate()`` — or the opt-in llm string-consensus mode, which issues its
+own create() from inside consolidation — would run TP collectives on rank 0
+alone and deadlock. The default (centroid) consensus mode never does this;
+the embeddings path (engine.embed/embed_dev) is collective-free by design.
 """
 
 from __future__ import annotations
