@@ -96,20 +96,21 @@ def store_kv(
 
 def _prefill_tiles(cu_seqlens: torch.Tensor, device) -> tuple:
     """Grouped q-tile metadata for the MFMA prefill kernel: 32-row q-tiles of
-    each sequence packed 4 per workgroup (the 4 waves share the K/V LDS
+    each sequence packed 8 per workgroup (the 8 waves share the K/V LDS
     staging); idle wave slots padded with -1. One host round-trip per prefill
     batch (once per generate() call)."""
+    NW = 8
     cu = cu_seqlens.cpu().tolist()
     grp_start, grp_len, grp_qpos0 = [], [], []
     for i in range(len(cu) - 1):
         s, e = cu[i], cu[i + 1]
         L = e - s
         tiles = list(range(0, L, 32))
-        for g0 in range(0, len(tiles), 4):
+        for g0 in range(0, len(tiles), NW):
             grp_start.append(s)
             grp_len.append(L)
-            four = tiles[g0 : g0 + 4]
-            grp_qpos0.extend(four + [-1] * (4 - len(four)))
+            grp = tiles[g0 : g0 + NW]
+            grp_qpos0.extend(grp + [-1] * (NW - len(grp)))
     t = lambda x: torch.tensor(x, dtype=torch.int32, device=device)
     return t(grp_start), t(grp_qpos0), t(grp_len)
 

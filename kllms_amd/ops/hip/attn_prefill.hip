@@ -1,9 +1,9 @@
 // Varlen causal prefill attention for gfx950 — flash-style MFMA kernel.
 //
 // Structure (cdna_hip_programming.md Appendix B "Fused attention prefill"):
-//   - a 4-WAVE workgroup serves 4 consecutive 32-row q-tiles of ONE
-//     (sequence, q-head): the K/V LDS staging is shared by the 4 waves, so
-//     its cost is amortized 4x over the 1-wave-per-block version;
+//   - an 8-WAVE workgroup serves 8 consecutive 32-row q-tiles of ONE
+//     (sequence, q-head): the K/V LDS staging is shared by the 8 waves, so
+//     its cost is amortized 8x over the 1-wave-per-block version;
 //   - KV tiles of 32 keys staged in LDS; K XOR-swizzled (G4: row-major tiles
 //     at D=128 are an up-to-16-way ds_read_b128 conflict; byte ^= (row&15)<<4
 //     makes the 16-lane group conflict-free); V staged TRANSPOSED
@@ -32,7 +32,7 @@ typedef __bf16 bf16x8_mfma __attribute__((ext_vector_type(8)));
 #define QBLK 32
 #define KVBLK 64                   // staged keys per barrier round (2 MFMA sub-tiles)
 #define DHEAD 128
-#define NWAVES 4
+#define NWAVES 8
 #define K_ROW_BYTES 256            // 128 bf16
 #define VT_ROW_SHORTS 72           // 64 keys + 8 pad (conflict-free b128 reads)
 
@@ -95,11 +95,10 @@ extern "C" __global__ void __launch_bounds__(64 * NWAVES) attn_prefill_kernel(
   for (int kt00 = 0; kt00 < kv_end_grp; kt00 += KVBLK) {
     const int nkeys_blk = min(KVBLK, kv_end_grp - kt00);
 
-    // ---- stage K (swizzled) + V^T, all 256 threads cooperatively ----------
-#pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
-      const int key = (tid & 31) + kk * 32;   // two keys per thread
-      const int piece = tid >> 5;             // 8 pieces of 16 dims
+    // ---- stage K (swizzled) + V^T, all 512 threads cooperatively ----------
+    {
+      const int key = tid >> 3;               // one key per 8 threads
+      const int piece = tid & 7;              // 16-dim sliver of that key
       const bool valid = key < nkeys_blk;
       const bf16_t* kp = k + (int64_t)(seq_start + kt00 + min(key, nkeys_blk - 1)) * kv_tstride + g_kv * DHEAD;
 #pragma unroll

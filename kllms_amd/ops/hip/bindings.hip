@@ -15,7 +15,7 @@ extern "C" __global__ void store_kv_kernel(const bf16_t*, const bf16_t*, bf16_t*
 extern "C" __global__ void store_kv_fp8_kernel(const bf16_t*, const bf16_t*, unsigned char*, unsigned char*, const int64_t*, int, int, int, int, int);
 extern "C" void launch_attn_decode_partial(float*, const bf16_t*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int, int, int, hipStream_t);
 extern "C" __global__ void attn_decode_reduce_kernel(bf16_t*, const float*, const int*, int, int, int, int);
-extern "C" __global__ void attn_prefill_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, const int*, float, int, int, int, int);  // grouped: [G], [G], [G*4]
+extern "C" __global__ void attn_prefill_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, const int*, float, int, int, int, int);  // grouped: [G], [G], [G*8]
 extern "C" __global__ void sample_kernel(int64_t*, float*, const float*, const float*, const float*, const int*, const int64_t*, const int64_t*, const uint32_t*, int, float*);
 extern "C" __global__ void mfma_selftest_kernel(float*, const bf16_t*, const bf16_t*);
 
@@ -141,8 +141,8 @@ void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k, torch::Te
   TORCH_CHECK(D == 128, "attn_prefill: head_dim must be 128");
   TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share the token stride");
   const int ngroups = grp_seq_start.size(0);
-  TORCH_CHECK(grp_qpos0.size(0) == ngroups * 4, "grp_qpos0 must be [G*4]");
-  hipLaunchKernelGGL(attn_prefill_kernel, dim3(ngroups, H), dim3(256), 0, cur_stream(),
+  TORCH_CHECK(grp_qpos0.size(0) == ngroups * 8, "grp_qpos0 must be [G*8]");
+  hipLaunchKernelGGL(attn_prefill_kernel, dim3(ngroups, H), dim3(512), 0, cur_stream(),
                      bf(out), cbf(q), cbf(k), cbf(v),
                      grp_seq_start.data_ptr<int>(), grp_seqlen.data_ptr<int>(),
                      grp_qpos0.data_ptr<int>(), (float)scale, H, KVH,
