@@ -71,6 +71,12 @@ MODEL_PRESETS: dict[str, ModelArchConfig] = {
         num_layers=2, num_heads=4, num_kv_heads=2, rope_theta=10000.0,
         max_position_embeddings=512,
     ),
+    # tiny variant with 4 KV heads so gloo TP=4 tests divide evenly
+    "tiny-llama-kv4": ModelArchConfig(
+        arch="llama", vocab_size=512, hidden_size=128, intermediate_size=256,
+        num_layers=2, num_heads=8, num_kv_heads=4, rope_theta=10000.0,
+        max_position_embeddings=512,
+    ),
     "tiny-mixtral": ModelArchConfig(
         arch="mixtral", vocab_size=512, hidden_size=64, intermediate_size=128,
         num_layers=2, num_heads=4, num_kv_heads=2, rope_theta=10000.0,

@@ -15,8 +15,9 @@ import pytest
 PORT = 29791
 
 
-def _mk_engine_kwargs(tp):
-    return dict(model="tiny-llama", tp_size=tp, max_kv_blocks=512,
+def _mk_engine_kwargs(tp, world=None):
+    model = "tiny-llama-kv4" if (world or tp) >= 4 else "tiny-llama"
+    return dict(model=model, tp_size=tp, max_kv_blocks=512,
                 use_hip_graphs=False, device="cpu", seed=0,
                 default_max_new_tokens=8, max_seq_len=512,
                 prefill_chunk_tokens=16)
@@ -26,7 +27,7 @@ def _serving_worker(rank: int, world_size: int, q):
     import torch.distributed as dist
 
     os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = str(PORT)
+    os.environ["MASTER_PORT"] = str(PORT + world_size)
     dist.init_process_group("gloo", rank=rank, world_size=world_size)
     try:
         from kllms_amd.engine.api import LocalEngineClient
@@ -75,22 +76,24 @@ def _serving_worker(rank: int, world_size: int, q):
         dist.destroy_process_group()
 
 
-@pytest.mark.timeout(300)
-def test_tp2_scheduler_serving_matches_tp1():
+@pytest.mark.timeout(600)
+@pytest.mark.parametrize("world", [2, 4])
+def test_tp_scheduler_serving_matches_tp1(world):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_serving_worker, args=(r, 2, q)) for r in range(2)]
+    procs = [ctx.Process(target=_serving_worker, args=(r, world, q)) for r in range(world)]
     for p in procs:
         p.start()
     results = {}
-    for _ in range(2):
-        rank, payload = q.get(timeout=240)
+    for _ in range(world):
+        rank, payload = q.get(timeout=480)
         results[rank] = payload
     for p in procs:
         p.join(timeout=60)
         assert p.exitcode == 0
 
-    assert results[1] == "follower-done"
+    for r in range(1, world):
+        assert results[r] == "follower-done"
     tp2 = results[0]
     assert len(tp2["r1"]) == 3 and len(tp2["r2"]) == 2 and len(tp2["r3"]) == 1
     # greedy n=3: all streams identical
@@ -104,7 +107,7 @@ def test_tp2_scheduler_serving_matches_tp1():
     class Answer(BaseModel):
         value: int
 
-    client = LocalEngineClient(**_mk_engine_kwargs(1))
+    client = LocalEngineClient(**_mk_engine_kwargs(1, world=world))
     r1 = client.chat_completions_create(True,
         messages=[{"role": "user", "content": "count to three"}],
         n=3, temperature=0.0, max_tokens=6, seed=5)
@@ -119,3 +122,57 @@ def test_tp2_scheduler_serving_matches_tp1():
     assert tp2["r1"] == [c.message.content for c in r1.choices]
     assert tp2["r2"] == [c.message.content for c in r2.choices]
     assert tp2["r3"] == [c.message.content for c in r3.choices]
+
+
+class TestCollectiveDispatch:
+    """CPU-testable logic of the size-switched custom collective."""
+
+    def test_should_use_gates(self):
+        import torch
+
+        from kllms_amd.parallel.collective import CustomAllReduce
+
+        car = CustomAllReduce.__new__(CustomAllReduce)   # no IPC init on CPU
+        car.max_bytes = 4 << 20
+        ok = torch.zeros(4096, dtype=torch.bfloat16)
+        assert car.should_use(ok)
+        assert not car.should_use(ok.float())                      # dtype
+        assert not car.should_use(torch.zeros(4097, dtype=torch.bfloat16)[:4095])  # numel%8
+        assert not car.should_use(torch.zeros(3 << 20, dtype=torch.bfloat16))      # too big
+        t = torch.zeros(16, 16, dtype=torch.bfloat16).t()
+        assert not car.should_use(t)                               # non-contiguous
+
+    def test_env_disable(self, monkeypatch):
+        import torch
+
+        from kllms_amd.parallel.collective import maybe_init_custom_allreduce
+        from kllms_amd.parallel.tp import ParallelContext
+
+        monkeypatch.setenv("KLLMS_CUSTOM_AR", "0")
+        ctx = ParallelContext(world_size=2, rank=0)
+        assert maybe_init_custom_allreduce(ctx, torch.device("cpu")) is None
+
+    def test_serialize_roundtrip_constraint(self):
+        """GenRequest (de)serialization across the TP control plane keeps
+        constrained-decoding state (schema + whitespace flag)."""
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.constrained import JsonSchemaConstraint
+        from kllms_amd.engine.engine import GenRequest, LLMEngine
+        from kllms_amd.engine.sampling import SamplingParams
+        from kllms_amd.parallel.serve import _deserialize_request, _serialize_request
+
+        eng = LLMEngine(EngineConfig(model="tiny-llama", max_kv_blocks=64,
+                                     use_hip_graphs=False, device="cpu"))
+        schema = {"type": "object", "properties": {"v": {"type": "integer"}}, "required": ["v"]}
+        c = JsonSchemaConstraint(schema, eng.tokenizer, whitespace=True)
+        req = GenRequest(prompt_ids=[1, 2, 3], n=2,
+                         sampling=SamplingParams(temperature=0.5, max_tokens=7, seed=3),
+                         constraint=c)
+        d = _serialize_request(req)
+        import pickle
+        d = pickle.loads(pickle.dumps(d))    # the broadcast pickles it
+        req2 = _deserialize_request(d, eng)
+        assert req2.prompt_ids == req.prompt_ids and req2.n == 2
+        assert req2.sampling.model_dump() == req.sampling.model_dump()
+        assert req2.constraint.whitespace is True
+        assert (req2.constraint.next_state == c.next_state).all()
