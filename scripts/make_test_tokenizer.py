@@ -44,6 +44,34 @@ SPECIALS = [
 ]
 
 
+def make_sp(out_path: str) -> None:
+    """SentencePiece-style fixture: Unigram + Metaspace (real ▁ markers) +
+    the full <0xNN> byte-fallback special block — the OTHER vocab format
+    HFTokenizer.token_bytes must decode (Llama-2/Mistral tokenizer.json
+    shape)."""
+    from tokenizers import decoders, models, pre_tokenizers, trainers
+
+    tok = Tokenizer(models.Unigram())
+    tok.pre_tokenizer = pre_tokenizers.Metaspace()
+    tok.decoder = decoders.Metaspace()
+    byte_fb = [f"<0x{i:02X}>" for i in range(256)]
+    trainer = trainers.UnigramTrainer(
+        vocab_size=1200, special_tokens=["<unk>", "<s>", "</s>"],
+        unk_token="<unk>", show_progress=False,
+        initial_alphabet=list("abcdefghijklmnopqrstuvwxyzABCDEFGHIJKLMNOPQRSTUVWXYZ0123456789{}[]:,.\"'+-_ "),
+    )
+    tok.train_from_iterator(CORPUS, trainer=trainer)
+    # byte-fallback pieces are ORDINARY vocab entries in real SP tokenizers
+    # (Llama-2 shape), not special added tokens
+    tok.add_tokens(byte_fb)
+    tok.save(out_path)
+    t2 = Tokenizer.from_file(out_path)
+    v = t2.get_vocab()
+    assert any(k.startswith("\u2581") or k.startswith("▁") for k in v), "no metaspace markers"
+    assert "<0x41>" in v
+    print(f"wrote {out_path}: vocab={t2.get_vocab_size()}")
+
+
 def main(out_path: str) -> None:
     tok = Tokenizer(models.BPE(unk_token=None))
     tok.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
@@ -69,7 +97,6 @@ def main(out_path: str) -> None:
 
 
 if __name__ == "__main__":
-    out = sys.argv[1] if len(sys.argv) > 1 else os.path.join(
-        os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "tests", "data", "bpe_tokenizer.json"
-    )
-    main(out)
+    data = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "tests", "data")
+    main(sys.argv[1] if len(sys.argv) > 1 else os.path.join(data, "bpe_tokenizer.json"))
+    make_sp(os.path.join(data, "sp_tokenizer.json"))
