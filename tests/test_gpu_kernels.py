@@ -660,3 +660,33 @@ class TestLevenshteinKernel:
         got = simmod.string_similarity("Paris", "paris!", "embeddings", None)
         want = simmod.levenshtein_similarity("Paris", "paris!")
         assert got == pytest.approx(want, abs=1e-9)
+
+
+class TestPrefixCacheGPU:
+    def test_shared_prefix_parity_on_gpu(self):
+        """Prefix-cache hits must be output-invisible with the HIP kernels:
+        the cached-head + decode-mode-tail path must reproduce the uncached
+        packed-prefill greedy decode exactly."""
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.engine import GenRequest, LLMEngine
+        from kllms_amd.engine.sampling import SamplingParams
+
+        shared = list(range(2, 300))   # 298 tokens >= prefix_cache_min_tokens=128
+        mk = lambda cache: LLMEngine(EngineConfig(
+            model="mid-llama", max_kv_blocks=1024, use_hip_graphs=True,
+            device=DEV, seed=0, default_max_new_tokens=12,
+            enable_prefix_caching=cache))
+        greedy = SamplingParams(temperature=0.0, max_tokens=12)
+
+        eng_c = mk(True)
+        eng_c.generate([GenRequest(prompt_ids=shared + [333], n=1, sampling=greedy)])
+        out_hit = eng_c.generate([GenRequest(prompt_ids=shared + [444, 445], n=2, sampling=greedy)])[0]
+        assert eng_c.prefix_cache.hits >= 1
+        assert eng_c.prefix_cache.tokens_saved >= 128
+
+        eng_u = mk(False)
+        eng_u.generate([GenRequest(prompt_ids=shared + [333], n=1, sampling=greedy)])
+        out_ref = eng_u.generate([GenRequest(prompt_ids=shared + [444, 445], n=2, sampling=greedy)])[0]
+
+        for a, b in zip(out_hit.streams, out_ref.streams):
+            assert a.token_ids == b.token_ids
