@@ -169,3 +169,93 @@ def test_tp2_engine_one_gpu_custom_ar_matches_tp1():
                                    sampling=SamplingParams(temperature=0.0, max_tokens=12))])[0]
     assert results[0][0] == out.streams[0].token_ids, (
         "TP=2 (custom one-shot AR) greedy decode != TP=1")
+
+
+def _tp_serving_worker(rank: int, world: int, q):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(PORT + 2)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.cuda.set_device(0)
+        from kllms_amd.engine.api import LocalEngineClient
+        from kllms_amd.parallel.collective import maybe_init_custom_allreduce
+        from kllms_amd.parallel.serve import TPCoordinator, TPFollower
+        from kllms_amd.parallel.tp import ParallelContext
+
+        ctx = ParallelContext(world_size=world, rank=rank)
+        maybe_init_custom_allreduce(ctx, torch.device("cuda:0"))
+        assert ctx.custom_ar is not None
+
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(
+            EngineConfig(model="mid-llama", tp_size=world, max_kv_blocks=1024,
+                         use_hip_graphs=True, device="cuda:0", seed=0,
+                         default_max_new_tokens=10),
+            parallel_ctx=ctx,
+        )
+        client = LocalEngineClient(llm_engine=eng)
+
+        if rank != 0:
+            TPFollower(eng).run()
+            q.put((rank, "follower-done", ctx.custom_ar.calls))
+            return
+
+        sched = client.scheduler
+        sched.coordinator = TPCoordinator(eng)
+
+        import concurrent.futures as cf
+
+        with cf.ThreadPoolExecutor(3) as pool:
+            futs = [pool.submit(client.chat_completions_create, True,
+                                messages=[{"role": "user", "content": f"request {i}"}],
+                                n=3, temperature=0.0, max_tokens=8, seed=i)
+                    for i in range(3)]
+            res = [f.result(180) for f in futs]
+        sched.shutdown()
+        sched.coordinator.stop()
+        q.put((rank, [[c.message.content for c in r.choices] for r in res], ctx.custom_ar.calls))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_tp2_serving_stack_on_gpu():
+    """The WHOLE TP serving stack on hardware: public client -> scheduler ->
+    rank-0 action broadcast -> follower replay -> TP layers with the custom
+    IPC one-shot all-reduce under hipGraph-captured decode (2 ranks, 1 GPU)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tp_serving_worker, args=(r, 2, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, payload, ar_calls = q.get(timeout=480)
+        results[rank] = (payload, ar_calls)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+
+    assert results[1][0] == "follower-done"
+    assert results[0][1] > 0 and results[1][1] > 0, "custom AR unused"
+    outs = results[0][0]
+    assert len(outs) == 3
+    for choices in outs:
+        assert len(choices) == 3
+        assert choices[0] == choices[1] == choices[2]  # greedy n=3
+
+    # TP=1 single-process reference
+    from kllms_amd.engine.api import LocalEngineClient
+
+    client = LocalEngineClient(model="mid-llama", max_kv_blocks=1024,
+                               use_hip_graphs=True, device="cuda:0", seed=0,
+                               default_max_new_tokens=10)
+    for i, choices in enumerate(outs):
+        r = client.chat_completions_create(
+            False, messages=[{"role": "user", "content": f"request {i}"}],
+            n=3, temperature=0.0, max_tokens=8, seed=i)
+        assert choices == [c.message.content for c in r.choices]
