@@ -285,7 +285,12 @@ def main():
             torch.cuda.synchronize()
 
     async def timed_run():
-        # ONE event loop (and one to_thread pool) for the whole bench
+        # ONE event loop (and one to_thread pool) for the whole bench;
+        # asyncio's default executor caps at 32 threads, which would split
+        # a >32-request burst into two admission waves — size it to the batch
+        import concurrent.futures as _cf
+        asyncio.get_running_loop().set_default_executor(
+            _cf.ThreadPoolExecutor(max_workers=batch + 8))
         for w in range(args.warmup):
             log(f"[bench] warmup {w + 1}/{args.warmup}")
             await run_step(-1 - w)
