@@ -144,6 +144,13 @@ class EngineConfig(BaseModel):
         import os
 
         cfg_path = os.path.join(self.weights_path or self.model, "config.json")
+        if not os.path.exists(cfg_path):
+            # OpenAI-style model-not-found error instead of a raw ENOENT
+            raise ValueError(
+                f"model {self.model!r} is neither a preset "
+                f"({', '.join(sorted(MODEL_PRESETS))}) nor a directory with a "
+                f"config.json (model_not_found)"
+            )
         with open(cfg_path) as f:
             hf = json.load(f)
         arch = "mixtral" if "mixtral" in hf.get("model_type", "").lower() or hf.get("num_local_experts") else "llama"

@@ -268,3 +268,11 @@ class TestReferenceScenarios:
         embs = asyncio.run(run())
         assert len(embs) == 3
         assert embs[0] == embs[1] != embs[2]
+
+
+def test_invalid_model_raises_model_not_found():
+    from kllms_amd import KLLMs
+
+    with pytest.raises(ValueError, match="model_not_found"):
+        KLLMs(model="no-such-model", device="cpu", max_kv_blocks=32).chat.completions.create(
+            model="no-such-model", messages=[{"role": "user", "content": "hi"}])
