@@ -109,6 +109,8 @@ class LocalEngineClient:
             presence_penalty=call_params.get("presence_penalty", 0.0),
             logprobs=bool(call_params.get("logprobs", False)),
             top_logprobs=int(call_params.get("top_logprobs") or 0),
+            logit_bias={int(k): float(v) for k, v in call_params["logit_bias"].items()}
+            if call_params.get("logit_bias") else None,
         )
         if sampling.top_logprobs and not sampling.logprobs:
             raise ValueError("top_logprobs requires logprobs=True")
@@ -233,6 +235,8 @@ class LocalEngineClient:
                 seed=call_params.get("seed"),
                 logprobs=bool(call_params.get("logprobs", False)),
                 top_logprobs=int(call_params.get("top_logprobs") or 0),
+                logit_bias={int(k): float(v) for k, v in call_params["logit_bias"].items()}
+                if call_params.get("logit_bias") else None,
             )
             constraint = None
             rf = call_params.get("response_format")

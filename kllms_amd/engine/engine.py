@@ -148,6 +148,8 @@ class LLMEngine:
 
         # monotonic per-request counter for unseeded-request RNG derivation
         self._seed_counter = 0
+        # device-tensor cache for per-request logit_bias dicts (keyed by id)
+        self._logit_bias_cache: dict = {}
         self.dtype = {"bfloat16": torch.bfloat16, "float16": torch.float16, "float32": torch.float32}[config.dtype]
         # CPU bf16 matmuls are slow and torch CPU attention paths prefer f32
         if self.device.type == "cpu" and self.dtype == torch.bfloat16:
@@ -245,19 +247,32 @@ class LLMEngine:
         return mask.to(self.device)
 
     def _apply_penalties(self, logits: torch.Tensor, streams: List[_Stream]) -> torch.Tensor:
-        """frequency/presence penalties (OpenAI semantics), applied only when
-        a stream requests them (rare; host-side composition)."""
-        if not any(s.sampling.frequency_penalty or s.sampling.presence_penalty for s in streams):
+        """frequency/presence penalties and logit_bias (OpenAI semantics),
+        applied only when a stream requests them (rare; host-side
+        composition)."""
+        if not any(s.sampling.frequency_penalty or s.sampling.presence_penalty
+                   or s.sampling.logit_bias for s in streams):
             return logits
         for i, s in enumerate(streams):
             fp, pp = s.sampling.frequency_penalty, s.sampling.presence_penalty
-            if not (fp or pp):
-                continue
-            if not s.out.token_ids:
-                continue
-            ids = torch.tensor(s.out.token_ids, device=logits.device)
-            counts = torch.bincount(ids, minlength=logits.shape[1]).to(logits.dtype)
-            logits[i] -= fp * counts + pp * (counts > 0).to(logits.dtype)
+            if (fp or pp) and s.out.token_ids:
+                ids = torch.tensor(s.out.token_ids, device=logits.device)
+                counts = torch.bincount(ids, minlength=logits.shape[1]).to(logits.dtype)
+                logits[i] -= fp * counts + pp * (counts > 0).to(logits.dtype)
+            lb = s.sampling.logit_bias
+            if lb:
+                key = id(lb)
+                cached = self._logit_bias_cache.get(key)
+                if cached is None:
+                    idx = torch.tensor([int(t) for t in lb], dtype=torch.long,
+                                       device=logits.device)
+                    val = torch.tensor([float(v) for v in lb.values()],
+                                       dtype=logits.dtype, device=logits.device)
+                    cached = (idx, val)
+                    self._logit_bias_cache[key] = cached
+                    if len(self._logit_bias_cache) > 256:
+                        self._logit_bias_cache.pop(next(iter(self._logit_bias_cache)))
+                logits[i, cached[0]] += cached[1]
         return logits
 
     # --- main entry -----------------------------------------------------------

@@ -4,7 +4,7 @@ k_llms/resources/completions/completions.py:42-64)."""
 
 from __future__ import annotations
 
-from typing import List, Optional, Union
+from typing import Dict, List, Optional, Union
 
 from pydantic import BaseModel
 
@@ -18,6 +18,10 @@ class SamplingParams(BaseModel):
     seed: Optional[int] = None
     frequency_penalty: float = 0.0
     presence_penalty: float = 0.0
+    # OpenAI `logit_bias`: {token_id: bias in [-100, 100]} added to the
+    # logits before sampling (the reference forwards it to the remote API;
+    # served natively here)
+    logit_bias: Optional[Dict[int, float]] = None
     logprobs: bool = False
     # number of top alternative tokens to report per position (OpenAI
     # `top_logprobs`, 0 = off; requires logprobs=True at the API layer)

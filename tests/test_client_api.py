@@ -276,3 +276,34 @@ def test_invalid_model_raises_model_not_found():
     with pytest.raises(ValueError, match="model_not_found"):
         KLLMs(model="no-such-model", device="cpu", max_kv_blocks=32).chat.completions.create(
             model="no-such-model", messages=[{"role": "user", "content": "hi"}])
+
+
+def test_logit_bias_served_natively():
+    """OpenAI `logit_bias` (the reference forwards it to the remote API):
+    +100 on a token forces it; -100 bans it."""
+    from kllms_amd import KLLMs
+
+    k = KLLMs(model="tiny-llama", device="cpu", max_kv_blocks=256,
+              use_hip_graphs=False, default_max_new_tokens=6)
+    force = 77
+    r = k.chat.completions.create(
+        model="tiny-llama", messages=[{"role": "user", "content": "hello"}],
+        n=2, temperature=1.0, max_tokens=5, seed=3,
+        logit_bias={str(force): 100.0})
+    eng = k.client.engine
+    # +100 dominates every random-init logit: all sampled tokens == force
+    ids = eng.tokenizer.encode(r.choices[1].message.content)
+    assert all(i == force for i in ids), ids
+
+    # ban: the greedy top token (raw stream ids, not a decode round-trip)
+    from kllms_amd.engine.engine import GenRequest
+    from kllms_amd.engine.sampling import SamplingParams
+
+    ids = eng.tokenizer.encode("hello")
+    base = eng.generate([GenRequest(prompt_ids=ids, n=1,
+                                    sampling=SamplingParams(temperature=0.0, max_tokens=3))])[0]
+    top = base.streams[0].token_ids[0]
+    banned = eng.generate([GenRequest(prompt_ids=ids, n=1,
+                                      sampling=SamplingParams(temperature=0.0, max_tokens=3,
+                                                              logit_bias={top: -100.0}))])[0]
+    assert banned.streams[0].token_ids[0] != top
