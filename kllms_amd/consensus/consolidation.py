@@ -122,6 +122,41 @@ def _build_consensus_choice(base_choice: Choice, content_str: str) -> Choice:
     )
 
 
+def consolidate_tool_call_completions(
+    completion: ChatCompletion,
+    get_openai_embeddings_from_text: SYNC_GET_EMBEDDINGS_FROM_TEXT_TYPE,
+    client: Any = None,
+    consensus_settings: ConsensusSettings = None,  # type: ignore[assignment]
+    aligner: str = "similarity",
+) -> KLLMsChatCompletion:
+    """Consensus over FORCED tool-call responses (beyond-reference: the
+    reference forwards `tools` to the remote API and its consolidator then
+    crashes on the all-None contents — tool arguments are JSON, which is
+    exactly what the consensus machinery consumes). The n choices' argument
+    strings run through the standard consolidation; every resulting choice
+    is re-wrapped as a tool call of the same function, with the CONSENSUS
+    arguments on choice 0 and `likelihoods` mirroring the argument schema."""
+    import uuid as _uuid
+
+    from ..types.openai_compat import ChatCompletionMessageToolCall, Function
+
+    name = completion.choices[0].message.tool_calls[0].function.name
+    shadow = completion.model_copy(deep=True)
+    for c in shadow.choices:
+        c.message.content = c.message.tool_calls[0].function.arguments
+        c.message.tool_calls = None
+    out = consolidate_chat_completions(
+        shadow, get_openai_embeddings_from_text, client, consensus_settings, aligner
+    )
+    orig_calls = [c.message.tool_calls[0] for c in completion.choices]
+    for i, c in enumerate(out.choices):
+        call_id = f"call_{_uuid.uuid4().hex[:24]}" if i == 0 else orig_calls[i - 1].id
+        c.message.tool_calls = [ChatCompletionMessageToolCall(
+            id=call_id, function=Function(name=name, arguments=c.message.content or ""))]
+        c.message.content = None
+    return out
+
+
 def consolidate_chat_completions(
     completions: Union[List[ChatCompletion], ChatCompletion],
     get_openai_embeddings_from_text: SYNC_GET_EMBEDDINGS_FROM_TEXT_TYPE,

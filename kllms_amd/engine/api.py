@@ -118,9 +118,12 @@ class LocalEngineClient:
             raise ValueError("top_logprobs must be between 0 and 20")
 
         constraint = None
+        tool_name = None
         response_format = call_params.get("response_format")
         if response_format is not None:
             constraint = self._build_constraint(response_format, constrained)
+        elif call_params.get("tools"):
+            constraint, tool_name = self._build_tool_constraint(call_params)
 
         eng = self.engine
         prompt = eng.tokenizer.apply_chat_template(messages)
@@ -133,7 +136,7 @@ class LocalEngineClient:
         else:
             with self._engine_lock:
                 out = eng.generate([req])[0]
-        return out, model, sampling
+        return out, model, sampling, tool_name
 
     def _fit_context(self, prompt_ids: List[int], sampling: SamplingParams) -> None:
         """Error on over-long prompts (never truncate — that would cut the
@@ -148,6 +151,39 @@ class LocalEngineClient:
         room = max_seq - len(prompt_ids)
         want = sampling.max_tokens or self.config.default_max_new_tokens
         sampling.max_tokens = min(want, room)
+
+    def _build_tool_constraint(self, call_params: Dict[str, Any]):
+        """FORCED tool calls served natively: with a specific function forced
+        via tool_choice (or tool_choice="required" and exactly one tool), the
+        function's JSON-schema parameters compile into the decoding DFA and
+        the output is returned as an OpenAI tool_calls message. tool_choice
+        "auto" is NOT model-decided locally — it generates plain content
+        (documented limitation; the reference delegates the decision to the
+        remote model)."""
+        from .constrained import JsonSchemaConstraint
+
+        tools = call_params.get("tools") or []
+        choice = call_params.get("tool_choice")
+        fn = None
+        if isinstance(choice, dict) and choice.get("type") == "function":
+            name = (choice.get("function") or {}).get("name")
+            for t in tools:
+                f = t.get("function", {})
+                if f.get("name") == name:
+                    fn = f
+                    break
+            if fn is None:
+                raise ValueError(f"tool_choice names unknown function {name!r}")
+        elif choice == "required" and len(tools) == 1:
+            fn = tools[0].get("function", {})
+        if fn is None:
+            return None, None
+        schema = fn.get("parameters") or {"type": "object"}
+        return (
+            JsonSchemaConstraint(schema, self.engine.tokenizer,
+                                 whitespace=getattr(self.config, "constrained_whitespace", False)),
+            fn.get("name"),
+        )
 
     def _build_constraint(self, response_format: Any, constrained: bool):
         from .constrained import JsonSchemaConstraint
@@ -272,14 +308,27 @@ class LocalEngineClient:
         return results
 
     def chat_completions_create(self, _scheduled: bool = False, **call_params: Any) -> ChatCompletion:
-        out, model, sampling = self._generate(call_params, constrained=False, scheduled=_scheduled)
+        out, model, sampling, tool_name = self._generate(call_params, constrained=False, scheduled=_scheduled)
         choices = []
         for i, s in enumerate(out.streams):
+            if tool_name is not None:
+                from ..types.openai_compat import ChatCompletionMessageToolCall, Function
+
+                msg = ChatCompletionMessage(
+                    role="assistant", content=None,
+                    tool_calls=[ChatCompletionMessageToolCall(
+                        id=f"call_{uuid.uuid4().hex[:24]}",
+                        function=Function(name=tool_name, arguments=s.text))],
+                )
+                fr = "tool_calls" if s.finish_reason == "stop" else s.finish_reason
+            else:
+                msg = ChatCompletionMessage(role="assistant", content=s.text)
+                fr = s.finish_reason
             choices.append(
                 Choice(
-                    finish_reason=s.finish_reason,
+                    finish_reason=fr,
                     index=i,
-                    message=ChatCompletionMessage(role="assistant", content=s.text),
+                    message=msg,
                     logprobs=self._mk_logprobs(s) if sampling.logprobs else None,
                 )
             )
@@ -296,7 +345,7 @@ class LocalEngineClient:
         import json
 
         response_format = call_params.get("response_format")
-        out, model, sampling = self._generate(call_params, constrained=True, scheduled=_scheduled)
+        out, model, sampling, _ = self._generate(call_params, constrained=True, scheduled=_scheduled)
         choices = []
         for i, s in enumerate(out.streams):
             parsed = None

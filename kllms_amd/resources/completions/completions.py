@@ -101,6 +101,15 @@ class Completions:
             call_params["n"] = n
         # ONE engine call: shared prefill, n decode streams fanned out.
         completion = self._wrapper.client.chat.completions.create(**call_params)
+        if completion.choices and all(c.message.tool_calls for c in completion.choices):
+            from ...consensus.consolidation import consolidate_tool_call_completions
+
+            return consolidate_tool_call_completions(
+                completion,
+                embeddings_wrapper,
+                client=self._wrapper.client,
+                aligner=getattr(self._wrapper, "consensus_aligner", "similarity"),
+            )
         return consolidate_chat_completions(
             completion,
             embeddings_wrapper,
@@ -180,6 +189,21 @@ class AsyncCompletions:
         if n and n > 1:
             call_params["n"] = n
         completion = await self._wrapper.client.chat.completions.acreate(**call_params)
+        if completion.choices and all(c.message.tool_calls for c in completion.choices):
+            import asyncio as _asyncio
+
+            from ...consensus.consolidation import consolidate_tool_call_completions
+            from ...consensus.aio import _bridge_async_embeddings
+
+            loop = _asyncio.get_running_loop()
+            return await _asyncio.to_thread(
+                consolidate_tool_call_completions,
+                completion,
+                _bridge_async_embeddings(embeddings_wrapper, loop),
+                self._wrapper.client,
+                None,
+                getattr(self._wrapper, "consensus_aligner", "similarity"),
+            )
         return await async_consolidate_chat_completions(
             completion,
             embeddings_wrapper,

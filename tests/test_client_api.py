@@ -307,3 +307,96 @@ def test_logit_bias_served_natively():
                                       sampling=SamplingParams(temperature=0.0, max_tokens=3,
                                                               logit_bias={top: -100.0}))])[0]
     assert banned.streams[0].token_ids[0] != top
+
+
+class TestToolCalls:
+    """FORCED function calls served natively: the function's parameters
+    schema compiles into the decoding DFA; the consensus layer treats the
+    result exactly as it treats the reference's remote tool-call responses
+    (consensus choice borrows tool_calls from choice 0)."""
+
+    TOOLS = [{"type": "function", "function": {
+        "name": "get_weather",
+        "parameters": {"type": "object",
+                       "properties": {"city": {"type": "string", "maxLength": 10},
+                                      "days": {"type": "integer", "minimum": 0, "maximum": 9}},
+                       "required": ["city", "days"]}}}]
+
+    def _client(self):
+        from kllms_amd.engine.api import LocalEngineClient
+
+        return LocalEngineClient(model="tiny-llama", device="cpu", max_kv_blocks=256,
+                                 use_hip_graphs=False, default_max_new_tokens=48)
+
+    def test_forced_tool_call_schema_valid(self):
+        import json as _json
+
+        client = self._client()
+        r = client.chat.completions.create(
+            model="tiny-llama", messages=[{"role": "user", "content": "weather in Paris?"}],
+            tools=self.TOOLS, tool_choice={"type": "function", "function": {"name": "get_weather"}},
+            n=2, temperature=0.8, max_tokens=48, seed=4)
+        for ch in r.choices:
+            assert ch.message.content is None
+            tc = ch.message.tool_calls
+            assert tc and tc[0].function.name == "get_weather"
+            if ch.finish_reason == "tool_calls":
+                args = _json.loads(tc[0].function.arguments)
+                assert set(args) == {"city", "days"} and 0 <= args["days"] <= 9
+
+    def test_required_single_tool(self):
+        client = self._client()
+        r = client.chat.completions.create(
+            model="tiny-llama", messages=[{"role": "user", "content": "go"}],
+            tools=self.TOOLS, tool_choice="required", n=1, temperature=0.0, max_tokens=48)
+        assert r.choices[0].message.tool_calls
+
+    def test_unknown_function_raises(self):
+        client = self._client()
+        with pytest.raises(ValueError, match="unknown function"):
+            client.chat.completions.create(
+                model="tiny-llama", messages=[{"role": "user", "content": "x"}],
+                tools=self.TOOLS, tool_choice={"type": "function", "function": {"name": "nope"}})
+
+    def test_auto_falls_back_to_content(self):
+        client = self._client()
+        r = client.chat.completions.create(
+            model="tiny-llama", messages=[{"role": "user", "content": "hi"}],
+            tools=self.TOOLS, tool_choice="auto", n=1, temperature=0.0, max_tokens=6)
+        assert r.choices[0].message.tool_calls is None
+        assert r.choices[0].message.content is not None
+
+    def test_consensus_layer_over_tool_calls(self):
+        """KLLMs().create with forced tools: n+1 choices, consensus choice 0
+        borrows tool_calls from original choice 0 (reference semantics)."""
+        from kllms_amd import KLLMs
+
+        k = KLLMs(model="tiny-llama", device="cpu", max_kv_blocks=256,
+                  use_hip_graphs=False, default_max_new_tokens=48)
+        r = k.chat.completions.create(
+            model="tiny-llama", messages=[{"role": "user", "content": "weather?"}],
+            tools=self.TOOLS, tool_choice={"type": "function", "function": {"name": "get_weather"}},
+            n=3, temperature=0.9, max_tokens=48, seed=9)
+        assert len(r.choices) == 4
+        assert r.choices[0].message.tool_calls is not None
+        assert r.choices[0].message.tool_calls[0].function.name == "get_weather"
+
+
+def test_async_tool_call_consensus():
+    import asyncio
+
+    from kllms_amd import AsyncKLLMs
+
+    async def run():
+        k = AsyncKLLMs(model="tiny-llama", device="cpu", max_kv_blocks=256,
+                       use_hip_graphs=False, default_max_new_tokens=48)
+        r = await k.chat.completions.create(
+            model="tiny-llama", messages=[{"role": "user", "content": "weather?"}],
+            tools=TestToolCalls.TOOLS,
+            tool_choice={"type": "function", "function": {"name": "get_weather"}},
+            n=3, temperature=0.9, max_tokens=48, seed=2)
+        assert len(r.choices) == 4
+        assert r.choices[0].message.tool_calls[0].function.name == "get_weather"
+        k.close()
+
+    asyncio.run(run())
