@@ -88,10 +88,26 @@ class PagedKVCache:
         self.v_all = torch.zeros_like(self.k_all)
         self.k_caches = [self.k_all[i] for i in range(num_layers)]
         self.v_caches = [self.v_all[i] for i in range(num_layers)]
+        # fp8 caches carry per-(token, head)-ROW dequant scales (s = amax/448
+        # at store time): outlier rows in real checkpoints would otherwise
+        # saturate e4m3's +-448 under a static scale. fp32 [NB, KVH, BS] per
+        # layer per cache = 1/32 of the fp8 payload; init 1.0 (neutral for
+        # never-written slots). bf16 caches carry no scales (None).
+        self.fp8 = dtype == torch.float8_e4m3fn
+        if self.fp8:
+            self.k_scale_all = torch.ones(num_layers, num_blocks, num_kv_heads, block_size,
+                                          device=device, dtype=torch.float32)
+            self.v_scale_all = torch.ones_like(self.k_scale_all)
+        else:
+            self.k_scale_all = self.v_scale_all = None
         self.allocator = BlockAllocator(num_blocks)
 
-    def layer_caches(self) -> List[Tuple[torch.Tensor, torch.Tensor]]:
-        return list(zip(self.k_caches, self.v_caches))
+    def layer_caches(self) -> List[Tuple[torch.Tensor, torch.Tensor, Optional[torch.Tensor], Optional[torch.Tensor]]]:
+        """Per layer (k, v, k_scale, v_scale); scales are None for bf16."""
+        if self.fp8:
+            return [(self.k_caches[i], self.v_caches[i], self.k_scale_all[i], self.v_scale_all[i])
+                    for i in range(self.num_layers)]
+        return [(self.k_caches[i], self.v_caches[i], None, None) for i in range(self.num_layers)]
 
     # --- sequence-level operations ------------------------------------------
     def alloc_sequence(self, num_tokens: int) -> SequenceKV:
@@ -130,6 +146,9 @@ class PagedKVCache:
         dst = self.allocator.alloc()
         self.k_all[:, dst].copy_(self.k_all[:, src])
         self.v_all[:, dst].copy_(self.v_all[:, src])
+        if self.fp8:
+            self.k_scale_all[:, dst].copy_(self.k_scale_all[:, src])
+            self.v_scale_all[:, dst].copy_(self.v_scale_all[:, src])
         return dst
 
     def append_slot(self, seq: SequenceKV) -> int:

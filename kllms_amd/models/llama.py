@@ -34,7 +34,7 @@ class ForwardBatch:
     mode: Literal["prefill", "decode"]
     positions: torch.Tensor              # [T] long
     slot_mapping: torch.Tensor           # [T] long — flat KV slots to write
-    kv_caches: List[Tuple[torch.Tensor, torch.Tensor]]  # per layer (k, v)
+    kv_caches: List[Tuple]  # per layer (k, v, k_scale|None, v_scale|None)
     cu_seqlens: Optional[torch.Tensor] = None   # [n_seq+1] int32 (prefill)
     block_tables: Optional[torch.Tensor] = None  # [B, max_blocks] int32 (decode)
     context_lens: Optional[torch.Tensor] = None  # [B] int32 (decode)
@@ -73,14 +73,15 @@ class LlamaAttention(nn.Module):
         v = v.view(T, self.num_kv_heads, self.head_dim)
         ops.rope_inplace(q, k, batch.positions, cos_sin)
 
-        k_cache, v_cache = batch.kv_caches[self.layer_idx]
-        ops.store_kv(k, v, k_cache, v_cache, batch.slot_mapping)
+        k_cache, v_cache, k_scale, v_scale = batch.kv_caches[self.layer_idx]
+        ops.store_kv(k, v, k_cache, v_cache, batch.slot_mapping, k_scale, v_scale)
 
         if batch.mode == "prefill":
             out = ops.attn_prefill_varlen(q, k, v, batch.cu_seqlens, self.scale)
         else:
             out = ops.attn_decode_paged(
-                q, k_cache, v_cache, batch.block_tables, batch.context_lens, self.scale
+                q, k_cache, v_cache, batch.block_tables, batch.context_lens, self.scale,
+                k_scale, v_scale,
             )
         return self.o_proj(out.reshape(T, -1))
 

@@ -87,11 +87,19 @@ def store_kv(
     k_cache: torch.Tensor,
     v_cache: torch.Tensor,
     slot_mapping: torch.Tensor,
+    k_scale: Optional[torch.Tensor] = None,
+    v_scale: Optional[torch.Tensor] = None,
 ) -> None:
+    """k_scale/v_scale: [NB, KVH, BS] fp32 per-row dequant scales, REQUIRED
+    when the caches are fp8 (written here: s = amax(|row|)/448), None for bf16."""
     if k.is_cuda and not _force_torch():
-        _hip_or_raise().store_kv(k, v, k_cache, v_cache, slot_mapping)
+        none = torch.empty(0, dtype=torch.float32, device=k.device)
+        _hip_or_raise().store_kv(
+            k, v, k_cache, v_cache, slot_mapping,
+            k_scale if k_scale is not None else none,
+            v_scale if v_scale is not None else none)
         return
-    torch_ref.store_kv(k, v, k_cache, v_cache, slot_mapping)
+    torch_ref.store_kv(k, v, k_cache, v_cache, slot_mapping, k_scale, v_scale)
 
 
 def _prefill_tiles(cu_seqlens: torch.Tensor, device) -> tuple:
@@ -133,12 +141,19 @@ def attn_decode_paged(
     block_tables: torch.Tensor,
     context_lens: torch.Tensor,
     scale: float,
+    k_scale: Optional[torch.Tensor] = None,
+    v_scale: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     if q.is_cuda and not _force_torch():
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-        _hip_or_raise().attn_decode_paged(out, q, k_cache, v_cache, block_tables, context_lens, scale)
+        none = torch.empty(0, dtype=torch.float32, device=q.device)
+        _hip_or_raise().attn_decode_paged(
+            out, q, k_cache, v_cache, block_tables, context_lens, scale,
+            k_scale if k_scale is not None else none,
+            v_scale if v_scale is not None else none)
         return out
-    return torch_ref.attn_decode_paged(q, k_cache, v_cache, block_tables, context_lens, scale)
+    return torch_ref.attn_decode_paged(q, k_cache, v_cache, block_tables, context_lens, scale,
+                                       k_scale, v_scale)
 
 
 def sample(

@@ -38,6 +38,8 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
     const bf16_t* __restrict__ q,        // [B, H, 128], row stride q_tstride
     const char* __restrict__ k_cache,    // [NB, KVH, BS, 128], bf16 or fp8
     const char* __restrict__ v_cache,
+    const float* __restrict__ k_scale,   // [NB, KVH, BS] per-row dequant (FP8 only)
+    const float* __restrict__ v_scale,
     const int* __restrict__ block_tables,  // [B, max_blocks]
     const int* __restrict__ context_lens,  // [B]
     float scale, int num_kv_heads,
@@ -59,6 +61,8 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
   float* alpha_lds = lstate + 8;                           // [GQA]
   float* o_scratch = alpha_lds + 8;                        // [4][GQA][128]
   int64_t* rowoff = reinterpret_cast<int64_t*>(o_scratch + 4 * GQA * 128);  // [TKV]
+  float* ks_lds = reinterpret_cast<float*>(rowoff + TKV);                   // [TKV] (FP8)
+  float* vs_lds = ks_lds + TKV;                                             // [TKV] (FP8)
 
   for (int i = tid; i < GQA * 128; i += NTHREADS) {
     const int g = i >> 7, d = i & 127;
@@ -76,11 +80,17 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
 
   for (int tile = c0; tile < c1; tile += TKV) {
     const int nkeys = min(TKV, c1 - tile);
-    // stage per-key cache-row element offsets (shared by K and V)
+    // stage per-key cache-row element offsets (shared by K and V) and, for
+    // fp8 caches, the per-row dequant scales (scale row index = rowoff/128)
     for (int i = tid; i < nkeys; i += NTHREADS) {
       const int gk = tile + i;
-      rowoff[i] = (((int64_t)bt[gk / block_size] * num_kv_heads + g_kv) * block_size +
-                   (gk % block_size)) * 128;
+      const int64_t off = (((int64_t)bt[gk / block_size] * num_kv_heads + g_kv) * block_size +
+                           (gk % block_size)) * 128;
+      rowoff[i] = off;
+      if constexpr (FP8) {
+        ks_lds[i] = k_scale[off >> 7];
+        vs_lds[i] = v_scale[off >> 7];
+      }
     }
     __syncthreads();
 
@@ -120,7 +130,9 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
             for (int j = 0; j < ELEMS; ++j) acc += af[j] * qg[j];
 #pragma unroll
             for (int off = KLANES / 2; off > 0; off >>= 1) acc += __shfl_xor(acc, off);
-            if (kl == 0) s_lds[g * TKV + key] = acc;
+            // fp8: fold the key row's dequant scale into the score (q already
+            // carries the softmax scale; s_k is per-key, shared across g)
+            if (kl == 0) s_lds[g * TKV + key] = FP8 ? acc * ks_lds[key] : acc;
           }
         }
       }
@@ -180,10 +192,11 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
         }
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
-          const float v0 = FP8 ? fp8_to_f32((unsigned char)(pairs[u] & 0xff))
-                               : bf16_to_f32((short)(pairs[u] & 0xffff));
-          const float v1 = FP8 ? fp8_to_f32((unsigned char)((pairs[u] >> 8) & 0xff))
-                               : bf16_to_f32((short)(pairs[u] >> 16));
+          const float vs = FP8 ? vs_lds[key + u] : 1.0f;  // per-row V dequant
+          const float v0 = vs * (FP8 ? fp8_to_f32((unsigned char)(pairs[u] & 0xff))
+                                     : bf16_to_f32((short)(pairs[u] & 0xffff)));
+          const float v1 = vs * (FP8 ? fp8_to_f32((unsigned char)((pairs[u] >> 8) & 0xff))
+                                     : bf16_to_f32((short)(pairs[u] >> 16)));
 #pragma unroll
           for (int g = 0; g < GQA; ++g) {
             const float p = s_lds[g * TKV + key + u];  // wave-uniform: LDS broadcast
@@ -198,10 +211,11 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
           pair = *reinterpret_cast<const unsigned short*>(v_cache + rowoff[key] + d0);
         else
           pair = *reinterpret_cast<const uint32_t*>(v_cache + 2 * (rowoff[key] + d0));
-        const float v0 = FP8 ? fp8_to_f32((unsigned char)(pair & 0xff))
-                             : bf16_to_f32((short)(pair & 0xffff));
-        const float v1 = FP8 ? fp8_to_f32((unsigned char)((pair >> 8) & 0xff))
-                             : bf16_to_f32((short)(pair >> 16));
+        const float vs = FP8 ? vs_lds[key] : 1.0f;
+        const float v0 = vs * (FP8 ? fp8_to_f32((unsigned char)(pair & 0xff))
+                                   : bf16_to_f32((short)(pair & 0xffff)));
+        const float v1 = vs * (FP8 ? fp8_to_f32((unsigned char)((pair >> 8) & 0xff))
+                                   : bf16_to_f32((short)(pair >> 16)));
 #pragma unroll
         for (int g = 0; g < GQA; ++g) {
           const float p = s_lds[g * TKV + key];
@@ -243,6 +257,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_decode_partial_t(
 // host-side dispatcher: picks the GQA template instantiation
 extern "C" void launch_attn_decode_partial(
     float* partials, const bf16_t* q, const void* k_cache, const void* v_cache,
+    const float* k_scale, const float* v_scale,
     const int* block_tables, const int* context_lens, float scale,
     int num_q_heads, int num_kv_heads, int block_size, int max_blocks,
     int max_chunks, int q_tstride, int chunk_keys, int B, int cache_fp8,
@@ -250,10 +265,11 @@ extern "C" void launch_attn_decode_partial(
   const int gqa = num_q_heads / num_kv_heads;
   const dim3 grid(B, num_kv_heads, max_chunks);
   const size_t lds = (gqa * 128 + gqa * TKV + 24 + 4 * gqa * 128) * sizeof(float) +
-                     TKV * sizeof(int64_t);
+                     TKV * sizeof(int64_t) + 2 * TKV * sizeof(float);
 #define LAUNCH(G, F)                                                                  \
   hipLaunchKernelGGL((attn_decode_partial_t<G, F>), grid, dim3(NTHREADS), lds, stream, \
                      partials, q, (const char*)k_cache, (const char*)v_cache,          \
+                     k_scale, v_scale,                                                 \
                      block_tables, context_lens, scale, num_kv_heads, block_size,      \
                      max_blocks, max_chunks, q_tstride, chunk_keys)
 #define DISPATCH(G) do { if (cache_fp8) LAUNCH(G, true); else LAUNCH(G, false); } while (0)

@@ -12,8 +12,8 @@ extern "C" __global__ void fused_add_rmsnorm_kernel(bf16_t*, const bf16_t*, bf16
 extern "C" __global__ void silu_mul_kernel(bf16_t*, const bf16_t*, const bf16_t*, int64_t, int, int);
 extern "C" __global__ void rope_kernel(bf16_t*, bf16_t*, const int64_t*, const float*, int, int, int, int, int);
 extern "C" __global__ void store_kv_kernel(const bf16_t*, const bf16_t*, bf16_t*, bf16_t*, const int64_t*, int, int, int, int, int);
-extern "C" __global__ void store_kv_fp8_kernel(const bf16_t*, const bf16_t*, unsigned char*, unsigned char*, const int64_t*, int, int, int, int, int);
-extern "C" void launch_attn_decode_partial(float*, const bf16_t*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int, int, int, hipStream_t);
+extern "C" __global__ void store_kv_fp8_kernel(const bf16_t*, const bf16_t*, unsigned char*, unsigned char*, float*, float*, const int64_t*, int, int, int, int, int);
+extern "C" void launch_attn_decode_partial(float*, const bf16_t*, const void*, const void*, const float*, const float*, const int*, const int*, float, int, int, int, int, int, int, int, int, int, hipStream_t);
 extern "C" __global__ void attn_decode_reduce_kernel(bf16_t*, const float*, const int*, int, int, int, int);
 extern "C" __global__ void attn_prefill_kernel(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, const int*, float, int, int, int, int);  // grouped: [G], [G], [G*8]
 extern "C" __global__ void sample_kernel(int64_t*, float*, const float*, const float*, const float*, const int*, const int64_t*, const int64_t*, const uint32_t*, int, float*);
@@ -82,20 +82,27 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions, tor
 }
 
 void store_kv(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache, torch::Tensor v_cache,
-              torch::Tensor slot_mapping) {
+              torch::Tensor slot_mapping, torch::Tensor k_scale, torch::Tensor v_scale) {
   CHECK_BF16_ROWS(k); CHECK_BF16_ROWS(v); CHECK_KV_CACHE(k_cache); CHECK_KV_CACHE(v_cache);
   TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share the token stride");
   const int T = k.size(0), KVH = k.size(1), D = k.size(2);
   const int BS = k_cache.size(2);
-  const int64_t total = (int64_t)T * KVH * (D / 8);
-  const int grid = (int)std::min<int64_t>((total + 255) / 256, 2048);
   if (k_cache.scalar_type() == at::kFloat8_e4m3fn) {
+    // per-row quantization: one wave per (token, head) row, 4 waves/block
+    TORCH_CHECK(k_scale.numel() == k_cache.size(0) * KVH * BS && k_scale.is_contiguous() &&
+                    k_scale.scalar_type() == at::kFloat && v_scale.numel() == k_scale.numel(),
+                "fp8 store_kv requires [NB, KVH, BS] fp32 k_scale/v_scale");
+    const int64_t rows = (int64_t)T * KVH;
+    const int grid = (int)std::min<int64_t>((rows + 3) / 4, 2048);
     hipLaunchKernelGGL(store_kv_fp8_kernel, dim3(grid), dim3(256), 0, cur_stream(),
                        cbf(k), cbf(v),
                        reinterpret_cast<unsigned char*>(k_cache.data_ptr()),
                        reinterpret_cast<unsigned char*>(v_cache.data_ptr()),
+                       k_scale.data_ptr<float>(), v_scale.data_ptr<float>(),
                        slot_mapping.data_ptr<int64_t>(), T, KVH, D, BS, (int)k.stride(0));
   } else {
+    const int64_t total = (int64_t)T * KVH * (D / 8);
+    const int grid = (int)std::min<int64_t>((total + 255) / 256, 2048);
     hipLaunchKernelGGL(store_kv_kernel, dim3(grid), dim3(256), 0, cur_stream(),
                        cbf(k), cbf(v), bf(k_cache), bf(v_cache),
                        slot_mapping.data_ptr<int64_t>(), T, KVH, D, BS, (int)k.stride(0));
@@ -104,7 +111,8 @@ void store_kv(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache, torch::Te
 
 void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                        torch::Tensor v_cache, torch::Tensor block_tables,
-                       torch::Tensor context_lens, double scale) {
+                       torch::Tensor context_lens, double scale,
+                       torch::Tensor k_scale, torch::Tensor v_scale) {
   CHECK_BF16_CONTIG(out); CHECK_BF16_ROWS(q); CHECK_KV_CACHE(k_cache); CHECK_KV_CACHE(v_cache);
   const int B = q.size(0), H = q.size(1), D = q.size(2);
   const int KVH = k_cache.size(1), BS = k_cache.size(2);
@@ -123,8 +131,15 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache
   auto partials = torch::empty({(int64_t)B * KVH * max_chunks * gqa * 130},
                                torch::dtype(torch::kFloat).device(q.device()));
   const int cache_fp8 = k_cache.scalar_type() == at::kFloat8_e4m3fn ? 1 : 0;
+  if (cache_fp8) {
+    TORCH_CHECK(k_scale.numel() == k_cache.size(0) * KVH * BS && k_scale.is_contiguous() &&
+                    k_scale.scalar_type() == at::kFloat && v_scale.numel() == k_scale.numel(),
+                "fp8 attn_decode requires [NB, KVH, BS] fp32 k_scale/v_scale");
+  }
   launch_attn_decode_partial(
       partials.data_ptr<float>(), cbf(q), k_cache.data_ptr(), v_cache.data_ptr(),
+      cache_fp8 ? k_scale.data_ptr<float>() : nullptr,
+      cache_fp8 ? v_scale.data_ptr<float>() : nullptr,
       block_tables.data_ptr<int>(), context_lens.data_ptr<int>(),
       (float)scale, H, KVH, BS, max_blocks, max_chunks, (int)q.stride(0), CHUNK_KEYS, B,
       cache_fp8, cur_stream());

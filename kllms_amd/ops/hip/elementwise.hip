@@ -172,36 +172,45 @@ extern "C" __global__ void __launch_bounds__(256) store_kv_kernel(
   }
 }
 
-// fp8 (e4m3) variant of the paged-KV scatter: bf16 activations quantized
-// into 1-byte cache elements (static scale 1.0; e4m3 covers +-448 which the
-// post-RoPE K / V magnitudes fit comfortably).
+// fp8 (e4m3) variant of the paged-KV scatter with PER-ROW dequant scales:
+// one 64-lane wave per (token, kv_head) row computes s = max(amax|row|/448,
+// 1e-8) over the row's K (and V) values, stores the scales to
+// k_scale/v_scale [NB, KVH, BS] and quantizes x/s into the 1-byte caches —
+// outlier rows in real checkpoints no longer saturate e4m3's +-448 range.
+// T*KVH rows per call is tiny next to the attention reads, so a plain
+// strided (non-vectorized) load pattern is fine here.
 extern "C" __global__ void __launch_bounds__(256) store_kv_fp8_kernel(
     const bf16_t* __restrict__ k, const bf16_t* __restrict__ v,
     unsigned char* __restrict__ k_cache, unsigned char* __restrict__ v_cache,
+    float* __restrict__ k_scale, float* __restrict__ v_scale,
     const int64_t* __restrict__ slots, int num_tokens, int kv_heads,
     int head_dim, int block_size, int kv_tstride) {
-  const int dvec = head_dim / 8;
-  const int64_t total = (int64_t)num_tokens * kv_heads * dvec;
-  for (int64_t idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (int64_t)gridDim.x * blockDim.x) {
-    const int dv = idx % dvec;
-    const int g = (idx / dvec) % kv_heads;
-    const int t = idx / ((int64_t)dvec * kv_heads);
+  const int64_t rows = (int64_t)num_tokens * kv_heads;
+  const int lane = threadIdx.x & 63;
+  for (int64_t row = blockIdx.x * 4 + (threadIdx.x >> 6); row < rows;
+       row += (int64_t)gridDim.x * 4) {
+    const int g = row % kv_heads;
+    const int t = row / kv_heads;
     const int64_t slot = slots[t];
     const int64_t blk = slot / block_size;
     const int off = slot % block_size;
-    const bf16x8_vec src_k =
-        reinterpret_cast<const bf16x8_vec*>(k + (int64_t)t * kv_tstride + g * head_dim)[dv];
-    const bf16x8_vec src_v =
-        reinterpret_cast<const bf16x8_vec*>(v + (int64_t)t * kv_tstride + g * head_dim)[dv];
-    u8x8_vec qk, qv;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      qk[j] = f32_to_fp8(bf16_to_f32(src_k[j]));
-      qv[j] = f32_to_fp8(bf16_to_f32(src_v[j]));
+    const bf16_t* src_k = k + (int64_t)t * kv_tstride + g * head_dim;
+    const bf16_t* src_v = v + (int64_t)t * kv_tstride + g * head_dim;
+    float amax_k = 0.f, amax_v = 0.f;
+    for (int d = lane; d < head_dim; d += 64) {
+      amax_k = fmaxf(amax_k, fabsf(bf16_to_f32(((const short*)src_k)[d])));
+      amax_v = fmaxf(amax_v, fabsf(bf16_to_f32(((const short*)src_v)[d])));
     }
-    const int64_t dst_off = ((blk * kv_heads + g) * block_size + off) * head_dim;
-    reinterpret_cast<u8x8_vec*>(k_cache + dst_off)[dv] = qk;
-    reinterpret_cast<u8x8_vec*>(v_cache + dst_off)[dv] = qv;
+    const float sk = fmaxf(wave_reduce_max(amax_k) * (1.0f / 448.0f), 1e-8f);
+    const float sv = fmaxf(wave_reduce_max(amax_v) * (1.0f / 448.0f), 1e-8f);
+    const int64_t srow = (blk * kv_heads + g) * block_size + off;
+    if (lane == 0) { k_scale[srow] = sk; v_scale[srow] = sv; }
+    const float isk = 1.0f / sk, isv = 1.0f / sv;
+    unsigned char* dst_k = k_cache + srow * head_dim;
+    unsigned char* dst_v = v_cache + srow * head_dim;
+    for (int d = lane; d < head_dim; d += 64) {
+      dst_k[d] = f32_to_fp8(bf16_to_f32(((const short*)src_k)[d]) * isk);
+      dst_v[d] = f32_to_fp8(bf16_to_f32(((const short*)src_v)[d]) * isv);
+    }
   }
 }

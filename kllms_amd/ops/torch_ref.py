@@ -77,14 +77,27 @@ def store_kv(
     k_cache: torch.Tensor,
     v_cache: torch.Tensor,
     slot_mapping: torch.Tensor,
+    k_scale=None,
+    v_scale=None,
 ) -> None:
     """Scatter new K/V ([T, KVH, D]) into the paged caches at flat slots.
 
     A flat slot s addresses (block = s // block_size, offset = s % block_size).
+    fp8 caches with k_scale/v_scale ([NB, KVH, BS] fp32): each (token, head)
+    row is quantized with its OWN scale s = max(amax(|row|)/448, 1e-8), the
+    scale written alongside — outlier rows no longer saturate e4m3.
     """
     num_blocks, kv_heads, block_size, head_dim = k_cache.shape
     blk = slot_mapping // block_size
     off = slot_mapping % block_size
+    if k_scale is not None:
+        sk = (k.float().abs().amax(-1) / 448.0).clamp_(min=1e-8)  # [T, KVH]
+        sv = (v.float().abs().amax(-1) / 448.0).clamp_(min=1e-8)
+        k_scale[blk, :, off] = sk
+        v_scale[blk, :, off] = sv
+        k_cache[blk, :, off, :] = (k.float() / sk.unsqueeze(-1)).to(k_cache.dtype)
+        v_cache[blk, :, off, :] = (v.float() / sv.unsqueeze(-1)).to(v_cache.dtype)
+        return
     k_cache[blk, :, off, :] = k.to(k_cache.dtype)
     v_cache[blk, :, off, :] = v.to(v_cache.dtype)
 
@@ -126,12 +139,15 @@ def attn_decode_paged(
     block_tables: torch.Tensor,
     context_lens: torch.Tensor,
     scale: float,
+    k_scale=None,
+    v_scale=None,
 ) -> torch.Tensor:
     """Single-token decode attention against the paged KV cache.
 
     q: [B, H, D]; caches [num_blocks, KVH, block_size, D];
     block_tables: [B, max_blocks] int32; context_lens: [B] (length INCLUDING
-    the token being decoded, whose K/V are already stored).
+    the token being decoded, whose K/V are already stored). k_scale/v_scale
+    ([NB, KVH, BS] fp32, fp8 caches only): per-row dequant multipliers.
     """
     B, H, D = q.shape
     _, KVH, BS, _ = k_cache.shape
@@ -143,6 +159,9 @@ def attn_decode_paged(
         blocks = block_tables[b, :nblk].long()
         kk = k_cache[blocks].permute(1, 0, 2, 3).reshape(KVH, nblk * BS, D)[:, :L].float()
         vv = v_cache[blocks].permute(1, 0, 2, 3).reshape(KVH, nblk * BS, D)[:, :L].float()
+        if k_scale is not None:
+            kk = kk * k_scale[blocks].permute(1, 0, 2).reshape(KVH, nblk * BS)[:, :L, None]
+            vv = vv * v_scale[blocks].permute(1, 0, 2).reshape(KVH, nblk * BS)[:, :L, None]
         qb = q[b].float()  # [H, D]
         # per-head GQA mapping without materializing repeats
         for h in range(H):

@@ -479,18 +479,34 @@ class TestFp8KVCacheGPU:
         T = NB * BS
         k = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=DEV) * 0.3
         v = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=DEV) * 0.3
+        # outlier row: would saturate e4m3's +-448 under a static scale; the
+        # per-row scale must absorb it
+        k[3, 1, 7] = 800.0
         slots = torch.arange(T, device=DEV)
-        ops.store_kv(k, v, kc, vc, slots)
+        ks = torch.ones(NB, KVH, BS, device=DEV)
+        vs = torch.ones_like(ks)
+        ops.store_kv(k, v, kc, vc, slots, ks, vs)
         # quantized store matches the torch reference (dequantized compare:
         # HIP's and torch's fp8 RNE can differ on exact ties, so bitwise
         # equality is the wrong assertion)
         ref_kc = torch.zeros_like(kc)
         ref_vc = torch.zeros_like(vc)
-        torch_ref.store_kv(k, v, ref_kc, ref_vc, slots)
-        diff = (kc.float() - ref_kc.float()).abs()
-        assert diff.max().item() <= 0.0625, f"fp8 store quantization diverges: {diff.max().item()}"
+        ref_ks = torch.ones_like(ks)
+        ref_vs = torch.ones_like(vs)
+        torch_ref.store_kv(k, v, ref_kc, ref_vc, slots, ref_ks, ref_vs)
+        assert torch.allclose(ks, ref_ks, rtol=1e-6), "per-row K scales diverge"
+        assert torch.allclose(vs, ref_vs, rtol=1e-6), "per-row V scales diverge"
+        deq = kc.float() * ks.unsqueeze(-1)
+        ref_deq = ref_kc.float() * ref_ks.unsqueeze(-1)
+        # one RNE tie at the top of the code range costs ulp(448)*s
+        thresh = (ks.max().item() * 32) + 1e-6
+        diff = (deq - ref_deq).abs()
+        assert diff.max().item() <= thresh, f"fp8 store quantization diverges: {diff.max().item()}"
         mismatch_frac = (kc.view(torch.uint8) != ref_kc.view(torch.uint8)).float().mean().item()
         assert mismatch_frac < 0.01, f"too many fp8 rounding mismatches: {mismatch_frac}"
+        # the outlier survives per-row quantization with e4m3 relative precision
+        got = deq[slots[3] // BS, 1, slots[3] % BS, 7]
+        assert abs(got.item() - 800.0) <= 800.0 * 0.0625, f"outlier lost: {got.item()}"
 
         max_blocks = max((c + BS - 1) // BS for c in ctx)
         perm = torch.randperm(NB).tolist()
@@ -502,9 +518,17 @@ class TestFp8KVCacheGPU:
         lens = torch.tensor(ctx, dtype=torch.int32, device=DEV)
         q = torch.randn(B, H, D, dtype=torch.bfloat16, device=DEV) * 0.5
         scale = D ** -0.5
-        out = ops.attn_decode_paged(q, kc, vc, bt, lens, scale)
-        ref = torch_ref.attn_decode_paged(q, kc, vc, bt, lens, scale)
+        out = ops.attn_decode_paged(q, kc, vc, bt, lens, scale, ks, vs)
+        ref = torch_ref.attn_decode_paged(q, kc, vc, bt, lens, scale, ref_ks, ref_vs)
         assert_close_bf16(out, ref, rtol=4e-2, atol=4e-2, msg="fp8 decode")
+        # and the whole fp8 path tracks a full-precision oracle: per-row
+        # scales keep the quantization error at e4m3's relative precision
+        # even with the injected outlier in the context
+        kc_bf = torch.zeros(NB, KVH, BS, D, dtype=torch.bfloat16, device=DEV)
+        vc_bf = torch.zeros_like(kc_bf)
+        torch_ref.store_kv(k, v, kc_bf, vc_bf, slots)
+        oracle = torch_ref.attn_decode_paged(q, kc_bf, vc_bf, bt, lens, scale)
+        assert_close_bf16(out, oracle, rtol=8e-2, atol=8e-2, msg="fp8 vs bf16 oracle")
 
     def test_engine_fp8_cache_generates(self):
         from kllms_amd.engine.config import EngineConfig

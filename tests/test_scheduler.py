@@ -137,7 +137,7 @@ class TestChunkedPrefill:
         # KV written by both paths is identical per slot
         slots_a = eng.kv.prefill_slot_mapping(seq_a)
         slots_b = eng.kv.prefill_slot_mapping(seq_b)
-        for (kc, vc) in eng.kv.layer_caches():
+        for (kc, vc, _ks, _vs) in eng.kv.layer_caches():
             bs = eng.kv.block_size
             ka = kc.view(-1, kc.shape[1], kc.shape[3])  # can't index slots directly; compare per slot
             for sa, sb in zip(slots_a, slots_b):
