@@ -205,6 +205,45 @@ class TestFp8KVCache:
         agree = sum(x == y for x, y in zip(a, b))
         assert agree >= len(a) - 2, f"fp8 cache diverged early: {a} vs {b}"
 
+    def test_per_row_scales_survive_outliers(self):
+        """An outlier K/V row beyond e4m3's +-448 would be clamped under a
+        static scale; the per-row amax/448 scale must preserve it to e4m3's
+        RELATIVE precision, and decode must dequantize with it."""
+        import torch
+        from kllms_amd.ops import torch_ref
+
+        KVH, D, BS, NB = 2, 128, 16, 8
+        kc = torch.zeros(NB, KVH, BS, D, dtype=torch.float8_e4m3fn)
+        vc = torch.zeros_like(kc)
+        ks = torch.ones(NB, KVH, BS)
+        vs = torch.ones_like(ks)
+        T = 20
+        k = torch.randn(T, KVH, D, dtype=torch.bfloat16) * 0.3
+        v = torch.randn(T, KVH, D, dtype=torch.bfloat16) * 0.3
+        k[5, 0, 3] = 1200.0
+        v[9, 1, 100] = -777.0
+        slots = torch.arange(T)
+        torch_ref.store_kv(k, v, kc, vc, slots, ks, vs)
+        # round-trip of the outliers (e4m3: 3 mantissa bits -> rel err <= 2^-4)
+        deq_k = kc.float()[0, 0, 5] * ks[0, 0, 5]
+        assert abs(deq_k[3].item() - 1200.0) <= 1200.0 * 0.0625
+        deq_v = vc.float()[0, 1, 9] * vs[0, 1, 9]
+        assert abs(deq_v[100].item() + 777.0) <= 777.0 * 0.0625
+        # non-outlier rows keep a tight absolute error too (own small scale)
+        err = (kc.float()[0, 1] * ks[0, 1, :, None] - k[:BS, 1].float()).abs()
+        assert err.max().item() <= 0.3 * 4 * 0.0625
+        # decode consumes the scales: matches a bf16-cache oracle
+        q = torch.randn(2, KVH * 2, D, dtype=torch.bfloat16) * 0.5
+        bt = torch.zeros(2, 2, dtype=torch.int32)
+        bt[:, 1] = 1
+        lens = torch.tensor([T, T - 6], dtype=torch.int32)
+        out = torch_ref.attn_decode_paged(q, kc, vc, bt, lens, D ** -0.5, ks, vs)
+        kb = torch.zeros(NB, KVH, BS, D, dtype=torch.bfloat16)
+        vb = torch.zeros_like(kb)
+        torch_ref.store_kv(k, v, kb, vb, slots)
+        oracle = torch_ref.attn_decode_paged(q, kb, vb, bt, lens, D ** -0.5)
+        assert torch.allclose(out.float(), oracle.float(), rtol=8e-2, atol=8e-2)
+
 
 class TestPenalties:
     def test_apply_penalties_math(self):
