@@ -103,6 +103,67 @@ def test_tp2_matches_tp1_logits():
     assert out.streams[0].token_ids == results[0][1], "TP=2 greedy diverged from TP=1"
 
 
+def _tp_fp8_worker(rank: int, world_size: int, q):
+    """TP=2 with the fp8 KV cache: per-row dequant scale tensors are sharded
+    with the KV heads; ranks must stay in lockstep (identical all-reduce
+    inputs require identical dequantized KV on both ranks)."""
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(PORT + 11)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.engine import GenRequest, LLMEngine
+        from kllms_amd.engine.sampling import SamplingParams
+        from kllms_amd.parallel.tp import ParallelContext
+
+        ctx = ParallelContext(world_size=world_size, rank=rank)
+        eng = LLMEngine(
+            EngineConfig(model="tiny-llama", tp_size=world_size, max_kv_blocks=128,
+                         use_hip_graphs=False, device="cpu", seed=0,
+                         kv_cache_dtype="fp8_e4m3"),
+            parallel_ctx=ctx,
+        )
+        assert eng.kv.fp8 and eng.kv.k_scale_all is not None
+        ids = list(range(1, 33))
+        out = eng.generate([GenRequest(prompt_ids=ids, n=2,
+                                       sampling=SamplingParams(temperature=0.0, max_tokens=8))])[0]
+        q.put((rank, out.streams[0].token_ids))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp2_fp8_cache_lockstep():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tp_fp8_worker, args=(r, 2, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, tokens = q.get(timeout=240)
+        results[rank] = tokens
+    for p in procs:
+        p.join(timeout=60)
+    assert results[0] == results[1], "fp8 TP ranks diverged"
+
+    # TP=1 fp8 baseline: same quantization path, so greedy should agree
+    # (allow the usual 2-token fp8 slack — reduction order differs under TP)
+    from kllms_amd.engine.config import EngineConfig
+    from kllms_amd.engine.engine import GenRequest, LLMEngine
+    from kllms_amd.engine.sampling import SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="tiny-llama", max_kv_blocks=128, use_hip_graphs=False,
+                                 device="cpu", seed=0, kv_cache_dtype="fp8_e4m3"))
+    out = eng.generate([GenRequest(prompt_ids=list(range(1, 33)), n=2,
+                                   sampling=SamplingParams(temperature=0.0, max_tokens=8))])[0]
+    a, b = out.streams[0].token_ids, results[0]
+    agree = sum(x == y for x, y in zip(a, b))
+    assert agree >= len(a) - 2, f"fp8 TP=2 diverged from TP=1: {a} vs {b}"
+
+
 def _dp_worker(rank: int, world_size: int, q):
     """tp_size=1 engines under an INITIALIZED process group must be fully
     independent (the bench's data-parallel weak-scaling mode): different
