@@ -201,9 +201,13 @@ class LLMEngine:
         tp = self.ctx.world_size
         kv_heads_local = max(1, a.num_kv_heads // tp)
         elem = torch.tensor([], dtype=self.dtype).element_size()
+        scale_bytes = 0
         if self.config.kv_cache_dtype == "fp8_e4m3":
             elem = 1
-        block_bytes = 2 * a.num_layers * kv_heads_local * self.config.kv_block_size * a.head_dim_ * elem
+            # per-row fp32 dequant scale sidecar: 4 B per (head_dim) row
+            scale_bytes = 2 * a.num_layers * kv_heads_local * self.config.kv_block_size * 4
+        block_bytes = (2 * a.num_layers * kv_heads_local * self.config.kv_block_size *
+                       a.head_dim_ * elem + scale_bytes)
         if self.config.max_kv_blocks is not None:
             num_blocks = self.config.max_kv_blocks
         elif self.device.type == "cuda":

@@ -339,3 +339,36 @@ class TestServingMetadata:
         for b, s in zip(batch, singles):
             assert [c.message.content for c in b.choices] == [c.message.content for c in s.choices]
             assert b.usage.prompt_tokens == s.usage.prompt_tokens
+
+
+class TestFp8ScaleProperties:
+    def test_round_trip_error_bound_randomized(self):
+        """Property: for ANY row magnitudes (including extreme outliers and
+        all-zero rows), per-row-scaled fp8 round-trip keeps every element's
+        error within e4m3's relative precision of the row amax."""
+        import torch
+        from kllms_amd.ops import torch_ref
+
+        g = torch.Generator().manual_seed(123)
+        KVH, D, BS, NB = 2, 128, 16, 4
+        for trial in range(25):
+            T = int(torch.randint(1, NB * BS + 1, (1,), generator=g))
+            mag = 10.0 ** float(torch.empty(1).uniform_(-3, 5, generator=g))
+            k = (torch.randn(T, KVH, D, generator=g) * mag).bfloat16()
+            v = (torch.randn(T, KVH, D, generator=g) * mag).bfloat16()
+            if trial % 5 == 0:
+                k[0, 0].zero_()  # all-zero row: scale floors at 1e-8, no NaN
+            kc = torch.zeros(NB, KVH, BS, D, dtype=torch.float8_e4m3fn)
+            vc = torch.zeros_like(kc)
+            ks = torch.ones(NB, KVH, BS)
+            vs = torch.ones_like(ks)
+            slots = torch.randperm(NB * BS, generator=g)[:T]
+            torch_ref.store_kv(k, v, kc, vc, slots, ks, vs)
+            blk, off = slots // BS, slots % BS
+            deq = kc.float()[blk, :, off] * ks[blk, :, off, None]
+            err = (deq - k.float()).abs()
+            # bound: half-ulp at the top of the scaled range = amax * 2^-4,
+            # plus the bf16 input's own rounding
+            amax = k.float().abs().amax(-1, keepdim=True)
+            assert torch.isfinite(deq).all()
+            assert (err <= amax * 0.0625 + 1e-6).all(), (trial, err.max(), amax.max())
