@@ -30,6 +30,8 @@ class ModelArchConfig(BaseModel):
     rms_norm_eps: float = 1e-5
     max_position_embeddings: int = 8192
     tie_word_embeddings: bool = False
+    # Qwen2-family: bias on the fused QKV projection (o_proj stays bias-free)
+    attention_qkv_bias: bool = False
     # MoE (mixtral only)
     num_experts: int = 0
     num_experts_per_tok: int = 2
@@ -60,6 +62,12 @@ MODEL_PRESETS: dict[str, ModelArchConfig] = {
         num_layers=2, num_heads=4, num_kv_heads=2, rope_theta=10000.0,
         max_position_embeddings=2048,
     ),
+    "mid-qwen": ModelArchConfig(
+        arch="llama", vocab_size=2048, hidden_size=512, intermediate_size=1024,
+        num_layers=2, num_heads=4, num_kv_heads=2, rope_theta=10000.0,
+        max_position_embeddings=2048, attention_qkv_bias=True,
+        tie_word_embeddings=True,
+    ),
     "mid-mixtral": ModelArchConfig(
         arch="mixtral", vocab_size=2048, hidden_size=512, intermediate_size=1024,
         num_layers=2, num_heads=4, num_kv_heads=2, rope_theta=10000.0,
@@ -81,6 +89,19 @@ MODEL_PRESETS: dict[str, ModelArchConfig] = {
         arch="mixtral", vocab_size=512, hidden_size=64, intermediate_size=128,
         num_layers=2, num_heads=4, num_kv_heads=2, rope_theta=10000.0,
         num_experts=4, num_experts_per_tok=2, max_position_embeddings=512,
+    ),
+    # Qwen2 family: Llama compute graph + QKV bias (+ tied embeddings on
+    # the small members); head_dim 128 so every HIP kernel applies as-is
+    "qwen2-7b": ModelArchConfig(
+        arch="llama", vocab_size=152064, hidden_size=3584, intermediate_size=18944,
+        num_layers=28, num_heads=28, num_kv_heads=4, rope_theta=1e6,
+        rms_norm_eps=1e-6, max_position_embeddings=32768, attention_qkv_bias=True,
+    ),
+    "tiny-qwen": ModelArchConfig(
+        arch="llama", vocab_size=512, hidden_size=64, intermediate_size=128,
+        num_layers=2, num_heads=4, num_kv_heads=2, rope_theta=10000.0,
+        max_position_embeddings=512, attention_qkv_bias=True,
+        tie_word_embeddings=True,
     ),
 }
 
@@ -170,4 +191,8 @@ class EngineConfig(BaseModel):
             tie_word_embeddings=hf.get("tie_word_embeddings", False),
             num_experts=hf.get("num_local_experts", 0),
             num_experts_per_tok=hf.get("num_experts_per_tok", 2),
+            # Qwen2 configs carry no attention_bias key but always use QKV
+            # bias; Llama-style configs say so explicitly
+            attention_qkv_bias=hf.get(
+                "attention_bias", hf.get("model_type", "").lower() == "qwen2"),
         )

@@ -61,10 +61,15 @@ def load_safetensors_weights(model, weights_dir: str, ctx: ParallelContext) -> N
                     assign("lm_head.weight", t)
                 elif ".self_attn.q_proj." in n or ".self_attn.k_proj." in n or ".self_attn.v_proj." in n:
                     layer = n.split(".")[1]
-                    key = f"layers.{layer}.self_attn.qkv_proj.weight"
+                    leaf = "bias" if n.endswith(".bias") else "weight"
+                    key = f"layers.{layer}.self_attn.qkv_proj.{leaf}"
+                    if leaf == "bias" and key not in params:
+                        raise RuntimeError(
+                            f"checkpoint has {name} but the model was built without "
+                            f"attention_qkv_bias (set it in the arch config)")
                     st = staged.setdefault(key, {})
                     which = "q" if ".q_proj." in n else ("k" if ".k_proj." in n else "v")
-                    # shard heads per rank, then fuse
+                    # shard heads per rank, then fuse (bias shards dim 0 too)
                     st[which] = _shard(t, 0, rank, world)
                     if len(st) == 3:
                         assign(key, torch.cat([st["q"], st["k"], st["v"]], dim=0))

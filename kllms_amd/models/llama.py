@@ -57,6 +57,7 @@ class LlamaAttention(nn.Module):
         self.qkv_proj = ColumnParallelLinear(
             cfg.hidden_size, q_size + 2 * kv_size, ctx, dtype=dtype,
             partition_sizes=[q_size, kv_size, kv_size],
+            bias=cfg.attention_qkv_bias,  # Qwen2-style
         )
         self.o_proj = RowParallelLinear(q_size, cfg.hidden_size, ctx, dtype=dtype)
         self._q = self.num_heads * self.head_dim
@@ -222,6 +223,9 @@ class LlamaForCausalLM(nn.Module):
                 else:
                     full = full.narrow(shard_dim, rank * size, size)
             p.copy_(full.to(p.dtype))
+        if self.cfg.tie_word_embeddings:
+            # mirror the weights loader: tied checkpoints share one matrix
+            self.lm_head.weight.copy_(self.embed_tokens.weight)
 
     def _owner_module(self, param_name: str):
         parts = param_name.split(".")[:-1]

@@ -64,7 +64,7 @@ class ColumnParallelLinear(nn.Module):
     """
 
     def __init__(self, in_features: int, out_features: int, ctx: ParallelContext, dtype=None,
-                 partition_sizes=None):
+                 partition_sizes=None, bias: bool = False):
         super().__init__()
         assert out_features % ctx.world_size == 0, (out_features, ctx.world_size)
         self.ctx = ctx
@@ -80,6 +80,14 @@ class ColumnParallelLinear(nn.Module):
         self.weight = nn.Parameter(
             torch.empty(self.out_features_per_rank, in_features, dtype=dtype), requires_grad=False
         )
+        if bias:
+            # output-sharded, so the bias shards with the weight's dim 0 and
+            # needs no communication (Qwen2-style attention qkv bias)
+            self.bias = nn.Parameter(
+                torch.empty(self.out_features_per_rank, dtype=dtype), requires_grad=False
+            )
+        else:
+            self.register_parameter("bias", None)
 
     def shard_full_tensor(self, full: torch.Tensor) -> torch.Tensor:
         """Slice this rank's shard out of the FULL fused weight (dim 0)."""
@@ -95,7 +103,7 @@ class ColumnParallelLinear(nn.Module):
         return torch.cat(pieces, dim=0)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return torch.nn.functional.linear(x, self.weight)
+        return torch.nn.functional.linear(x, self.weight, self.bias)
 
 
 class RowParallelLinear(nn.Module):

@@ -766,3 +766,28 @@ class TestRealVocabGPU:
                 assert ch.message.parsed is not None
                 obj = _json.loads(ch.message.content)
                 assert set(obj) == {"city", "n"} and 0 <= obj["n"] <= 99
+
+
+class TestQwenGPU:
+    def test_qwen_bias_engine_generates(self):
+        """Qwen2-style fused-QKV bias through the full GPU path (rocBLAS
+        GEMM + bias epilogue, HIP attention, hipGraph decode capture)."""
+        from kllms_amd.engine.config import EngineConfig
+        from kllms_amd.engine.engine import GenRequest, LLMEngine
+        from kllms_amd.engine.sampling import SamplingParams
+
+        eng = LLMEngine(EngineConfig(model="mid-qwen", max_kv_blocks=512, use_hip_graphs=True,
+                                     hip_graph_batch_sizes=[1, 2, 4], max_seq_len=512, seed=5))
+        assert eng.model.layers[0].self_attn.qkv_proj.bias is not None
+        assert torch.equal(eng.model.lm_head.weight, eng.model.embed_tokens.weight)
+        out = eng.generate([GenRequest(prompt_ids=list(range(1, 60)), n=2,
+                                       sampling=SamplingParams(temperature=0.0, max_tokens=10))])[0]
+        assert eng._graph_runner is not None and eng._graph_runner._enabled
+        assert out.streams[0].token_ids == out.streams[1].token_ids
+        # the bias is live: zeroing it changes the greedy continuation path
+        with torch.no_grad():
+            for layer in eng.model.layers:
+                layer.self_attn.qkv_proj.bias.zero_()
+        out2 = eng.generate([GenRequest(prompt_ids=list(range(1, 60)), n=1,
+                                        sampling=SamplingParams(temperature=0.0, max_tokens=10))])[0]
+        assert out2.streams[0].token_ids != out.streams[0].token_ids

@@ -14,7 +14,9 @@ from kllms_amd.parallel.tp import ParallelContext
 
 
 def _make_hf_llama_checkpoint(tmp_path, cfg, seed=0):
-    """Write a random HF-style (unfused q/k/v, gate/up) checkpoint."""
+    """Write a random HF-style (unfused q/k/v, gate/up) checkpoint.
+    Honors cfg.attention_qkv_bias (Qwen2-style q/k/v bias tensors) and
+    cfg.tie_word_embeddings (no lm_head.weight tensor in the files)."""
     from safetensors.torch import save_file
 
     g = torch.Generator().manual_seed(seed)
@@ -23,13 +25,18 @@ def _make_hf_llama_checkpoint(tmp_path, cfg, seed=0):
     tensors = {
         "model.embed_tokens.weight": torch.randn(cfg.vocab_size, H, generator=g),
         "model.norm.weight": torch.randn(H, generator=g),
-        "lm_head.weight": torch.randn(cfg.vocab_size, H, generator=g),
     }
+    if not cfg.tie_word_embeddings:
+        tensors["lm_head.weight"] = torch.randn(cfg.vocab_size, H, generator=g)
     for L in range(cfg.num_layers):
         p = f"model.layers.{L}"
         tensors[f"{p}.self_attn.q_proj.weight"] = torch.randn(cfg.num_heads * D, H, generator=g)
         tensors[f"{p}.self_attn.k_proj.weight"] = torch.randn(cfg.num_kv_heads * D, H, generator=g)
         tensors[f"{p}.self_attn.v_proj.weight"] = torch.randn(cfg.num_kv_heads * D, H, generator=g)
+        if cfg.attention_qkv_bias:
+            tensors[f"{p}.self_attn.q_proj.bias"] = torch.randn(cfg.num_heads * D, generator=g)
+            tensors[f"{p}.self_attn.k_proj.bias"] = torch.randn(cfg.num_kv_heads * D, generator=g)
+            tensors[f"{p}.self_attn.v_proj.bias"] = torch.randn(cfg.num_kv_heads * D, generator=g)
         tensors[f"{p}.self_attn.o_proj.weight"] = torch.randn(H, cfg.num_heads * D, generator=g)
         tensors[f"{p}.mlp.gate_proj.weight"] = torch.randn(cfg.intermediate_size, H, generator=g)
         tensors[f"{p}.mlp.up_proj.weight"] = torch.randn(cfg.intermediate_size, H, generator=g)
