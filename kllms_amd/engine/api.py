@@ -89,12 +89,35 @@ class LocalEngineClient:
         return self.tokenizer.crop_to_tokens(text, max_tokens)
 
     # --- core generation ------------------------------------------------------
+    @staticmethod
+    def _validate_call_params(call_params: Dict[str, Any]) -> None:
+        """OpenAI-style 400s for out-of-range sampling params (the reference
+        relies on the remote API to raise these; locally we mirror the
+        documented ranges so a switched-over client sees the same errors)."""
+        t = call_params.get("temperature")
+        if t is not None and not 0.0 <= float(t) <= 2.0:
+            raise ValueError(f"temperature must be between 0 and 2, got {t}")
+        p = call_params.get("top_p")
+        if p is not None and not 0.0 <= float(p) <= 1.0:
+            raise ValueError(f"top_p must be between 0 and 1, got {p}")
+        n = call_params.get("n")
+        if n is not None and not 1 <= int(n) <= 128:
+            raise ValueError(f"n must be between 1 and 128, got {n}")
+        mt = call_params.get("max_tokens")
+        if mt is not None and int(mt) < 1:
+            raise ValueError(f"max_tokens must be at least 1, got {mt}")
+        for key in ("frequency_penalty", "presence_penalty"):
+            v = call_params.get(key)
+            if v is not None and not -2.0 <= float(v) <= 2.0:
+                raise ValueError(f"{key} must be between -2 and 2, got {v}")
+
     def _generate(self, call_params: Dict[str, Any], constrained: bool, scheduled: bool = False) -> tuple:
         from .engine import GenRequest
 
         messages: List[Dict[str, Any]] = call_params["messages"]
         if not messages:
             raise ValueError("messages must be a non-empty list")
+        self._validate_call_params(call_params)
         model: str = call_params.get("model", self.config.model)
         n: int = int(call_params.get("n") or 1)
 
@@ -261,6 +284,7 @@ class LocalEngineClient:
         samplings = []
         for call_params in call_params_list:
             messages = call_params["messages"]
+            self._validate_call_params(call_params)
             n = int(call_params.get("n") or 1)
             sampling = SamplingParams(
                 temperature=call_params.get("temperature", 1.0),

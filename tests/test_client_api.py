@@ -400,3 +400,40 @@ def test_async_tool_call_consensus():
         k.close()
 
     asyncio.run(run())
+
+
+class TestParamValidation:
+    """OpenAI-documented sampling-param ranges raise locally (the reference
+    lets the remote API 400 these; a switched-over client sees the same)."""
+
+    def _client(self):
+        from kllms_amd import KLLMs
+        return KLLMs(model="tiny-llama", device="cpu", max_kv_blocks=128,
+                     use_hip_graphs=False, seed=0)
+
+    def test_out_of_range_params_raise(self):
+        c = self._client()
+        msgs = [{"role": "user", "content": "hi"}]
+        import pytest as _pytest
+        for bad in (dict(temperature=2.5), dict(temperature=-0.1),
+                    dict(top_p=1.5), dict(top_p=-0.01),
+                    dict(n=129), dict(max_tokens=0),
+                    dict(frequency_penalty=3.0), dict(presence_penalty=-2.5)):
+            kw = dict(max_tokens=4)
+            kw.update(bad)
+            with _pytest.raises(ValueError):
+                c.chat.completions.create(messages=msgs, model="tiny-llama", **kw)
+        # n=0 is NOT an error: the wrapper only forwards n>1 (reference
+        # completions.py:72 shape), so it behaves as the default n=1
+        out = c.chat.completions.create(messages=msgs, model="tiny-llama",
+                                        max_tokens=2, n=0)
+        assert len(out.choices) == 1
+
+    def test_boundary_values_accepted(self):
+        c = self._client()
+        msgs = [{"role": "user", "content": "hi"}]
+        out = c.chat.completions.create(messages=msgs, model="tiny-llama",
+                                        temperature=2.0, top_p=1.0, n=1,
+                                        max_tokens=2, frequency_penalty=2.0,
+                                        presence_penalty=-2.0)
+        assert out.choices
