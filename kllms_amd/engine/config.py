@@ -132,6 +132,13 @@ class EngineConfig(BaseModel):
     max_seq_len: int = 8192
     max_batch_size: int = 256
 
+    # embeddings.create backend: "token_mean" mean-pools the model's token
+    # embeddings (semantically meaningful once real weights are loaded);
+    # "ngram" is a deterministic signed char-3-gram hashing embedder whose
+    # cosine tracks STRING similarity with no weights at all; "auto" picks
+    # ngram under random init and token_mean when a checkpoint is loaded
+    embedding_mode: Literal["auto", "token_mean", "ngram"] = "auto"
+
     # Decode-step hipGraph capture
     use_hip_graphs: bool = True
     # constrained decoding: allow optional JSON whitespace between tokens
@@ -158,6 +165,20 @@ class EngineConfig(BaseModel):
 
     # Device ("cuda" is ROCm/HIP under torch-rocm; "cpu" for tests)
     device: Optional[str] = None
+
+    def effective_weights_dir(self) -> Optional[str]:
+        """The directory to load safetensors from: explicit weights_path, or
+        the model path itself when it is a directory holding safetensors
+        (README contract: model = preset name OR path to a weights dir).
+        None = deterministic random init."""
+        import glob as _glob
+        import os as _os
+
+        if self.weights_path:
+            return self.weights_path
+        if _os.path.isdir(str(self.model)) and _glob.glob(_os.path.join(self.model, "*.safetensors")):
+            return self.model
+        return None
 
     def resolve_arch(self) -> ModelArchConfig:
         if self.model in MODEL_PRESETS:

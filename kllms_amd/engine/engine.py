@@ -185,10 +185,11 @@ class LLMEngine:
 
         cls = MixtralForCausalLM if self.arch.arch == "mixtral" else LlamaForCausalLM
         model = cls(self.arch, self.ctx, dtype=self.dtype)
-        if self.config.weights_path:
+        wdir = self.config.effective_weights_dir()
+        if wdir:
             from .weights import load_safetensors_weights
 
-            load_safetensors_weights(model, self.config.weights_path, self.ctx)
+            load_safetensors_weights(model, wdir, self.ctx)
             model.to_device(self.device)
         else:
             model.to_device(self.device)
@@ -738,11 +739,47 @@ class LLMEngine:
         out = unit.cpu().tolist()
         return out, total
 
+    NGRAM_DIM = 512
+
+    def _embedding_mode(self) -> str:
+        mode = getattr(self.config, "embedding_mode", "auto")
+        if mode != "auto":
+            return mode
+        # random-init token embeddings are semantic noise;
+        # the signed n-gram hash tracks string similarity with no weights
+        return "token_mean" if self.config.effective_weights_dir() else "ngram"
+
+    def _embed_ngram_dev(self, texts: List[str]) -> Tuple[torch.Tensor, int]:
+        """Deterministic signed char-3-gram hashing embedder: each lowercased
+        text's padded 3-grams hash (BLAKE2b) into one of NGRAM_DIM buckets
+        with a ±1 sign bit; the L2-normalized count vector's cosine is a
+        Jaccard-like string similarity — semantically meaningful for the
+        consensus "embeddings" method without any model weights."""
+        import hashlib as _hl
+
+        out = torch.zeros(len(texts), self.NGRAM_DIM, dtype=torch.float32)
+        total_tokens = 0
+        for i, t in enumerate(texts):
+            total_tokens += len(self.tokenizer.encode(t))
+            if not t.strip():
+                continue  # zero row for empty texts, matching token_mean
+            s = f"  {t.lower()} "
+            row = out[i]
+            for j in range(len(s) - 2):
+                h = int.from_bytes(_hl.blake2b(s[j:j + 3].encode(), digest_size=4).digest(), "little")
+                row[h % self.NGRAM_DIM] += 1.0 if (h >> 20) & 1 else -1.0
+            n = row.norm()
+            if n > 1e-12:
+                row /= n
+        return out.to(self.device), total_tokens
+
     def embed_dev(self, texts: List[str]) -> Tuple[torch.Tensor, int]:
         """Device-resident variant of embed(): returns the [N, H] float32
         unit-vector tensor ON the engine device (zero rows for empty texts),
         so the consensus accel can run its cosine GEMM without a host round
         trip (consolidation similarity math on-device)."""
+        if self._embedding_mode() == "ngram":
+            return self._embed_ngram_dev(texts)
         emb = self.model.embed_tokens.weight
         H = emb.shape[1]
         ids_per_text = [self.tokenizer.encode(t) for t in texts]

@@ -372,3 +372,82 @@ class TestFp8ScaleProperties:
             amax = k.float().abs().amax(-1, keepdim=True)
             assert torch.isfinite(deq).all()
             assert (err <= amax * 0.0625 + 1e-6).all(), (trial, err.max(), amax.max())
+
+
+class TestEmbeddingModes:
+    def test_ngram_similarity_is_semantic(self):
+        """VERDICT r1 'missing' item 5: under random init the embedder must
+        still produce STRING-similarity-meaningful cosines. Signed char-3-gram
+        hashing: near-duplicate long strings ≫ unrelated strings."""
+        eng = LLMEngine(EngineConfig(model="tiny-llama", max_kv_blocks=64,
+                                     use_hip_graphs=False, device="cpu", seed=0))
+        assert eng._embedding_mode() == "ngram"  # auto + random init
+        texts = [
+            "The quarterly revenue increased by fifteen percent compared to last year",
+            "The quarterly revenue increased by fourteen percent compared to last year",
+            "Bananas are an excellent source of potassium and vitamin B6 for athletes",
+        ]
+        v, total = eng.embed_dev(texts)
+        sim = v @ v.T
+        assert sim[0, 1] > 0.8 > sim[0, 2], sim
+        assert total > 0
+        # deterministic; unit rows; zero row for empty text
+        v2, _ = eng.embed_dev(texts)
+        assert torch.equal(v, v2)
+        assert torch.allclose(v.norm(dim=1), torch.ones(3), atol=1e-5)
+        ve, _ = eng.embed_dev(["", "abc"])
+        assert ve[0].abs().sum() == 0
+
+    def test_mode_override_and_auto_with_weights(self, tmp_path):
+        pytest.importorskip("safetensors")
+        import sys
+        sys.path.insert(0, "tests")
+        from test_weights_io import _make_hf_llama_checkpoint
+
+        from kllms_amd.engine.config import MODEL_PRESETS
+
+        eng = LLMEngine(EngineConfig(model="tiny-llama", max_kv_blocks=64,
+                                     use_hip_graphs=False, device="cpu", seed=0,
+                                     embedding_mode="token_mean"))
+        assert eng._embedding_mode() == "token_mean"
+        v, _ = eng.embed_dev(["hello world"])
+        assert v.shape[1] == eng.model.embed_tokens.weight.shape[1]
+        # with a real checkpoint on disk, auto resolves to token_mean
+        cfg = MODEL_PRESETS["tiny-llama"]
+        _make_hf_llama_checkpoint(tmp_path, cfg)
+        import json as _json
+        (tmp_path / "config.json").write_text(_json.dumps({
+            "model_type": "llama", "vocab_size": cfg.vocab_size,
+            "hidden_size": cfg.hidden_size, "intermediate_size": cfg.intermediate_size,
+            "num_hidden_layers": cfg.num_layers, "num_attention_heads": cfg.num_heads,
+            "num_key_value_heads": cfg.num_kv_heads, "rope_theta": cfg.rope_theta,
+            "rms_norm_eps": cfg.rms_norm_eps, "max_position_embeddings": 512,
+        }))
+        eng2 = LLMEngine(EngineConfig(model=str(tmp_path), max_kv_blocks=64,
+                                      use_hip_graphs=False, device="cpu", seed=0))
+        assert eng2._embedding_mode() == "token_mean"
+
+    def test_model_dir_loads_weights_without_explicit_path(self, tmp_path):
+        """README contract: model may be a PATH to a weights dir — weights
+        must actually load (previously random-init unless weights_path set)."""
+        pytest.importorskip("safetensors")
+        import sys
+        sys.path.insert(0, "tests")
+        from test_weights_io import _make_hf_llama_checkpoint
+
+        from kllms_amd.engine.config import MODEL_PRESETS
+
+        cfg = MODEL_PRESETS["tiny-llama"]
+        tensors = _make_hf_llama_checkpoint(tmp_path, cfg)
+        import json as _json
+        (tmp_path / "config.json").write_text(_json.dumps({
+            "model_type": "llama", "vocab_size": cfg.vocab_size,
+            "hidden_size": cfg.hidden_size, "intermediate_size": cfg.intermediate_size,
+            "num_hidden_layers": cfg.num_layers, "num_attention_heads": cfg.num_heads,
+            "num_key_value_heads": cfg.num_kv_heads, "rope_theta": cfg.rope_theta,
+            "rms_norm_eps": cfg.rms_norm_eps, "max_position_embeddings": 512,
+        }))
+        eng = LLMEngine(EngineConfig(model=str(tmp_path), max_kv_blocks=64,
+                                     use_hip_graphs=False, device="cpu", seed=0))
+        got = eng.model.embed_tokens.weight.float()
+        assert torch.allclose(got, tensors["model.embed_tokens.weight"].to(got.dtype).float())
