@@ -151,8 +151,38 @@ class HFTokenizer(BaseTokenizer):
             return None
 
         self.bos_id = _first_id("<|begin_of_text|>", "<s>")
-        self.eos_id = _first_id("<|eot_id|>", "<|end_of_text|>", "</s>")
+        self.eos_id = _first_id("<|eot_id|>", "<|im_end|>", "<|end_of_text|>", "</s>", "<|endoftext|>")
         self.pad_id = self.eos_id if self.eos_id is not None else 0
+
+        # Real-checkpoint chat fidelity: HF checkpoints ship the model's own
+        # prompt format as a Jinja2 `chat_template` in tokenizer_config.json
+        # (Qwen ChatML, Llama-3 headers, ...). Load it when present; the
+        # hard-coded Llama-3 format stays the fallback.
+        self._chat_template = None
+        self._cfg_bos = self._cfg_eos = None
+        tc_path = os.path.join(os.path.dirname(os.path.abspath(tokenizer_json)),
+                               "tokenizer_config.json")
+        if os.path.exists(tc_path):
+            try:
+                with open(tc_path, "r", encoding="utf-8") as f:
+                    tc = _json.load(f)
+                ct = tc.get("chat_template")
+                if isinstance(ct, list):  # newer multi-template form
+                    ct = next((e.get("template") for e in ct
+                               if e.get("name") == "default"), None) or (
+                        ct[0].get("template") if ct else None)
+                if isinstance(ct, str) and ct.strip():
+                    self._chat_template = ct
+                def _tok_str(v):
+                    return v.get("content") if isinstance(v, dict) else v
+                self._cfg_bos = _tok_str(tc.get("bos_token"))
+                self._cfg_eos = _tok_str(tc.get("eos_token"))
+                # prefer the checkpoint's declared eos over the name guess
+                if self._cfg_eos in vocab:
+                    self.eos_id = vocab[self._cfg_eos]
+                    self.pad_id = self.eos_id
+            except Exception:
+                self._chat_template = None
 
         # id -> vocab surface string, and the set of special/added ids
         self._id_to_token = {i: t for t, i in vocab.items()}
@@ -173,6 +203,35 @@ class HFTokenizer(BaseTokenizer):
 
         self._byte_level = _has_bytelevel(spec.get("decoder")) or _has_bytelevel(spec.get("pre_tokenizer"))
         self._tb_cache: dict = {}
+
+    def apply_chat_template(self, messages: list[dict]) -> str:
+        if not self._chat_template:
+            return super().apply_chat_template(messages)
+        try:
+            import json as _json
+            from datetime import datetime
+
+            import jinja2
+
+            env = jinja2.Environment(trim_blocks=True, lstrip_blocks=True)
+            env.filters["tojson"] = lambda v, **kw: _json.dumps(v, **kw)
+
+            def raise_exception(msg):  # HF templates call this on bad input
+                raise jinja2.TemplateError(msg)
+
+            out = env.from_string(self._chat_template).render(
+                messages=messages,
+                add_generation_prompt=True,
+                bos_token=self._cfg_bos or "",
+                eos_token=self._cfg_eos or "",
+                raise_exception=raise_exception,
+                strftime_now=lambda fmt: datetime.now().strftime(fmt),
+                tools=None,
+            )
+            return out
+        except Exception:
+            # template requires features we don't model (tool schemas, ...)
+            return super().apply_chat_template(messages)
 
     def encode(self, text: str, add_bos: bool = False) -> list[int]:
         ids = self._tok.encode(text, add_special_tokens=False).ids

@@ -1,6 +1,7 @@
 """Tokenizer tests: byte-level round trips, chat templating, crop semantics
 (the local equivalents of the reference's tiktoken usage)."""
 
+import os
 import pytest
 
 hypothesis = pytest.importorskip("hypothesis")
@@ -60,3 +61,69 @@ class TestCrop:
 
     def test_crop_zero(self, tok):
         assert tok.encode(tok.crop_to_tokens("abc", 0)) == []
+
+
+class TestHFChatTemplate:
+    """HF tokenizer_config.json chat_template (Jinja2) drives the prompt
+    format for real checkpoints; the Llama-3 format stays the fallback."""
+
+    CHATML = (
+        "{% for message in messages %}"
+        "{{ '<|im_start|>' + message['role'] + '\n' + message['content'] + '<|im_end|>' + '\n' }}"
+        "{% endfor %}"
+        "{% if add_generation_prompt %}{{ '<|im_start|>assistant\n' }}{% endif %}"
+    )
+
+    def _fixture_dir(self, tmp_path):
+        import shutil
+        src = os.path.join(os.path.dirname(__file__), "data", "bpe_tokenizer.json")
+        shutil.copy(src, tmp_path / "tokenizer.json")
+        return tmp_path
+
+    def test_template_rendered(self, tmp_path):
+        import json
+        d = self._fixture_dir(tmp_path)
+        (d / "tokenizer_config.json").write_text(json.dumps({
+            "chat_template": self.CHATML,
+            "eos_token": "<|eot_id|>",
+        }))
+        from kllms_amd.engine.tokenizer import HFTokenizer
+        tok = HFTokenizer(str(d / "tokenizer.json"))
+        out = tok.apply_chat_template([
+            {"role": "system", "content": "be brief"},
+            {"role": "user", "content": "hi"},
+        ])
+        assert out == ("<|im_start|>system\nbe brief<|im_end|>\n"
+                       "<|im_start|>user\nhi<|im_end|>\n"
+                       "<|im_start|>assistant\n")
+
+    def test_no_config_falls_back_to_llama3(self, tmp_path):
+        d = self._fixture_dir(tmp_path)
+        from kllms_amd.engine.tokenizer import HFTokenizer
+        tok = HFTokenizer(str(d / "tokenizer.json"))
+        out = tok.apply_chat_template([{"role": "user", "content": "x"}])
+        assert "<|start_header_id|>user<|end_header_id|>" in out
+
+    def test_broken_template_falls_back(self, tmp_path):
+        import json
+        d = self._fixture_dir(tmp_path)
+        (d / "tokenizer_config.json").write_text(json.dumps({
+            "chat_template": "{{ raise_exception('nope') }}",
+        }))
+        from kllms_amd.engine.tokenizer import HFTokenizer
+        tok = HFTokenizer(str(d / "tokenizer.json"))
+        out = tok.apply_chat_template([{"role": "user", "content": "x"}])
+        assert "<|start_header_id|>" in out  # fell back, did not crash
+
+    def test_declared_eos_overrides_name_guess(self, tmp_path):
+        import json
+        d = self._fixture_dir(tmp_path)
+        from kllms_amd.engine.tokenizer import HFTokenizer
+        base = HFTokenizer(str(d / "tokenizer.json"))
+        vocab_eos = base._id_to_token.get(base.eos_id)
+        # declare a DIFFERENT token from the vocab as eos
+        other = next(t for t, i in base._tok.get_vocab().items()
+                     if t.startswith("<|") and t != vocab_eos and t.endswith("|>"))
+        (d / "tokenizer_config.json").write_text(json.dumps({"eos_token": other}))
+        tok = HFTokenizer(str(d / "tokenizer.json"))
+        assert tok._id_to_token[tok.eos_id] == other
