@@ -10,7 +10,7 @@ Llama-3-70B (TP=8), Mixtral-8x7B (MoE), plus tiny CPU-testable variants.
 
 from __future__ import annotations
 
-from typing import Literal, Optional
+from typing import List, Literal, Optional, Union
 
 from pydantic import BaseModel, Field
 
@@ -161,7 +161,10 @@ class EngineConfig(BaseModel):
 
     # Generation defaults
     default_max_new_tokens: int = 512
-    eos_token_id: Optional[int] = None  # default: tokenizer's
+    # default: the checkpoint's generation_config.json eos_token_id (which
+    # may be a LIST — Llama-3 stops on <|eot_id|> OR <|end_of_text|>),
+    # else the tokenizer's
+    eos_token_id: Optional[Union[int, List[int]]] = None
 
     # Device ("cuda" is ROCm/HIP under torch-rocm; "cpu" for tests)
     device: Optional[str] = None

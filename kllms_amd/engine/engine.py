@@ -161,7 +161,18 @@ class LLMEngine:
             config.max_seq_len = self.arch.max_position_embeddings
 
         self.tokenizer: BaseTokenizer = load_tokenizer(config.model, config.weights_path, self.arch.vocab_size)
-        self.eos_token_id = config.eos_token_id if config.eos_token_id is not None else self.tokenizer.eos_id
+        eos_cfg = config.eos_token_id
+        if eos_cfg is None:
+            eos_cfg = self._generation_config_eos()
+        if eos_cfg is None:
+            eos_cfg = self.tokenizer.eos_id
+        # a checkpoint may declare SEVERAL stop ids (generation_config.json
+        # eos_token_id list, e.g. Llama-3's [<|end_of_text|>, <|eot_id|>]);
+        # eos_token_id stays the primary (used where ONE id is needed, e.g.
+        # the constrained-decode DFA's accepting-state bit)
+        ids = eos_cfg if isinstance(eos_cfg, (list, tuple)) else [eos_cfg]
+        self.eos_token_id = ids[0]
+        self.eos_token_ids = frozenset(i for i in ids if i is not None)
 
         self.model = self._build_model()
         self.kv = self._build_kv_cache()
@@ -179,6 +190,25 @@ class LLMEngine:
             self._graph_runner = DecodeGraphRunner(self, config.hip_graph_batch_sizes)
 
     # --- construction --------------------------------------------------------
+    def _generation_config_eos(self):
+        """eos_token_id from the checkpoint's generation_config.json (int or
+        list), or None when absent / random init."""
+        import json as _json
+        import os as _os
+
+        wdir = self.config.effective_weights_dir() or (
+            self.config.model if _os.path.isdir(str(self.config.model)) else None)
+        if not wdir:
+            return None
+        p = _os.path.join(wdir, "generation_config.json")
+        if not _os.path.exists(p):
+            return None
+        try:
+            with open(p) as f:
+                return _json.load(f).get("eos_token_id")
+        except Exception:
+            return None
+
     def _build_model(self):
         from ..models.llama import LlamaForCausalLM
         from ..models.mixtral import MixtralForCausalLM
@@ -702,7 +732,7 @@ class LLMEngine:
 
     def _check_stop(self, s: _Stream) -> None:
         max_new = s.sampling.max_tokens or self.config.default_max_new_tokens
-        if s.last_token == self.eos_token_id:
+        if s.last_token in self.eos_token_ids:
             s.out.token_ids.pop()  # EOS itself is not part of the content
             s.out.logprobs.pop()
             if len(s.out.top_logprobs) > len(s.out.token_ids):

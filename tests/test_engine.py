@@ -451,3 +451,51 @@ class TestEmbeddingModes:
                                      use_hip_graphs=False, device="cpu", seed=0))
         got = eng.model.embed_tokens.weight.float()
         assert torch.allclose(got, tensors["model.embed_tokens.weight"].to(got.dtype).float())
+
+
+class TestMultiEos:
+    def test_config_list_and_generation_config(self, tmp_path):
+        """A checkpoint's generation_config.json eos_token_id list (Llama-3
+        style) makes the engine stop on ANY of the ids."""
+        eng = LLMEngine(EngineConfig(model="tiny-llama", max_kv_blocks=64,
+                                     use_hip_graphs=False, device="cpu", seed=0,
+                                     eos_token_id=[7, 9]))
+        assert eng.eos_token_id == 7 and eng.eos_token_ids == frozenset({7, 9})
+
+        # generation_config.json next to a weights dir is honored
+        pytest.importorskip("safetensors")
+        import json as _json
+        import sys
+        sys.path.insert(0, "tests")
+        from test_weights_io import _make_hf_llama_checkpoint
+
+        from kllms_amd.engine.config import MODEL_PRESETS
+        cfg = MODEL_PRESETS["tiny-llama"]
+        _make_hf_llama_checkpoint(tmp_path, cfg)
+        (tmp_path / "config.json").write_text(_json.dumps({
+            "model_type": "llama", "vocab_size": cfg.vocab_size,
+            "hidden_size": cfg.hidden_size, "intermediate_size": cfg.intermediate_size,
+            "num_hidden_layers": cfg.num_layers, "num_attention_heads": cfg.num_heads,
+            "num_key_value_heads": cfg.num_kv_heads, "max_position_embeddings": 512,
+        }))
+        (tmp_path / "generation_config.json").write_text(_json.dumps({
+            "eos_token_id": [11, 23]}))
+        eng2 = LLMEngine(EngineConfig(model=str(tmp_path), max_kv_blocks=64,
+                                      use_hip_graphs=False, device="cpu", seed=0))
+        assert eng2.eos_token_ids == frozenset({11, 23})
+
+    def test_secondary_eos_stops_stream(self):
+        """Force the model's greedy next token to be a SECONDARY eos id and
+        check the stream finishes with reason 'stop'."""
+        eng = LLMEngine(EngineConfig(model="tiny-llama", max_kv_blocks=64,
+                                     use_hip_graphs=False, device="cpu", seed=0))
+        probe = eng.generate([GenRequest(prompt_ids=[1, 2, 3], n=1, sampling=greedy(1))])[0]
+        tok = probe.streams[0].token_ids
+        if not tok:
+            pytest.skip("first greedy token was already eos")
+        eng2 = LLMEngine(EngineConfig(model="tiny-llama", max_kv_blocks=64,
+                                      use_hip_graphs=False, device="cpu", seed=0,
+                                      eos_token_id=[eng.eos_token_id, tok[0]]))
+        out = eng2.generate([GenRequest(prompt_ids=[1, 2, 3], n=1, sampling=greedy(8))])[0]
+        assert out.streams[0].finish_reason == "stop"
+        assert out.streams[0].token_ids == []  # eos content excluded
