@@ -293,6 +293,8 @@ class LocalEngineClient:
                 max_tokens=call_params.get("max_tokens"),
                 stop=call_params.get("stop"),
                 seed=call_params.get("seed"),
+                frequency_penalty=call_params.get("frequency_penalty", 0.0),
+                presence_penalty=call_params.get("presence_penalty", 0.0),
                 logprobs=bool(call_params.get("logprobs", False)),
                 top_logprobs=int(call_params.get("top_logprobs") or 0),
                 logit_bias={int(k): float(v) for k, v in call_params["logit_bias"].items()}

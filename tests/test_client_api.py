@@ -437,3 +437,23 @@ class TestParamValidation:
                                         max_tokens=2, frequency_penalty=2.0,
                                         presence_penalty=-2.0)
         assert out.choices
+
+
+def test_create_many_honors_penalties():
+    """Regression: the packed-batch path (chat_completions_create_many, the
+    bench/serving entry) silently dropped frequency/presence penalties."""
+    from kllms_amd import KLLMs
+
+    c = KLLMs(model="tiny-llama", device="cpu", max_kv_blocks=256,
+              use_hip_graphs=False, seed=0).client
+    base = dict(messages=[{"role": "user", "content": "repeat"}],
+                model="tiny-llama", n=1, temperature=0.0, max_tokens=12, seed=1)
+    plain = c.chat_completions_create_many([dict(base)])[0]
+    penal = c.chat_completions_create_many([dict(base, frequency_penalty=1.9,
+                                                 presence_penalty=1.9)])[0]
+    # greedy with strong penalties must diverge from unpenalized greedy
+    assert plain.choices[0].message.content != penal.choices[0].message.content
+    # and must match the single-request path with the same penalties
+    single = c.chat_completions_create(**dict(base, frequency_penalty=1.9,
+                                              presence_penalty=1.9))
+    assert penal.choices[0].message.content == single.choices[0].message.content
