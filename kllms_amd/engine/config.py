@@ -32,6 +32,9 @@ class ModelArchConfig(BaseModel):
     tie_word_embeddings: bool = False
     # Qwen2-family: bias on the fused QKV projection (o_proj stays bias-free)
     attention_qkv_bias: bool = False
+    # HF rope_scaling dict (Llama-3.1 "llama3" banded NTK, "linear");
+    # None = plain RoPE
+    rope_scaling: Optional[dict] = None
     # MoE (mixtral only)
     num_experts: int = 0
     num_experts_per_tok: int = 2
@@ -219,4 +222,5 @@ class EngineConfig(BaseModel):
             # bias; Llama-style configs say so explicitly
             attention_qkv_bias=hf.get(
                 "attention_bias", hf.get("model_type", "").lower() == "qwen2"),
+            rope_scaling=hf.get("rope_scaling"),
         )

@@ -160,7 +160,8 @@ class LlamaForCausalLM(nn.Module):
         self.lm_head.weight.requires_grad_(False)
         self.register_buffer(
             "cos_sin",
-            build_cos_sin_cache(cfg.head_dim_, cfg.max_position_embeddings, cfg.rope_theta, "cpu"),
+            build_cos_sin_cache(cfg.head_dim_, cfg.max_position_embeddings, cfg.rope_theta, "cpu",
+                                rope_scaling=cfg.rope_scaling),
             persistent=False,
         )
 

@@ -37,9 +37,32 @@ def fused_add_rmsnorm(
 
 # --- rotary embedding ---------------------------------------------------------
 
-def build_cos_sin_cache(head_dim: int, max_pos: int, theta: float, device, dtype=torch.float32) -> torch.Tensor:
+def build_cos_sin_cache(head_dim: int, max_pos: int, theta: float, device,
+                        dtype=torch.float32, rope_scaling=None) -> torch.Tensor:
     half = head_dim // 2
     inv_freq = 1.0 / (theta ** (torch.arange(0, half, dtype=torch.float64, device=device) / half))
+    if rope_scaling:
+        rt = rope_scaling.get("rope_type") or rope_scaling.get("type")
+        if rt == "llama3":
+            # Llama-3.1 frequency-banded NTK scaling: low-frequency bands
+            # (wavelength > old context) are divided by `factor`; high-
+            # frequency bands are untouched; the band between interpolates.
+            import math as _math
+
+            factor = float(rope_scaling.get("factor", 8.0))
+            lo_f = float(rope_scaling.get("low_freq_factor", 1.0))
+            hi_f = float(rope_scaling.get("high_freq_factor", 4.0))
+            old_ctx = float(rope_scaling.get("original_max_position_embeddings", 8192))
+            wavelen = 2 * _math.pi / inv_freq
+            lo_wl = old_ctx / lo_f
+            hi_wl = old_ctx / hi_f
+            smooth = ((old_ctx / wavelen - lo_f) / (hi_f - lo_f)).clamp(0.0, 1.0)
+            scaled = (1 - smooth) * inv_freq / factor + smooth * inv_freq
+            inv_freq = torch.where(wavelen > lo_wl, inv_freq / factor,
+                                   torch.where(wavelen < hi_wl, inv_freq, scaled))
+        elif rt == "linear":
+            inv_freq = inv_freq / float(rope_scaling.get("factor", 1.0))
+        # unknown types: serve unscaled rather than failing (documented)
     t = torch.arange(max_pos, dtype=torch.float64, device=device)
     freqs = torch.outer(t, inv_freq)
     return torch.cat([freqs.cos(), freqs.sin()], dim=-1).to(dtype)

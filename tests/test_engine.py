@@ -499,3 +499,61 @@ class TestMultiEos:
         out = eng2.generate([GenRequest(prompt_ids=[1, 2, 3], n=1, sampling=greedy(8))])[0]
         assert out.streams[0].finish_reason == "stop"
         assert out.streams[0].token_ids == []  # eos content excluded
+
+
+class TestRopeScaling:
+    def test_llama3_banded_scaling_formula(self):
+        """Llama-3.1 rope_scaling: low-frequency bands /factor, high-frequency
+        untouched, smooth interpolation between — checked against a direct
+        transcription of the published formula."""
+        import math
+
+        from kllms_amd.ops.torch_ref import build_cos_sin_cache
+
+        D, P, theta = 128, 64, 500000.0
+        rs = {"rope_type": "llama3", "factor": 8.0, "low_freq_factor": 1.0,
+              "high_freq_factor": 4.0, "original_max_position_embeddings": 8192}
+        got = build_cos_sin_cache(D, P, theta, "cpu", rope_scaling=rs)
+        plain = build_cos_sin_cache(D, P, theta, "cpu")
+        assert not torch.equal(got, plain)
+
+        half = D // 2
+        inv = 1.0 / (theta ** (torch.arange(0, half, dtype=torch.float64) / half))
+        out = []
+        for f in inv.tolist():
+            wl = 2 * math.pi / f
+            if wl > 8192 / 1.0:
+                out.append(f / 8.0)
+            elif wl < 8192 / 4.0:
+                out.append(f)
+            else:
+                s = (8192 / wl - 1.0) / (4.0 - 1.0)
+                out.append((1 - s) * f / 8.0 + s * f)
+        t = torch.arange(P, dtype=torch.float64)
+        freqs = torch.outer(t, torch.tensor(out, dtype=torch.float64))
+        want = torch.cat([freqs.cos(), freqs.sin()], dim=-1).float()
+        assert torch.allclose(got, want, atol=1e-6)
+
+        # high-frequency (early) bands must be IDENTICAL to unscaled
+        hi_cols = [i for i, f in enumerate(inv.tolist()) if 2 * math.pi / f < 8192 / 4.0]
+        assert hi_cols, "expected some high-frequency bands"
+        assert torch.allclose(got[:, hi_cols], plain[:, hi_cols], atol=1e-7)
+
+    def test_linear_scaling_and_engine_wiring(self, tmp_path):
+        from kllms_amd.ops.torch_ref import build_cos_sin_cache
+
+        lin = build_cos_sin_cache(64, 32, 10000.0, "cpu",
+                                  rope_scaling={"type": "linear", "factor": 2.0})
+        plain = build_cos_sin_cache(64, 64, 10000.0, "cpu")
+        # position 2p under factor-2 linear == position p unscaled
+        assert torch.allclose(lin[2], plain[1], atol=1e-6)
+
+        # arch config threads rope_scaling into the model's cache
+        from kllms_amd.engine.config import MODEL_PRESETS
+        from kllms_amd.models.llama import LlamaForCausalLM
+        from kllms_amd.parallel.tp import ParallelContext
+        cfg = MODEL_PRESETS["tiny-llama"].model_copy(update={
+            "rope_scaling": {"type": "linear", "factor": 2.0}})
+        m = LlamaForCausalLM(cfg, ParallelContext(), dtype=torch.float32)
+        m2 = LlamaForCausalLM(MODEL_PRESETS["tiny-llama"], ParallelContext(), dtype=torch.float32)
+        assert not torch.equal(m.cos_sin, m2.cos_sin)
