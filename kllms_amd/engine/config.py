@@ -50,6 +50,14 @@ MODEL_PRESETS: dict[str, ModelArchConfig] = {
         arch="llama", vocab_size=128256, hidden_size=4096, intermediate_size=14336,
         num_layers=32, num_heads=32, num_kv_heads=8, rope_theta=500000.0,
     ),
+    # Llama-3.1: same dims, banded-NTK rope scaling to 128k context
+    "llama-3.1-8b": ModelArchConfig(
+        arch="llama", vocab_size=128256, hidden_size=4096, intermediate_size=14336,
+        num_layers=32, num_heads=32, num_kv_heads=8, rope_theta=500000.0,
+        max_position_embeddings=131072,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0, "low_freq_factor": 1.0,
+                      "high_freq_factor": 4.0, "original_max_position_embeddings": 8192},
+    ),
     "llama-3-70b": ModelArchConfig(
         arch="llama", vocab_size=128256, hidden_size=8192, intermediate_size=28672,
         num_layers=80, num_heads=64, num_kv_heads=8, rope_theta=500000.0,
