@@ -170,9 +170,12 @@ class LLMEngine:
         # eos_token_id list, e.g. Llama-3's [<|end_of_text|>, <|eot_id|>]);
         # eos_token_id stays the primary (used where ONE id is needed, e.g.
         # the constrained-decode DFA's accepting-state bit)
-        ids = eos_cfg if isinstance(eos_cfg, (list, tuple)) else [eos_cfg]
+        ids = list(eos_cfg) if isinstance(eos_cfg, (list, tuple)) else [eos_cfg]
+        ids = [i for i in ids if i is not None]
+        if not ids:  # e.g. an empty list in generation_config.json
+            ids = [self.tokenizer.eos_id]
         self.eos_token_id = ids[0]
-        self.eos_token_ids = frozenset(i for i in ids if i is not None)
+        self.eos_token_ids = frozenset(ids)
 
         self.model = self._build_model()
         self.kv = self._build_kv_cache()
