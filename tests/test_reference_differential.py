@@ -574,3 +574,23 @@ def test_numerical_similarity_matches_reference(ref, a, b):
     from kllms_amd.consensus.similarity import numerical_similarity
 
     assert numerical_similarity(a, b) == pytest.approx(ref.numerical_similarity(a, b), abs=1e-12)
+
+
+@settings(max_examples=100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(_record_lists)
+def test_fuzzy_key_selection_matches_reference(ref_key, lists_of_records):
+    """C43: canonicalized (fuzzy) key cascade and the fuzzy-fallback picker
+    must agree with the reference, incl. the stability-tuple tie-break."""
+    import _refk.fuzzy_key_selection as ref_fuzzy  # type: ignore[import-not-found]
+
+    from kllms_amd.consensus import fuzzy_key_selection as our_fuzzy
+
+    records = [r for lst in lists_of_records for r in lst]
+    try:
+        want = ref_fuzzy.select_best_keys_with_fuzzy_fallback(list(records))
+    except Exception as e:
+        with pytest.raises(type(e)):
+            our_fuzzy.select_best_keys_with_fuzzy_fallback(list(records))
+        return
+    got = our_fuzzy.select_best_keys_with_fuzzy_fallback(list(records))
+    _deep_eq(got.model_dump(), want.model_dump(), "fuzzy_selected_keys")
