@@ -457,3 +457,14 @@ def test_create_many_honors_penalties():
     single = c.chat_completions_create(**dict(base, frequency_penalty=1.9,
                                               presence_penalty=1.9))
     assert penal.choices[0].message.content == single.choices[0].message.content
+
+
+def test_cli_entry_help():
+    import subprocess
+    import sys
+    r = subprocess.run([sys.executable, "-m", "kllms_amd", "help"],
+                       capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0 and "serve" in r.stdout and "bench" in r.stdout
+    r2 = subprocess.run([sys.executable, "-m", "kllms_amd", "nonsense"],
+                        capture_output=True, text=True, timeout=120)
+    assert r2.returncode == 2
