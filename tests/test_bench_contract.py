@@ -43,3 +43,33 @@ def test_bench_json_contract(tmp_path):
               "batch_wall_s"):
         assert k in cfg, k
     assert cfg["parallelism"] == "dp1 tp1"
+
+
+@pytest.mark.timeout(600)
+def test_bench_driver_multiproc_invocation(tmp_path):
+    """The EXACT shape the driver uses for SCALE_rNN: torch.distributed.run
+    with N ranks — on CPU this exercises the gloo/dp path end to end and
+    must still print exactly ONE JSON line (rank 0) with the whole-job
+    aggregate value."""
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29841", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "0", "--model", "tiny-llama",
+         "--batch", "2", "--max-new", "6", "--prompt-len", "48"],
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        capture_output=True, text=True, timeout=500, env=env,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines()
+             if l.strip().startswith("{")]
+    assert len(lines) == 1, f"exactly ONE json line expected: {out.stdout[-800:]}"
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2
+    assert d["scaling"] == "weak"
+    assert d["config"]["parallelism"].startswith("dp2")
+    # whole-job aggregate: 2 ranks x per-rank batch
+    assert d["config"]["global_batch"] == 4
+    assert d["value"] > 0
