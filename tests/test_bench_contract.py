@@ -73,3 +73,29 @@ def test_bench_driver_multiproc_invocation(tmp_path):
     # whole-job aggregate: 2 ranks x per-rank batch
     assert d["config"]["global_batch"] == 4
     assert d["value"] > 0
+
+
+@pytest.mark.timeout(600)
+def test_bench_tp_serving_invocation(tmp_path):
+    """bench.py --parallel tp: rank 0 drives the scheduler, rank 1 replays
+    (TPCoordinator/TPFollower) — the full TP serving stack under the bench
+    entry, on CPU/gloo."""
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29843", "bench.py", "--gpus", "2",
+         "--parallel", "tp", "--steps", "2", "--warmup", "0",
+         "--model", "tiny-llama", "--batch", "2", "--max-new", "6",
+         "--prompt-len", "48"],
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        capture_output=True, text=True, timeout=500, env=env,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines()
+             if l.strip().startswith("{")]
+    assert len(lines) == 1, f"exactly ONE json line expected: {out.stdout[-800:]}"
+    d = json.loads(lines[0])
+    assert d["config"]["parallelism"] == "tp2"
+    assert d["value"] > 0
