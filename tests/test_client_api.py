@@ -468,3 +468,17 @@ def test_cli_entry_help():
     r2 = subprocess.run([sys.executable, "-m", "kllms_amd", "nonsense"],
                         capture_output=True, text=True, timeout=120)
     assert r2.returncode == 2
+
+
+def test_key_aligner_through_client():
+    """consensus_aligner="key" (the reference's commented-in L1b engine)
+    must serve end to end through the public client."""
+    from kllms_amd import KLLMs
+
+    c = KLLMs(model="tiny-llama", device="cpu", max_kv_blocks=256,
+              use_hip_graphs=False, seed=0, consensus_aligner="key")
+    out = c.chat.completions.create(
+        messages=[{"role": "user", "content": "list items"}],
+        model="tiny-llama", n=3, temperature=0.9, max_tokens=10, seed=5)
+    assert len(out.choices) == 4 and out.choices[0].index == 0
+    assert out.likelihoods is not None
