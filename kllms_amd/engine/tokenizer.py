@@ -212,14 +212,18 @@ class HFTokenizer(BaseTokenizer):
             from datetime import datetime
 
             import jinja2
+            from jinja2.sandbox import ImmutableSandboxedEnvironment
 
-            env = jinja2.Environment(trim_blocks=True, lstrip_blocks=True)
-            env.filters["tojson"] = lambda v, **kw: _json.dumps(v, **kw)
+            if getattr(self, "_compiled_template", None) is None:
+                # sandboxed + trim/lstrip to match transformers' rendering
+                env = ImmutableSandboxedEnvironment(trim_blocks=True, lstrip_blocks=True)
+                env.filters["tojson"] = lambda v, **kw: _json.dumps(v, **kw)
+                self._compiled_template = env.from_string(self._chat_template)
 
             def raise_exception(msg):  # HF templates call this on bad input
                 raise jinja2.TemplateError(msg)
 
-            out = env.from_string(self._chat_template).render(
+            out = self._compiled_template.render(
                 messages=messages,
                 add_generation_prompt=True,
                 bos_token=self._cfg_bos or "",
