@@ -31,3 +31,17 @@ def fake_embed():
         return out
 
     return embed
+
+
+# Deep-fuzz profile: KLLMS_FUZZ_DEEP=<mult> multiplies every test's
+# max_examples (used for long validation campaigns; default off)
+try:
+    from hypothesis import settings as _hyp_settings
+
+    _mult = int(os.environ.get("KLLMS_FUZZ_DEEP", "0"))
+    if _mult > 1:
+        _hyp_settings.register_profile(
+            "deep", max_examples=_mult * 100, deadline=None, print_blob=True)
+        _hyp_settings.load_profile("deep")
+except Exception:
+    pass

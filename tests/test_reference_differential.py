@@ -153,7 +153,11 @@ json_values = st.recursive(
 )
 
 
-@settings(max_examples=150, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+# KLLMS_FUZZ_DEEP=<mult> scales every property's example count for long
+# validation campaigns (the explicit @settings below absorb the multiplier)
+_DEEP = max(1, int(os.environ.get("KLLMS_FUZZ_DEEP", "1")))
+
+@settings(max_examples=_DEEP * 150, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(json_values, json_values)
 def test_generic_similarity_matches_reference(ref, a, b):
     ours = generic_similarity(a, b, "levenshtein", fake_embed)
@@ -161,7 +165,7 @@ def test_generic_similarity_matches_reference(ref, a, b):
     assert ours == pytest.approx(theirs, abs=1e-9), f"{a!r} vs {b!r}"
 
 
-@settings(max_examples=120, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 120, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(st.lists(json_values, min_size=1, max_size=5))
 def test_consensus_values_matches_reference(ref, values):
     ours_settings = ConsensusSettings(string_similarity_method="levenshtein")
@@ -179,7 +183,7 @@ def test_consensus_values_matches_reference(ref, values):
     _deep_eq(got_conf, want_conf, "confidence")
 
 
-@settings(max_examples=80, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 80, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(st.lists(json_values, min_size=2, max_size=4))
 def test_recursive_alignment_matches_reference(ref, values):
     try:
@@ -194,7 +198,7 @@ def test_recursive_alignment_matches_reference(ref, values):
     _deep_eq(got_km, want_km, "key_mappings")
 
 
-@settings(max_examples=80, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 80, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(st.lists(st.lists(ascii_text, max_size=4), min_size=2, max_size=4))
 def test_lists_alignment_matches_reference(ref, lists):
     def ours_sim(a, b):
@@ -231,7 +235,7 @@ clustered_numbers = st.lists(
 )
 
 
-@settings(max_examples=200, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 200, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(clustered_numbers)
 def test_numeric_consensus_matches_reference(ref, values):
     """Hybrid numeric clustering incl. tie-breaks by cross-cluster support,
@@ -246,7 +250,7 @@ def test_numeric_consensus_matches_reference(ref, values):
     _deep_eq(got[1], want[1], "confidence")
 
 
-@settings(max_examples=150, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 150, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(json_values, json_values, st.sampled_from(["jaccard", "hamming", "embeddings"]))
 def test_similarity_methods_match_reference(ref, a, b, method):
     """All four string-similarity methods, incl. the >50-char embeddings gate
@@ -256,7 +260,7 @@ def test_similarity_methods_match_reference(ref, a, b, method):
     assert ours == pytest.approx(theirs, abs=1e-9), f"{method}: {a!r} vs {b!r}"
 
 
-@settings(max_examples=60, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 60, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(st.lists(st.text(alphabet="abcdefgh ", min_size=45, max_size=70), min_size=2, max_size=4))
 def test_long_string_embeddings_consensus_matches_reference(ref, values):
     """string_consensus 'centroid' on long strings routes through the
@@ -349,7 +353,7 @@ content_strings = st.one_of(
 )
 
 
-@settings(max_examples=100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(st.lists(content_strings, min_size=1, max_size=5))
 def test_consolidate_chat_completions_matches_reference(ref_consolidation, contents):
     from kllms_amd.consensus.consolidation import consolidate_chat_completions
@@ -377,7 +381,7 @@ def test_consolidate_chat_completions_matches_reference(ref_consolidation, conte
     assert got.usage.model_dump() == want.usage.model_dump()
 
 
-@settings(max_examples=60, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 60, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(st.lists(content_strings, min_size=1, max_size=4))
 def test_async_consolidation_matches_reference(ref_consolidation, contents):
     """The reference's ASYNC consolidation is hand-duplicated code (not a
@@ -458,7 +462,7 @@ _record = st.dictionaries(
 _record_lists = st.lists(st.lists(_record, max_size=4), min_size=2, max_size=4)
 
 
-@settings(max_examples=100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(st.one_of(_record_lists, st.lists(json_values, min_size=2, max_size=4)))
 def test_key_based_recursive_align_matches_reference(ref_key, values):
     from kllms_amd.consensus.key_based_alignment import recursive_align
@@ -473,7 +477,7 @@ def test_key_based_recursive_align_matches_reference(ref_key, values):
     _deep_eq(got_km, want_km, "key_mappings")
 
 
-@settings(max_examples=100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(_record_lists)
 def test_key_selection_matches_reference(ref_key, lists_of_records):
     """The cascade's chosen alignment keys must match (C42/C43)."""
@@ -494,7 +498,7 @@ def test_key_selection_matches_reference(ref_key, lists_of_records):
     _deep_eq(got.model_dump(), want.model_dump(), "selected_keys")
 
 
-@settings(max_examples=100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(st.lists(json_values, min_size=1, max_size=5))
 def test_compute_similarity_scores_matches_reference(ref, values):
     from kllms_amd.consensus.similarity import compute_similarity_scores
@@ -508,7 +512,7 @@ def test_compute_similarity_scores_matches_reference(ref, values):
     _deep_eq(list(got), list(want), "scores")
 
 
-@settings(max_examples=100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(st.lists(st.lists(st.integers(min_value=0, max_value=9), min_size=1, max_size=5),
                 min_size=2, max_size=4))
 def test_sort_by_original_majority_matches_reference(ref, originals):
@@ -532,7 +536,7 @@ def test_sort_by_original_majority_matches_reference(ref, originals):
     _deep_eq(list(got[1]), list(want[1]), "orig_idx")
 
 
-@settings(max_examples=120, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 120, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(json_values)
 def test_intermediary_cleanup_matches_reference(ref, value):
     from kllms_amd.consensus.values import intermediary_consensus_cleanup
@@ -544,7 +548,7 @@ def test_intermediary_cleanup_matches_reference(ref, value):
     )
 
 
-@settings(max_examples=120, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 120, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(st.lists(st.floats(min_value=0.0, max_value=1.0, allow_nan=False), max_size=10))
 def test_outlier_cutoff_matches_reference(ref, data):
     from kllms_amd.consensus import low_cutoff_bound, remove_outliers
@@ -553,7 +557,7 @@ def test_outlier_cutoff_matches_reference(ref, data):
     _deep_eq(remove_outliers(list(data)), ref.remove_outliers(list(data)), "outliers")
 
 
-@settings(max_examples=200, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 200, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(ascii_text)
 def test_string_normalization_matches_reference(ref, s):
     """normalize_string + sanitize_value: the voting/similarity equivalence
@@ -565,7 +569,7 @@ def test_string_normalization_matches_reference(ref, s):
     assert sanitize_value(s) == ref.sanitize_value(s)
 
 
-@settings(max_examples=200, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 200, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(st.one_of(st.booleans(), st.integers(-10**6, 10**6),
                  st.floats(allow_nan=False, allow_infinity=False)),
        st.one_of(st.booleans(), st.integers(-10**6, 10**6),
@@ -576,7 +580,7 @@ def test_numerical_similarity_matches_reference(ref, a, b):
     assert numerical_similarity(a, b) == pytest.approx(ref.numerical_similarity(a, b), abs=1e-12)
 
 
-@settings(max_examples=100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@settings(max_examples=_DEEP * 100, deadline=None, suppress_health_check=[HealthCheck.too_slow])
 @given(_record_lists)
 def test_fuzzy_key_selection_matches_reference(ref_key, lists_of_records):
     """C43: canonicalized (fuzzy) key cascade and the fuzzy-fallback picker
