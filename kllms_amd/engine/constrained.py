@@ -105,10 +105,90 @@ def _json_string_ir(min_len: int = 0, max_len: Optional[int] = None) -> _Node:
     return Seq([Lit(b'"'), _bounded(char, min_len, max_len), Lit(b'"')])
 
 
+def _digits_any(n: int) -> _Node:
+    return Seq([Cls(_DIGITS) for _ in range(n)]) if n else Lit(b"")
+
+
+def _ge_str(s: bytes) -> _Node:
+    """len(s)-digit strings (leading zeros allowed) numerically >= s."""
+    if not s:
+        return Lit(b"")
+    d = s[0]
+    opts: List[_Node] = [Seq([Lit(bytes([d])), _ge_str(s[1:])])]
+    hi = {c for c in _DIGITS if c > d}
+    if hi:
+        opts.append(Seq([Cls(hi), _digits_any(len(s) - 1)]))
+    return Alt(opts)
+
+
+def _le_str(s: bytes, min_first: int = 0x30) -> _Node:
+    """len(s)-digit strings numerically <= s, first digit >= min_first
+    (min_first=ord('1') forbids a leading zero)."""
+    if not s:
+        return Lit(b"")
+    d = s[0]
+    opts: List[_Node] = []
+    if d >= min_first:
+        opts.append(Seq([Lit(bytes([d])), _le_str(s[1:])]))
+    lo = {c for c in _DIGITS if min_first <= c < d}
+    if lo:
+        opts.append(Seq([Cls(lo), _digits_any(len(s) - 1)]))
+    return Alt(opts) if opts else Alt([])
+
+
+def _between_str(lo_s: bytes, hi_s: bytes) -> _Node:
+    """Equal-length digit strings numerically in [lo_s, hi_s]."""
+    if not lo_s:
+        return Lit(b"")
+    l0, h0 = lo_s[0], hi_s[0]
+    if l0 == h0:
+        return Seq([Lit(bytes([l0])), _between_str(lo_s[1:], hi_s[1:])])
+    opts: List[_Node] = [
+        Seq([Lit(bytes([l0])), _ge_str(lo_s[1:])]),
+        Seq([Lit(bytes([h0])), _le_str(hi_s[1:])]),
+    ]
+    mid = {c for c in _DIGITS if l0 < c < h0}
+    if mid:
+        opts.append(Seq([Cls(mid), _digits_any(len(lo_s) - 1)]))
+    return Alt(opts)
+
+
+def _nonneg_range_ir(a: int, b: int) -> _Node:
+    """Decimal integers in [a, b] (0 <= a <= b), no leading zeros."""
+    sa, sb = str(a).encode(), str(b).encode()
+    opts: List[_Node] = []
+    for L in range(len(sa), len(sb) + 1):
+        if L == len(sa) == len(sb):
+            opts.append(_between_str(sa, sb))
+        elif L == len(sa):
+            opts.append(_ge_str(sa))  # [a, 10^L - 1]; sa has no leading zero
+        elif L == len(sb):
+            opts.append(_le_str(sb, min_first=0x31))  # [10^(L-1), b]
+        else:
+            opts.append(Seq([Cls(set(b"123456789")), _digits_any(L - 1)]))
+    return Alt(opts)
+
+
+def _int_range_ir(lo: int, hi: int) -> _Node:
+    """EXACT JSON-integer range [lo, hi]: the DFA admits precisely the
+    integers in range (tight-prefix construction, O(digits^2) IR) — a
+    schema's minimum/maximum are enforced by construction, not by the
+    digit-count approximation."""
+    assert lo <= hi
+    if hi < 0:
+        return Seq([Lit(b"-"), _nonneg_range_ir(-hi, -lo)])
+    opts: List[_Node] = []
+    if lo < 0:
+        opts.append(Seq([Lit(b"-"), _nonneg_range_ir(1, -lo)]))  # JSON has no -0
+        lo = 0
+    opts.append(_nonneg_range_ir(lo, hi))
+    return Alt(opts)
+
+
 def _integer_ir(max_digits: Optional[int] = None, allow_negative: bool = True) -> _Node:
-    """JSON integer; with bounds, the digit COUNT is capped from the schema's
-    minimum/maximum (a digit-count bound, not a full range check — standard
-    guided-decoding practice: 5000 -> up to 4 digits, so 9999 is admitted)."""
+    """JSON integer; with one-sided bounds, the digit COUNT is capped from
+    the schema's minimum/maximum (two-sided bounds take the exact
+    _int_range_ir path instead)."""
     sign = Opt(Lit(b"-")) if allow_negative else Lit(b"")
     if max_digits is None:
         body = Alt([Lit(b"0"), Seq([Cls(set(b"123456789")), Star(Cls(_DIGITS))])])
@@ -159,6 +239,17 @@ def _schema_ir_body(schema: Dict[str, Any], defs: Dict[str, Any], depth: int, ws
         return _json_string_ir(schema.get("minLength", 0), schema.get("maxLength"))
     if t == "integer":
         mn, mx = schema.get("minimum"), schema.get("maximum")
+        # pydantic gt/lt emit exclusive bounds (JSON Schema draft 2020)
+        if schema.get("exclusiveMinimum") is not None:
+            emn = int(schema["exclusiveMinimum"]) + 1
+            mn = emn if mn is None else max(int(mn), emn)
+        if schema.get("exclusiveMaximum") is not None:
+            emx = int(schema["exclusiveMaximum"]) - 1
+            mx = emx if mx is None else min(int(mx), emx)
+        if mn is not None and mx is not None:
+            if int(mn) > int(mx):
+                raise SchemaCompileError(f"empty integer range [{mn}, {mx}]")
+            return _int_range_ir(int(mn), int(mx))
         max_digits = None
         if mx is not None or mn is not None:
             bound = max(abs(int(mx)) if mx is not None else 0,

@@ -121,3 +121,30 @@ def test_mutated_byte_never_reaches_accept_silently(pair):
     mid = len(text) // 2
     corrupted = text[:mid] + "\x01" + text[mid + 1:]
     assert not _accepts(schema, corrupted, ws=False)
+
+
+@settings(max_examples=300, deadline=None, suppress_health_check=[HealthCheck.too_slow])
+@given(st.integers(-10**7, 10**7), st.integers(0, 10**7), st.integers(-10**7 - 1000, 10**7 + 1000))
+def test_integer_range_is_exact(lo, span, probe):
+    """Two-sided integer bounds compile to an EXACT range DFA: str(x) is
+    accepted iff minimum <= x <= maximum (no digit-count slack)."""
+    hi = lo + span
+    sch = {"type": "integer", "minimum": lo, "maximum": hi}
+    for x in {probe, lo, hi, lo - 1, hi + 1}:
+        assert _accepts(sch, str(x), ws=False) == (lo <= x <= hi), (lo, hi, x)
+
+
+def test_integer_range_edges():
+    sch = {"type": "integer", "minimum": 0, "maximum": 120}
+    assert _accepts(sch, "120", ws=False) and _accepts(sch, "0", ws=False)
+    for bad in ("121", "323", "999", "-1", "-0", "05", "00"):
+        assert not _accepts(sch, bad, ws=False), bad
+    # exclusive bounds (pydantic gt/lt)
+    sch = {"type": "integer", "exclusiveMinimum": 0, "exclusiveMaximum": 10}
+    assert _accepts(sch, "1", ws=False) and _accepts(sch, "9", ws=False)
+    assert not _accepts(sch, "0", ws=False) and not _accepts(sch, "10", ws=False)
+    # negative-only range
+    sch = {"type": "integer", "minimum": -250, "maximum": -3}
+    assert _accepts(sch, "-3", ws=False) and _accepts(sch, "-250", ws=False)
+    for bad in ("-2", "-251", "0", "3"):
+        assert not _accepts(sch, bad, ws=False), bad
