@@ -198,10 +198,16 @@ def _integer_ir(max_digits: Optional[int] = None, allow_negative: bool = True) -
     return Seq([sign, body])
 
 
-def _number_ir() -> _Node:
+def _number_ir(max_int_digits: Optional[int] = None, allow_negative: bool = True) -> _Node:
+    """JSON number. With schema bounds the INTEGER PART's digit count is
+    capped (exact float ranges are not DFA-expressible; the cap bounds the
+    magnitude so maximum=100 can't emit 12345.0) and the exponent is
+    dropped (an exponent reintroduces unbounded magnitude)."""
     frac = Seq([Lit(b"."), Cls(_DIGITS), Star(Cls(_DIGITS))])
     exp = Seq([Cls(set(b"eE")), Opt(Cls(set(b"+-"))), Cls(_DIGITS), Star(Cls(_DIGITS))])
-    return Seq([_integer_ir(), Opt(frac), Opt(exp)])
+    if max_int_digits is None:
+        return Seq([_integer_ir(None, allow_negative), Opt(frac), Opt(exp)])
+    return Seq([_integer_ir(max_int_digits, allow_negative), Opt(frac)])
 
 
 class SchemaCompileError(ValueError):
@@ -258,7 +264,18 @@ def _schema_ir_body(schema: Dict[str, Any], defs: Dict[str, Any], depth: int, ws
         allow_neg = mn is None or mn < 0
         return _integer_ir(max_digits, allow_neg)
     if t == "number":
-        return _number_ir()
+        mn, mx = schema.get("minimum"), schema.get("maximum")
+        if schema.get("exclusiveMinimum") is not None:
+            mn = schema["exclusiveMinimum"] if mn is None else max(mn, schema["exclusiveMinimum"])
+        if schema.get("exclusiveMaximum") is not None:
+            mx = schema["exclusiveMaximum"] if mx is None else min(mx, schema["exclusiveMaximum"])
+        max_int_digits = None
+        if mn is not None or mx is not None:
+            bound = max(abs(float(mx)) if mx is not None else 0.0,
+                        abs(float(mn)) if mn is not None else 0.0)
+            max_int_digits = max(1, len(str(int(bound))))
+        allow_neg = mn is None or mn < 0
+        return _number_ir(max_int_digits, allow_neg)
     if t == "boolean":
         return Alt([Lit(b"true"), Lit(b"false")])
     if t == "null":

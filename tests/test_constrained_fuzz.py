@@ -148,3 +148,17 @@ def test_integer_range_edges():
     assert _accepts(sch, "-3", ws=False) and _accepts(sch, "-250", ws=False)
     for bad in ("-2", "-251", "0", "3"):
         assert not _accepts(sch, bad, ws=False), bad
+
+
+def test_number_bounds_cap_magnitude():
+    """number fields with bounds cap the integer-part digits and drop the
+    exponent form (magnitude guard; exact float ranges aren't a DFA)."""
+    sch = {"type": "number", "minimum": 0, "maximum": 100}
+    for ok in ("0", "100", "99.75", "3.14159", "0.001"):
+        assert _accepts(sch, ok, ws=False), ok
+    for bad in ("12345", "1234.5", "-1.0", "1e9", "2E2"):
+        assert not _accepts(sch, bad, ws=False), bad
+    # unbounded numbers keep the full JSON grammar incl. exponent
+    sch = {"type": "number"}
+    for ok in ("-12345.67", "1e9", "6.02E23"):
+        assert _accepts(sch, ok, ws=False), ok
