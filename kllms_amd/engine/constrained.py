@@ -105,6 +105,37 @@ def _json_string_ir(min_len: int = 0, max_len: Optional[int] = None) -> _Node:
     return Seq([Lit(b'"'), _bounded(char, min_len, max_len), Lit(b'"')])
 
 
+def _format_string_ir(fmt: str) -> Optional[_Node]:
+    """Fixed-shape DFAs for the JSON-Schema string formats pydantic emits
+    (uuid.UUID, datetime.date/time/datetime fields). Without these a
+    formatted field samples an arbitrary string and pydantic re-validation
+    rejects essentially every stream; with them the sampled surface parses.
+    Date shapes are calendar-plausible (day 01-31 regardless of month —
+    pydantic still rejects e.g. Feb 31, a rare draw), times are 24h."""
+    H = _HEX
+    hexs = lambda n: [Cls(H) for _ in range(n)]
+    d = lambda: Cls(_DIGITS)
+    if fmt == "uuid":
+        return Seq(hexs(8) + [Lit(b"-")] + hexs(4) + [Lit(b"-")] + hexs(4)
+                   + [Lit(b"-")] + hexs(4) + [Lit(b"-")] + hexs(12))
+    month = Alt([Seq([Lit(b"0"), Cls(set(b"123456789"))]), Seq([Lit(b"1"), Cls(set(b"012"))])])
+    day = Alt([Seq([Lit(b"0"), Cls(set(b"123456789"))]),
+               Seq([Cls(set(b"12")), d()]), Seq([Lit(b"3"), Cls(set(b"01"))])])
+    date = Seq([d(), d(), d(), d(), Lit(b"-"), month, Lit(b"-"), day])
+    hh = Alt([Seq([Cls(set(b"01")), d()]), Seq([Lit(b"2"), Cls(set(b"0123"))])])
+    m60 = Seq([Cls(set(b"012345")), d()])
+    time = Seq([hh, Lit(b":"), m60, Lit(b":"), m60,
+                Opt(Seq([Lit(b"."), Cls(_DIGITS), Star(Cls(_DIGITS))]))])
+    if fmt == "date":
+        return date
+    if fmt == "time":
+        return time
+    if fmt == "date-time":
+        tz = Alt([Lit(b"Z"), Seq([Cls(set(b"+-")), hh, Lit(b":"), m60])])
+        return Seq([date, Lit(b"T"), time, Opt(tz)])
+    return None
+
+
 def _digits_any(n: int) -> _Node:
     return Seq([Cls(_DIGITS) for _ in range(n)]) if n else Lit(b"")
 
@@ -242,6 +273,11 @@ def _schema_ir_body(schema: Dict[str, Any], defs: Dict[str, Any], depth: int, ws
     if isinstance(t, list):
         return Alt([_schema_ir_body({**schema, "type": ti}, defs, depth + 1, ws) for ti in t])
     if t == "string":
+        if schema.get("format"):
+            body = _format_string_ir(schema["format"])
+            if body is not None:
+                return Seq([Lit(b'"'), body, Lit(b'"')])
+            # unknown formats fall through to the plain string grammar
         return _json_string_ir(schema.get("minLength", 0), schema.get("maxLength"))
     if t == "integer":
         mn, mx = schema.get("minimum"), schema.get("maximum")

@@ -162,3 +162,56 @@ def test_number_bounds_cap_magnitude():
     sch = {"type": "number"}
     for ok in ("-12345.67", "1e9", "6.02E23"):
         assert _accepts(sch, ok, ws=False), ok
+
+
+def test_format_strings_accept_valid_reject_garbage():
+    """uuid/date/time/date-time formats compile to fixed-shape DFAs so
+    pydantic UUID/date/datetime fields validate from the sampled surface."""
+    cases = {
+        "uuid": (["01234567-89ab-CDEF-0123-456789abcdef"],
+                 ["0123456789ab-CDEF-0123-456789abcdef", "zz234567-89ab-cdef-0123-456789abcdef",
+                  "01234567-89ab-cdef-0123-456789abcde"]),
+        "date": (["2024-02-29", "0001-01-01", "9999-12-31"],
+                 ["2024-13-01", "2024-00-10", "2024-01-32", "24-01-01", "2024/01/01"]),
+        "time": (["00:00:00", "23:59:59.123456"],
+                 ["24:00:00", "12:60:00", "12:00:61", "1:00:00"]),
+        "date-time": (["2024-06-15T08:30:00Z", "2024-06-15T08:30:00.5+05:30",
+                       "2024-06-15T23:59:59"],
+                      ["2024-06-15 08:30:00", "2024-06-15T25:00:00Z"]),
+    }
+    for fmt, (goods, bads) in cases.items():
+        sch = {"type": "string", "format": fmt}
+        for g in goods:
+            assert _accepts(sch, json.dumps(g), ws=False), (fmt, g)
+        for b in bads:
+            assert not _accepts(sch, json.dumps(b), ws=False), (fmt, b)
+    # unknown format falls back to the plain string grammar
+    sch = {"type": "string", "format": "hostname"}
+    assert _accepts(sch, json.dumps("anything at all"), ws=False)
+
+
+def test_pydantic_uuid_datetime_fields_parse_end_to_end():
+    import datetime
+    import uuid as _uuid
+
+    from pydantic import BaseModel
+
+    from kllms_amd import KLLMs
+
+    class Event(BaseModel):
+        id: _uuid.UUID
+        day: datetime.date
+        ts: datetime.datetime
+
+    c = KLLMs(model="tiny-llama", device="cpu", max_kv_blocks=512,
+              use_hip_graphs=False, seed=0)
+    r = c.chat.completions.parse(
+        messages=[{"role": "user", "content": "emit an event"}],
+        model="tiny-llama", response_format=Event, n=6,
+        max_tokens=128, temperature=1.0, seed=4)
+    finished = [ch for ch in r.choices[1:] if ch.finish_reason == "stop"]
+    assert finished, "no stream finished within budget"
+    parsed = [ch for ch in finished if ch.message.parsed is not None]
+    # calendar-impossible draws (Feb 31) are the only allowed misses
+    assert len(parsed) >= max(1, len(finished) - 1), \
+        [ch.message.content for ch in finished]
