@@ -336,6 +336,18 @@ def _schema_ir_body(schema: Dict[str, Any], defs: Dict[str, Any], depth: int, ws
     if t == "object" or "properties" in schema:
         props = schema.get("properties", {})
         if not props:
+            ap = schema.get("additionalProperties")
+            if isinstance(ap, dict):
+                # pydantic Dict[str, T]: free-form keys, TYPED values — the
+                # value subschema is enforced (an untyped fallback here let
+                # Dict[str, int] sample string values that pydantic rejects)
+                val = _lead_ws(schema_to_ir(ap, defs, depth + 1, ws), ws)
+                key = _lead_ws(_json_string_ir(), ws)
+                colon = _lead_ws(Lit(b":"), ws)
+                member = Seq([key, colon, val])
+                return Seq([Lit(b"{"),
+                            Opt(Seq([member, Star(Seq([_lead_ws(Lit(b","), ws), member]))])),
+                            _lead_ws(Lit(b"}"), ws)])
             return _any_object_ir(defs, depth + 1, ws)
         parts: List[_Node] = [Lit(b"{")]
         required = set(schema.get("required", list(props.keys())))

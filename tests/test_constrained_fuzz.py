@@ -215,3 +215,34 @@ def test_pydantic_uuid_datetime_fields_parse_end_to_end():
     # calendar-impossible draws (Feb 31) are the only allowed misses
     assert len(parsed) >= max(1, len(finished) - 1), \
         [ch.message.content for ch in finished]
+
+
+def test_additional_properties_value_type_enforced():
+    """pydantic Dict[str, T]: free keys but TYPED values by construction."""
+    sch = {"type": "object", "additionalProperties": {"type": "integer",
+                                                      "minimum": 0, "maximum": 99}}
+    for ok in ('{}', '{"a":1}', '{"a":1,"bc":42}'):
+        assert _accepts(sch, ok, ws=False), ok
+    for bad in ('{"a":"x"}', '{"a":100}', '{"a":[1]}', '{"a":1,}'):
+        assert not _accepts(sch, bad, ws=False), bad
+
+    # end to end: Dict[str, int] parses from the sampled surface
+    from typing import Dict as _Dict
+
+    from pydantic import BaseModel
+
+    from kllms_amd import KLLMs
+
+    class Scores(BaseModel):
+        scores: _Dict[str, int]
+
+    c = KLLMs(model="tiny-llama", device="cpu", max_kv_blocks=512,
+              use_hip_graphs=False, seed=0)
+    r = c.chat.completions.parse(
+        messages=[{"role": "user", "content": "score things"}],
+        model="tiny-llama", response_format=Scores, n=6,
+        max_tokens=64, temperature=1.0, seed=9)
+    finished = [ch for ch in r.choices[1:] if ch.finish_reason == "stop"]
+    assert finished
+    assert all(ch.message.parsed is not None for ch in finished), \
+        [ch.message.content for ch in finished]
