@@ -317,6 +317,29 @@ def _schema_ir_body(schema: Dict[str, Any], defs: Dict[str, Any], depth: int, ws
     if t == "null":
         return Lit(b"null")
     if t == "array":
+        pref = schema.get("prefixItems")
+        if pref and schema.get("minItems", 0) >= len(pref):
+            # pydantic Tuple[...]: POSITIONAL subschemas enforced (the
+            # any-value fallback admitted ["x", 3] for Tuple[int, str]);
+            # items beyond the prefix follow the `items` schema
+            parts: List[_Node] = [Lit(b"[")]
+            for i, sub in enumerate(pref):
+                if i:
+                    parts.append(_lead_ws(Lit(b","), ws))
+                parts.append(_lead_ws(schema_to_ir(sub, defs, depth + 1, ws), ws))
+            n = len(pref)
+            max_items = schema.get("maxItems")
+            tail_sch = schema.get("items", {})
+            if tail_sch is not False and (max_items is None or max_items > n):
+                tail = lambda: Seq([
+                    _lead_ws(Lit(b","), ws),
+                    _lead_ws(schema_to_ir(tail_sch, defs, depth + 1, ws) if tail_sch
+                             else _any_value_ir(defs, depth + 1, ws=ws), ws),
+                ])
+                parts.append(_bounded(tail, max(0, schema.get("minItems", n) - n),
+                                      None if max_items is None else max_items - n))
+            parts.append(_lead_ws(Lit(b"]"), ws))
+            return Seq(parts)
         item = schema.get("items", {})
 
         def item_ir() -> _Node:

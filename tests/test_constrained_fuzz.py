@@ -246,3 +246,25 @@ def test_additional_properties_value_type_enforced():
     assert finished
     assert all(ch.message.parsed is not None for ch in finished), \
         [ch.message.content for ch in finished]
+
+
+def test_prefix_items_positional_types_enforced():
+    """pydantic Tuple[int, str]: each position's subschema is enforced."""
+    from typing import Tuple as _Tuple
+
+    from pydantic import BaseModel
+
+    class M(BaseModel):
+        pair: _Tuple[int, str]
+
+    sch = M.model_json_schema()
+    assert _accepts(sch, '{"pair":[3,"x"]}', ws=False)
+    for bad in ('{"pair":["x",3]}', '{"pair":[3,"x",1]}', '{"pair":[3]}', '{"pair":[]}'):
+        assert not _accepts(sch, bad, ws=False), bad
+    # variadic tail: Tuple[int, ...]-style (prefix + typed items)
+    sch2 = {"type": "array", "minItems": 1, "maxItems": 3,
+            "prefixItems": [{"type": "string"}], "items": {"type": "integer"}}
+    for ok in ('["a"]', '["a",1]', '["a",1,2]'):
+        assert _accepts(sch2, ok, ws=False), ok
+    for bad in ('["a",1,2,3]', '["a","b"]', '[1]'):
+        assert not _accepts(sch2, bad, ws=False), bad
