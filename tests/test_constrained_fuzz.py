@@ -268,3 +268,62 @@ def test_prefix_items_positional_types_enforced():
         assert _accepts(sch2, ok, ws=False), ok
     for bad in ('["a",1,2,3]', '["a","b"]', '[1]'):
         assert not _accepts(sch2, bad, ws=False), bad
+
+
+def test_kitchen_sink_model_streams_all_validate():
+    """Every field type the constrained compiler supports, in ONE model:
+    all stop-finished streams must pydantic-validate (calendar-impossible
+    dates are the single tolerated miss)."""
+    import datetime
+    import enum
+    import uuid as _uuid
+    from typing import Dict as _Dict
+    from typing import List as _List
+    from typing import Literal as _Literal
+    from typing import Optional as _Optional
+    from typing import Tuple as _Tuple
+
+    from pydantic import BaseModel, Field
+
+    from kllms_amd import KLLMs
+
+    class Color(str, enum.Enum):
+        red = "red"
+        blue = "blue"
+
+    class Inner(BaseModel):
+        name: str = Field(max_length=6)
+        score: int = Field(ge=-5, le=120)
+
+    class Sink(BaseModel):
+        id: _uuid.UUID
+        day: datetime.date
+        kind: _Literal["a", "b"]
+        color: Color
+        ratio: float = Field(ge=0, le=10)
+        flag: bool
+        note: _Optional[str] = Field(default=None, max_length=5)
+        pair: _Tuple[int, str]
+        items: _List[Inner] = Field(max_length=2)
+        counts: _Dict[str, int]
+
+    c = KLLMs(model="tiny-llama", device="cpu", max_kv_blocks=1024,
+              use_hip_graphs=False, seed=0, max_seq_len=512)
+    r = c.chat.completions.parse(
+        messages=[{"role": "user", "content": "emit"}],
+        model="tiny-llama", response_format=Sink, n=8,
+        max_tokens=400, temperature=1.0, seed=13)
+    finished = [ch for ch in r.choices[1:] if ch.finish_reason == "stop"]
+    assert finished, "no stream finished"
+    misses = [ch.message.content for ch in finished if ch.message.parsed is None]
+    # tolerated: impossible calendar dates (e.g. Feb 31) and floats beyond
+    # le=10 (digit-cap only); everything else must validate
+    for m in misses:
+        import json as _json
+        doc = _json.loads(m)
+        day_ok = True
+        try:
+            datetime.date.fromisoformat(doc["day"])
+        except ValueError:
+            day_ok = False
+        assert (not day_ok) or not (0 <= doc["ratio"] <= 10), m
