@@ -105,6 +105,178 @@ def _json_string_ir(min_len: int = 0, max_len: Optional[int] = None) -> _Node:
     return Seq([Lit(b'"'), _bounded(char, min_len, max_len), Lit(b'"')])
 
 
+_RE_CLASSES = {
+    ord("d"): set(b"0123456789"),
+    ord("w"): set(b"abcdefghijklmnopqrstuvwxyzABCDEFGHIJKLMNOPQRSTUVWXYZ0123456789_"),
+    ord("s"): set(b" \t"),  # JSON strings can't hold raw \n\r; escaped ws only
+}
+# the printable-ASCII universe of the constrained string alphabet (no '"'
+# or '\\' — those need JSON escapes, which patterns rarely intend)
+_RE_ANY = {c for c in range(0x20, 0x7F)} - {0x22, 0x5C}
+
+
+class _PatternUnsupported(Exception):
+    pass
+
+
+def _compile_pattern_ir(pat: str) -> _Node:
+    """A safe regex SUBSET -> IR, for JSON-Schema `pattern` (pydantic
+    Field(pattern=...)): literals, escapes (\\d \\w \\s \\. etc.),
+    [...] classes with ranges and negation, groups, alternation `|`,
+    quantifiers `* + ? {m} {m,} {m,n}`, and `^`/`$` anchors (stripped —
+    compilation is full-match, which is strictly stronger than the spec's
+    re.search, so every emitted string still validates). Unsupported
+    constructs (backrefs, lookaround, non-ASCII, nested quantifier edge
+    cases) raise _PatternUnsupported and the caller falls back to the
+    plain string grammar."""
+    b = pat.encode("ascii", errors="strict").lstrip(b"^")
+    if b.endswith(b"$") and not b.endswith(b"\\$"):
+        b = b[:-1]
+    pos = 0
+
+    def peek():
+        return b[pos] if pos < len(b) else None
+
+    def parse_class() -> set:
+        nonlocal pos
+        assert b[pos] == 0x5B  # [
+        pos += 1
+        neg = peek() == 0x5E  # ^
+        if neg:
+            pos += 1
+        chars: set = set()
+        first = True
+        while True:
+            c = peek()
+            if c is None:
+                raise _PatternUnsupported("unterminated class")
+            if c == 0x5D and not first:  # ]
+                pos += 1
+                break
+            first = False
+            if c == 0x5C:  # escape
+                pos += 1
+                e = peek()
+                if e is None:
+                    raise _PatternUnsupported("trailing backslash")
+                pos += 1
+                low = e | 0x20
+                if low in _RE_CLASSES:
+                    cls = _RE_CLASSES[low]
+                    chars |= (_RE_ANY - cls) if (e < 0x61) else cls
+                    continue
+                chars.add(e)
+                continue
+            pos += 1
+            if peek() == 0x2D and pos + 1 < len(b) and b[pos + 1] != 0x5D:  # range
+                pos += 1
+                hi = b[pos]
+                pos += 1
+                chars |= set(range(c, hi + 1))
+            else:
+                chars.add(c)
+        out = (_RE_ANY - chars) if neg else (chars & _RE_ANY)
+        if not out:
+            raise _PatternUnsupported("empty class")
+        return out
+
+    def parse_atom() -> _Node:
+        nonlocal pos
+        c = peek()
+        if c is None:
+            raise _PatternUnsupported("dangling quantifier")
+        if c == 0x28:  # (
+            pos += 1
+            if peek() == 0x3F:  # (?: or lookaround
+                if pos + 1 < len(b) and b[pos + 1] == 0x3A:
+                    pos += 2
+                else:
+                    raise _PatternUnsupported("lookaround/named group")
+            node = parse_alt()
+            if peek() != 0x29:
+                raise _PatternUnsupported("unbalanced group")
+            pos += 1
+            return node
+        if c == 0x5B:  # [
+            return Cls(parse_class())
+        if c == 0x2E:  # .
+            pos += 1
+            return Cls(set(_RE_ANY))
+        if c == 0x5C:  # escape
+            pos += 1
+            e = peek()
+            if e is None:
+                raise _PatternUnsupported("trailing backslash")
+            pos += 1
+            low = e | 0x20
+            if low in _RE_CLASSES:
+                cls = _RE_CLASSES[low]
+                return Cls(set(cls) if e >= 0x61 else set(_RE_ANY - cls))
+            if e in (0x62, 0x42) or 0x30 <= e <= 0x39:  # \b, backrefs
+                raise _PatternUnsupported("anchor/backref escape")
+            return Lit(bytes([e]))
+        if c in (0x2A, 0x2B, 0x3F, 0x7B, 0x29, 0x7C):  # * + ? { ) |
+            raise _PatternUnsupported("misplaced metachar")
+        if c in (0x22, 0x5C) or not 0x20 <= c < 0x7F:
+            raise _PatternUnsupported("char needs JSON escaping")
+        pos += 1
+        return Lit(bytes([c]))
+
+    def parse_quant(node: _Node) -> _Node:
+        nonlocal pos
+        c = peek()
+        if c == 0x2A:  # *
+            pos += 1
+            return Star(node)
+        if c == 0x2B:  # +
+            pos += 1
+            return Seq([node, Star(node)])
+        if c == 0x3F:  # ?
+            pos += 1
+            return Opt(node)
+        if c == 0x7B:  # {m,n}
+            end = b.find(b"}", pos)
+            if end < 0:
+                raise _PatternUnsupported("unterminated {}")
+            body = b[pos + 1:end].decode()
+            pos = end + 1
+            try:
+                if "," in body:
+                    lo_s, hi_s = body.split(",", 1)
+                    m = int(lo_s)
+                    n = int(hi_s) if hi_s else None
+                else:
+                    m = n = int(body)
+            except ValueError:
+                raise _PatternUnsupported("bad {} bound")
+            if m > 64 or (n is not None and n > 64):
+                raise _PatternUnsupported("repetition too large")
+            head = [node] * m
+            if n is None:
+                return Seq(head + [Star(node)])
+            return Seq(head + [Opt(node) for _ in range(n - m)])
+        return node
+
+    def parse_seq() -> _Node:
+        parts: List[_Node] = []
+        while peek() is not None and peek() not in (0x7C, 0x29):
+            parts.append(parse_quant(parse_atom()))
+        return Seq(parts) if parts else Lit(b"")
+
+    def parse_alt() -> _Node:
+        nonlocal pos
+        opts = [parse_seq()]
+        while peek() == 0x7C:
+            pos += 1
+            opts.append(parse_seq())
+        return opts[0] if len(opts) == 1 else Alt(opts)
+
+    node = parse_alt()
+    if pos != len(b):
+        raise _PatternUnsupported("trailing garbage")
+    return node
+
+
 def _format_string_ir(fmt: str) -> Optional[_Node]:
     """Fixed-shape DFAs for the JSON-Schema string formats pydantic emits
     (uuid.UUID, datetime.date/time/datetime fields). Without these a
@@ -278,6 +450,12 @@ def _schema_ir_body(schema: Dict[str, Any], defs: Dict[str, Any], depth: int, ws
             if body is not None:
                 return Seq([Lit(b'"'), body, Lit(b'"')])
             # unknown formats fall through to the plain string grammar
+        if schema.get("pattern"):
+            try:
+                body = _compile_pattern_ir(schema["pattern"])
+                return Seq([Lit(b'"'), body, Lit(b'"')])
+            except (_PatternUnsupported, UnicodeEncodeError):
+                pass  # unsupported construct: plain string grammar
         return _json_string_ir(schema.get("minLength", 0), schema.get("maxLength"))
     if t == "integer":
         mn, mx = schema.get("minimum"), schema.get("maximum")

@@ -327,3 +327,56 @@ def test_kitchen_sink_model_streams_all_validate():
         except ValueError:
             day_ok = False
         assert (not day_ok) or not (0 <= doc["ratio"] <= 10), m
+
+
+def test_pattern_subset_matches_python_re():
+    """JSON-Schema `pattern` (pydantic Field(pattern=...)): the compiled
+    DFA agrees with python re.fullmatch on the supported subset; anything
+    unsupported falls back to the plain string grammar (never over-
+    constrains into an unsatisfiable schema)."""
+    import random
+    import re
+    import string as strmod
+
+    cases = [
+        (r"^[A-Z]{2}-\d{4}$", ["AB-1234"], ["ab-1234", "AB-123", "AB_1234"]),
+        (r"^\d+(\.\d+)?$", ["1", "3.14"], ["", ".5", "1."]),
+        (r"cat|dog|bird", ["cat", "dog"], ["cats", "catdog"]),
+        (r"a{2,4}b?", ["aa", "aaaab"], ["a", "aaaaa", "ab"]),
+        (r"[^0-9]+", ["abc"], ["a1", ""]),
+        (r"(?:ab)+c", ["abc", "ababc"], ["ac", "abab"]),
+    ]
+    rng = random.Random(0)
+    alpha = strmod.ascii_letters + strmod.digits + "-_. "
+    for pat, goods, bads in cases:
+        sch = {"type": "string", "pattern": pat}
+        for g in goods:
+            assert _accepts(sch, f'"{g}"', ws=False), (pat, g)
+        for b in bads:
+            assert not _accepts(sch, f'"{b}"', ws=False), (pat, b)
+        rx = re.compile(pat.lstrip("^").rstrip("$"))
+        for _ in range(200):
+            s = "".join(rng.choice(alpha) for _ in range(rng.randint(0, 8)))
+            assert _accepts(sch, f'"{s}"', ws=False) == (rx.fullmatch(s) is not None), (pat, s)
+    for pat in (r"(?=x)a", r"\bword\b", r"(a)\1", "héllo"):
+        assert _accepts({"type": "string", "pattern": pat}, '"anything"', ws=False)
+
+
+def test_pattern_field_end_to_end():
+    from pydantic import BaseModel, Field
+
+    from kllms_amd import KLLMs
+
+    class Order(BaseModel):
+        code: str = Field(pattern=r"^[A-Z]{2}-\d{4}$")
+
+    c = KLLMs(model="tiny-llama", device="cpu", max_kv_blocks=512,
+              use_hip_graphs=False, seed=0)
+    r = c.chat.completions.parse(
+        messages=[{"role": "user", "content": "make an order code"}],
+        model="tiny-llama", response_format=Order, n=6,
+        max_tokens=48, temperature=1.0, seed=21)
+    finished = [ch for ch in r.choices[1:] if ch.finish_reason == "stop"]
+    assert finished
+    assert all(ch.message.parsed is not None for ch in finished), \
+        [ch.message.content for ch in finished]
