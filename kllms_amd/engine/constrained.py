@@ -542,13 +542,20 @@ def _schema_ir_body(schema: Dict[str, Any], defs: Dict[str, Any], depth: int, ws
                 # pydantic Dict[str, T]: free-form keys, TYPED values — the
                 # value subschema is enforced (an untyped fallback here let
                 # Dict[str, int] sample string values that pydantic rejects)
-                val = _lead_ws(schema_to_ir(ap, defs, depth + 1, ws), ws)
-                key = _lead_ws(_json_string_ir(), ws)
-                colon = _lead_ws(Lit(b":"), ws)
-                member = Seq([key, colon, val])
-                return Seq([Lit(b"{"),
-                            Opt(Seq([member, Star(Seq([_lead_ws(Lit(b","), ws), member]))])),
-                            _lead_ws(Lit(b"}"), ws)])
+                def member() -> _Node:
+                    return Seq([_lead_ws(_json_string_ir(), ws), _lead_ws(Lit(b":"), ws),
+                                _lead_ws(schema_to_ir(ap, defs, depth + 1, ws), ws)])
+                min_p = schema.get("minProperties", 0)
+                max_p = schema.get("maxProperties")
+                more = lambda: Seq([_lead_ws(Lit(b","), ws), member()])
+                if min_p == 0:
+                    rest = _bounded(more, 0, None if max_p is None else max(0, max_p - 1))
+                    body: _Node = Opt(Seq([member(), rest]))
+                else:
+                    head = [member()] + [more() for _ in range(min_p - 1)]
+                    rest = _bounded(more, 0, None if max_p is None else max(0, max_p - min_p))
+                    body = Seq(head + [rest])
+                return Seq([Lit(b"{"), body, _lead_ws(Lit(b"}"), ws)])
             return _any_object_ir(defs, depth + 1, ws)
         parts: List[_Node] = [Lit(b"{")]
         required = set(schema.get("required", list(props.keys())))

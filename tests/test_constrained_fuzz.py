@@ -380,3 +380,12 @@ def test_pattern_field_end_to_end():
     assert finished
     assert all(ch.message.parsed is not None for ch in finished), \
         [ch.message.content for ch in finished]
+
+
+def test_dict_min_max_properties():
+    sch = {"type": "object", "additionalProperties": {"type": "integer"},
+           "minProperties": 1, "maxProperties": 2}
+    for ok in ('{"a":1}', '{"a":1,"b":2}'):
+        assert _accepts(sch, ok, ws=False), ok
+    for bad in ('{}', '{"a":1,"b":2,"c":3}'):
+        assert not _accepts(sch, bad, ws=False), bad
