@@ -37,6 +37,17 @@ from .config import EngineConfig
 from .sampling import SamplingParams
 
 
+def _parse_logit_bias(lb):
+    """OpenAI logit_bias {token_id: bias}; malformed keys/values raise a
+    clean ValueError (a 400), not a TypeError."""
+    if not lb:
+        return None
+    try:
+        return {int(k): float(v) for k, v in lb.items()}
+    except (TypeError, ValueError, AttributeError):
+        raise ValueError("logit_bias must map token ids to numeric biases")
+
+
 class ContextLengthExceededError(ValueError):
     """Prompt does not fit in the model's context window (OpenAI-compatible
     behavior: the API raises context_length_exceeded rather than silently
@@ -94,22 +105,68 @@ class LocalEngineClient:
         """OpenAI-style 400s for out-of-range sampling params (the reference
         relies on the remote API to raise these; locally we mirror the
         documented ranges so a switched-over client sees the same errors)."""
-        t = call_params.get("temperature")
-        if t is not None and not 0.0 <= float(t) <= 2.0:
+        def num(key):
+            v = call_params.get(key)
+            if v is None:
+                return None
+            try:
+                return float(v) if not isinstance(v, bool) else 1.0 * v
+            except (TypeError, ValueError):
+                raise ValueError(f"{key} must be a number, got {type(v).__name__}")
+
+        t = num("temperature")
+        if t is not None and not 0.0 <= t <= 2.0:
             raise ValueError(f"temperature must be between 0 and 2, got {t}")
-        p = call_params.get("top_p")
-        if p is not None and not 0.0 <= float(p) <= 1.0:
+        p = num("top_p")
+        if p is not None and not 0.0 <= p <= 1.0:
             raise ValueError(f"top_p must be between 0 and 1, got {p}")
-        n = call_params.get("n")
-        if n is not None and not 1 <= int(n) <= 128:
+        n = num("n")
+        if n is not None and not 1 <= n <= 128:
             raise ValueError(f"n must be between 1 and 128, got {n}")
-        mt = call_params.get("max_tokens")
-        if mt is not None and int(mt) < 1:
+        mt = num("max_tokens")
+        if mt is not None and mt < 1:
             raise ValueError(f"max_tokens must be at least 1, got {mt}")
         for key in ("frequency_penalty", "presence_penalty"):
-            v = call_params.get(key)
-            if v is not None and not -2.0 <= float(v) <= 2.0:
+            v = num(key)
+            if v is not None and not -2.0 <= v <= 2.0:
                 raise ValueError(f"{key} must be between -2 and 2, got {v}")
+        for key in ("seed", "top_logprobs"):
+            v = call_params.get(key)
+            if v is not None and not isinstance(v, int):
+                raise ValueError(f"{key} must be an integer, got {type(v).__name__}")
+        lb = call_params.get("logit_bias")
+        if lb is not None and not isinstance(lb, dict):
+            raise ValueError(f"logit_bias must be an object, got {type(lb).__name__}")
+        msgs = call_params.get("messages")
+        if msgs is not None:
+            for m in msgs:
+                if not isinstance(m, dict) or not isinstance(m.get("role", ""), str):
+                    raise ValueError("each message must be an object with a string 'role'")
+                c = m.get("content")
+                if c is None or isinstance(c, str):
+                    continue
+                if isinstance(c, list):
+                    # OpenAI content-part arrays: text parts are supported
+                    # (flattened before templating); image/audio parts are not
+                    for part in c:
+                        if not isinstance(part, dict) or part.get("type") != "text" \
+                                or not isinstance(part.get("text"), str):
+                            raise ValueError(
+                                "only text content parts are supported by the local engine")
+                    continue
+                raise ValueError("message 'content' must be a string, part list or null")
+
+    @staticmethod
+    def _flatten_messages(messages: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
+        """Flatten OpenAI content-part arrays to plain strings for the chat
+        template (validation guarantees text-only parts)."""
+        out = []
+        for m in messages:
+            c = m.get("content")
+            if isinstance(c, list):
+                m = {**m, "content": "".join(p["text"] for p in c)}
+            out.append(m)
+        return out
 
     def _generate(self, call_params: Dict[str, Any], constrained: bool, scheduled: bool = False) -> tuple:
         from .engine import GenRequest
@@ -132,8 +189,7 @@ class LocalEngineClient:
             presence_penalty=call_params.get("presence_penalty", 0.0),
             logprobs=bool(call_params.get("logprobs", False)),
             top_logprobs=int(call_params.get("top_logprobs") or 0),
-            logit_bias={int(k): float(v) for k, v in call_params["logit_bias"].items()}
-            if call_params.get("logit_bias") else None,
+            logit_bias=_parse_logit_bias(call_params.get("logit_bias")),
         )
         if sampling.top_logprobs and not sampling.logprobs:
             raise ValueError("top_logprobs requires logprobs=True")
@@ -149,7 +205,7 @@ class LocalEngineClient:
             constraint, tool_name = self._build_tool_constraint(call_params)
 
         eng = self.engine
-        prompt = eng.tokenizer.apply_chat_template(messages)
+        prompt = eng.tokenizer.apply_chat_template(self._flatten_messages(messages))
         prompt_ids = eng.tokenizer.encode(prompt)
         self._fit_context(prompt_ids, sampling)
 
@@ -297,14 +353,13 @@ class LocalEngineClient:
                 presence_penalty=call_params.get("presence_penalty", 0.0),
                 logprobs=bool(call_params.get("logprobs", False)),
                 top_logprobs=int(call_params.get("top_logprobs") or 0),
-                logit_bias={int(k): float(v) for k, v in call_params["logit_bias"].items()}
-                if call_params.get("logit_bias") else None,
+                logit_bias=_parse_logit_bias(call_params.get("logit_bias")),
             )
             constraint = None
             rf = call_params.get("response_format")
             if rf is not None:
                 constraint = self._build_constraint(rf, constrained=False)
-            prompt = eng.tokenizer.apply_chat_template(messages)
+            prompt = eng.tokenizer.apply_chat_template(self._flatten_messages(messages))
             prompt_ids = eng.tokenizer.encode(prompt)
             self._fit_context(prompt_ids, sampling)
             reqs.append(GenRequest(prompt_ids=prompt_ids, n=n, sampling=sampling, constraint=constraint))

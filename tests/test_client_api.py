@@ -482,3 +482,25 @@ def test_key_aligner_through_client():
         model="tiny-llama", n=3, temperature=0.9, max_tokens=10, seed=5)
     assert len(out.choices) == 4 and out.choices[0].index == 0
     assert out.likelihoods is not None
+
+
+def test_content_part_arrays_supported():
+    """OpenAI content-part arrays (text parts) are flattened for the local
+    engine; non-text parts get a clean error."""
+    from kllms_amd import KLLMs
+
+    c = KLLMs(model="tiny-llama", device="cpu", max_kv_blocks=256,
+              use_hip_graphs=False, seed=0)
+    parts = [{"type": "text", "text": "Hello "}, {"type": "text", "text": "world"}]
+    a = c.chat.completions.create(messages=[{"role": "user", "content": parts}],
+                                  model="tiny-llama", max_tokens=4, temperature=0.0)
+    b = c.chat.completions.create(messages=[{"role": "user", "content": "Hello world"}],
+                                  model="tiny-llama", max_tokens=4, temperature=0.0)
+    assert a.choices[0].message.content == b.choices[0].message.content
+
+    import pytest as _pytest
+    with _pytest.raises(ValueError, match="text content parts"):
+        c.chat.completions.create(
+            messages=[{"role": "user", "content": [{"type": "image_url",
+                                                    "image_url": {"url": "x"}}]}],
+            model="tiny-llama", max_tokens=4)

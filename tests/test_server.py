@@ -137,3 +137,26 @@ class TestConcurrency:
         again = one(3)
         assert [c["message"]["content"] for c in again["choices"]] == \
                [c["message"]["content"] for c in results[3]["choices"]]
+
+
+class TestFuzz:
+    def test_random_bodies_never_500(self, http):
+        """Adversarial/garbage request bodies: always a clean 4xx/200,
+        never an unhandled 500."""
+        import random
+
+        rng = random.Random(0)
+        vals = [None, True, 0, -1, 1.5, "x", [], {}, {"role": "user"},
+                [{"role": "user", "content": "hi"}], "유니코드", {"a": [1, {"b": None}]},
+                float("1e308"), [[]], {"content": "y"}]
+        keys = ["model", "messages", "n", "temperature", "top_p", "max_tokens",
+                "stop", "seed", "logprobs", "top_logprobs", "logit_bias",
+                "response_format", "tools", "tool_choice", "stream", "bogus"]
+        for trial in range(60):
+            body = {rng.choice(keys): rng.choice(vals)
+                    for _ in range(rng.randint(0, 4))}
+            if rng.random() < 0.5:
+                body["messages"] = [{"role": "user", "content": "hi"}]
+                body.setdefault("max_tokens", 4)
+            r = http.post("/v1/chat/completions", json=body)
+            assert r.status_code < 500, (trial, body, r.text[:300])
