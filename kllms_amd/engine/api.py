@@ -209,7 +209,10 @@ class LocalEngineClient:
         prompt_ids = eng.tokenizer.encode(prompt)
         self._fit_context(prompt_ids, sampling)
 
-        req = GenRequest(prompt_ids=prompt_ids, n=n, sampling=sampling, constraint=constraint)
+        tmo = call_params.get("timeout")
+        deadline = (time.monotonic() + float(tmo)) if tmo else None
+        req = GenRequest(prompt_ids=prompt_ids, n=n, sampling=sampling, constraint=constraint,
+                         deadline=deadline)
         if scheduled:
             out = self.scheduler.submit(req).result()
         else:

@@ -33,6 +33,11 @@ class GenRequest:
     n: int = 1
     sampling: SamplingParams = field(default_factory=SamplingParams)
     constraint: Optional[Any] = None  # kllms_amd.engine.constrained.JsonSchemaConstraint
+    # absolute time.monotonic() cutoff (client `timeout`): streams past it
+    # finish with reason "length" at the next step boundary. TP>1 ignores
+    # deadlines — a wall-clock stop is rank-divergent and would break the
+    # followers' lockstep replay (serve.py)
+    deadline: Optional[float] = None
 
 
 @dataclass
@@ -67,6 +72,7 @@ class _Stream:
         self.sampling = sampling
         self.seed = seed
         self.out = StreamOutput(stream_idx=stream_idx)
+        self.deadline: Optional[float] = None
         self.done = False
         self.step = 0
         self.constraint = constraint
@@ -510,6 +516,7 @@ class LLMEngine:
                 cstate = req.constraint.init_state() if req.constraint is not None else None
                 st = _Stream(ri, si, seq, req.sampling, seed=base_seed + 7919 * si,
                              constraint=req.constraint, constraint_state=cstate)
+                st.deadline = req.deadline if self.ctx.world_size == 1 else None
                 streams.append(st)
                 new_streams.append(st)
             self.kv.free_sequence(parent_seqs[ri])  # streams hold their own refs
@@ -748,6 +755,10 @@ class LLMEngine:
             s.done = True
             return
         if len(s.out.token_ids) >= max_new or s.seq.num_tokens >= self.config.max_seq_len - 1:
+            s.out.finish_reason = "length"
+            s.done = True
+            return
+        if s.deadline is not None and time.monotonic() >= s.deadline:
             s.out.finish_reason = "length"
             s.done = True
             return

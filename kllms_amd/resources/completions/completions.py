@@ -67,6 +67,14 @@ def _build_call_params(
     return call_params
 
 
+def _apply_wrapper_timeout(call_params: Dict[str, Any], wrapper) -> None:
+    """Client-level `timeout` (reference C1 holds it for the HTTP client;
+    locally it bounds generation wall-time — streams past the deadline
+    finish with reason "length")."""
+    if "timeout" not in call_params and getattr(wrapper, "timeout", None):
+        call_params["timeout"] = wrapper.timeout
+
+
 class Completions:
     def __init__(self, wrapper: "KLLMs"):
         self._wrapper = wrapper
@@ -93,6 +101,7 @@ class Completions:
             messages, model, temperature, max_tokens, top_p, frequency_penalty,
             presence_penalty, stop, seed, response_format, kwargs, force_stream_false=True,
         )
+        _apply_wrapper_timeout(call_params, self._wrapper)
 
         def embeddings_wrapper(texts: List[str]) -> List[List[float]]:
             return self._wrapper.get_embeddings(texts, "text-embedding-3-small", 2048, False)
@@ -140,6 +149,7 @@ class Completions:
             messages, model, temperature, max_tokens, top_p, frequency_penalty,
             presence_penalty, stop, seed, response_format, kwargs, force_stream_false=False,
         )
+        _apply_wrapper_timeout(call_params, self._wrapper)
 
         def embeddings_wrapper(texts: List[str]) -> List[List[float]]:
             return self._wrapper.get_embeddings(texts, "text-embedding-3-small", 2048, False)
@@ -182,6 +192,7 @@ class AsyncCompletions:
             messages, model, temperature, max_tokens, top_p, frequency_penalty,
             presence_penalty, stop, seed, response_format, kwargs, force_stream_false=True,
         )
+        _apply_wrapper_timeout(call_params, self._wrapper)
 
         async def embeddings_wrapper(texts: List[str]) -> List[List[float]]:
             return await self._wrapper.get_embeddings(texts, "text-embedding-3-small", 2048, False)
@@ -232,6 +243,7 @@ class AsyncCompletions:
             messages, model, temperature, max_tokens, top_p, frequency_penalty,
             presence_penalty, stop, seed, response_format, kwargs, force_stream_false=False,
         )
+        _apply_wrapper_timeout(call_params, self._wrapper)
 
         async def embeddings_wrapper(texts: List[str]) -> List[List[float]]:
             return await self._wrapper.get_embeddings(texts, "text-embedding-3-small", 2048, False)

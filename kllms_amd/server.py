@@ -84,7 +84,7 @@ def create_app(client):
         messages = body.get("messages")
         if not isinstance(messages, list) or not messages:
             return _error_response("'messages' must be a non-empty list", 400)
-        known = {"model", "messages", "n", "temperature", "top_p", "top_k",
+        known = {"model", "messages", "n", "temperature", "top_p", "top_k", "timeout",
                  "max_tokens", "max_completion_tokens", "stop", "seed",
                  "frequency_penalty", "presence_penalty", "logprobs",
                  "top_logprobs", "logit_bias", "response_format", "tools",

@@ -504,3 +504,30 @@ def test_content_part_arrays_supported():
             messages=[{"role": "user", "content": [{"type": "image_url",
                                                     "image_url": {"url": "x"}}]}],
             model="tiny-llama", max_tokens=4)
+
+
+def test_request_timeout_caps_generation():
+    """Client-level `timeout` bounds generation wall time: the stream
+    finishes with reason "length" at the next step boundary and returns
+    what it has, instead of running out max_tokens."""
+    import time as _time
+
+    from kllms_amd import KLLMs
+
+    c = KLLMs(model="tiny-llama", device="cpu", max_kv_blocks=512,
+              use_hip_graphs=False, seed=0, timeout=1e-6)
+    t0 = _time.monotonic()
+    out = c.chat.completions.create(messages=[{"role": "user", "content": "go"}],
+                                    model="tiny-llama", max_tokens=400,
+                                    temperature=0.0)
+    assert _time.monotonic() - t0 < 30
+    assert out.choices[0].finish_reason == "length"
+    assert len(out.choices[0].message.content or "") < 400
+
+    # per-call override beats the client default
+    c2 = KLLMs(model="tiny-llama", device="cpu", max_kv_blocks=512,
+               use_hip_graphs=False, seed=0)
+    out2 = c2.chat.completions.create(messages=[{"role": "user", "content": "go"}],
+                                      model="tiny-llama", max_tokens=6,
+                                      temperature=0.0, timeout=1e-6)
+    assert out2.choices[0].finish_reason == "length"
