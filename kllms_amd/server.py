@@ -72,6 +72,33 @@ def create_app(client):
             }],
         }
 
+    @app.post("/v1/embeddings")
+    async def embeddings(request: Request):
+        import asyncio as _asyncio
+
+        try:
+            body: Dict[str, Any] = await request.json()
+        except Exception:
+            return _error_response("request body must be JSON", 400)
+        inp = body.get("input")
+        if isinstance(inp, str):
+            inp = [inp]
+        if not isinstance(inp, list) or not all(isinstance(t, str) for t in inp):
+            return _error_response("'input' must be a string or list of strings", 400)
+        model = body.get("model", "text-embedding-3-small")
+        from .client import MAX_TOKENS_PER_MODEL
+
+        if model not in MAX_TOKENS_PER_MODEL:
+            return _error_response(
+                f"unknown embedding model {model!r} "
+                f"(available: {sorted(MAX_TOKENS_PER_MODEL)})", 404, code="model_not_found")
+        try:
+            resp = await _asyncio.to_thread(
+                client.client.embeddings_create, input=inp, model=model)
+        except ValueError as e:
+            return _error_response(str(e), 400)
+        return resp.model_dump()
+
     @app.post("/v1/chat/completions")
     async def chat_completions(request: Request):
         try:

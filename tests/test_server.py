@@ -160,3 +160,31 @@ class TestFuzz:
                 body.setdefault("max_tokens", 4)
             r = http.post("/v1/chat/completions", json=body)
             assert r.status_code < 500, (trial, body, r.text[:300])
+
+
+class TestEmbeddingsRoute:
+    def test_embeddings_shape(self, http):
+        r = http.post("/v1/embeddings", json={
+            "input": ["hello world", "hello world!", "totally different topic here"],
+            "model": "text-embedding-3-small"})
+        assert r.status_code == 200, r.text
+        body = r.json()
+        assert body["object"] == "list" and len(body["data"]) == 3
+        assert body["data"][0]["index"] == 0
+        v = body["data"][0]["embedding"]
+        assert isinstance(v, list) and len(v) > 0
+        assert body["usage"]["prompt_tokens"] > 0
+        # near-duplicates more similar than unrelated text (ngram embedder)
+        import math
+        def cos(a, b):
+            num = sum(x * y for x, y in zip(a, b))
+            den = math.sqrt(sum(x * x for x in a)) * math.sqrt(sum(y * y for y in b))
+            return num / den if den else 0.0
+        e = [d["embedding"] for d in body["data"]]
+        assert cos(e[0], e[1]) > cos(e[0], e[2])
+
+    def test_embeddings_errors(self, http):
+        assert http.post("/v1/embeddings", json={"input": 5}).status_code == 400
+        r = http.post("/v1/embeddings", json={"input": "x", "model": "bogus-model"})
+        assert r.status_code == 404
+        assert r.json()["error"]["code"] == "model_not_found"
