@@ -48,6 +48,13 @@ def _parse_logit_bias(lb):
         raise ValueError("logit_bias must map token ids to numeric biases")
 
 
+# OpenAI responses carry a backend build fingerprint; ours names the local
+# engine so response provenance is visible in logs/dumps
+from .. import __version__ as _pkg_version
+
+_FINGERPRINT = f"kllms_amd-{_pkg_version}-gfx950"
+
+
 class ContextLengthExceededError(ValueError):
     """Prompt does not fit in the model's context window (OpenAI-compatible
     behavior: the API raises context_length_exceeded rather than silently
@@ -385,6 +392,7 @@ class LocalEngineClient:
                     id=f"chatcmpl-{uuid.uuid4().hex[:24]}",
                     choices=choices,
                     created=int(time.time()),
+            system_fingerprint=_FINGERPRINT,
                     model=call_params.get("model", self.config.model),
                     usage=self._mk_usage(out),
                 )
@@ -420,6 +428,7 @@ class LocalEngineClient:
             id=f"chatcmpl-{uuid.uuid4().hex[:24]}",
             choices=choices,
             created=int(time.time()),
+            system_fingerprint=_FINGERPRINT,
             model=model,
             usage=self._mk_usage(out),
             timings=self._mk_timings(out),
@@ -450,6 +459,7 @@ class LocalEngineClient:
             id=f"chatcmpl-{uuid.uuid4().hex[:24]}",
             choices=choices,
             created=int(time.time()),
+            system_fingerprint=_FINGERPRINT,
             model=model,
             usage=self._mk_usage(out),
             timings=self._mk_timings(out),
