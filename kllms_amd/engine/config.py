@@ -135,7 +135,7 @@ class EngineConfig(BaseModel):
     # (token, head) row is stored as x/s with s = amax(|row|)/448 kept in a
     # parallel fp32 [NB, KVH, BS] tensor (~3% overhead), so outlier-heavy
     # real checkpoints don't saturate e4m3's +-448 range)
-    kv_cache_dtype: str = "bf16"
+    kv_cache_dtype: Literal["bf16", "fp8_e4m3"] = "bf16"
     # Fraction of free HBM given to the KV cache after weights are resident
     kv_memory_fraction: float = 0.70
     max_kv_blocks: Optional[int] = None  # explicit cap (used on CPU/tests)
