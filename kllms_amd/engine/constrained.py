@@ -440,6 +440,16 @@ def _schema_ir_body(schema: Dict[str, Any], defs: Dict[str, Any], depth: int, ws
     if "anyOf" in schema or "oneOf" in schema:
         opts = schema.get("anyOf") or schema.get("oneOf")
         return Alt([_schema_ir_body(s, defs, depth + 1, ws) for s in opts])
+    if "allOf" in schema:
+        # pydantic v1-style wrapper: a single-element allOf is just the
+        # element (plus sibling annotations like description). True
+        # multi-constraint intersections aren't DFA-composable here; merge
+        # sibling keys into a single-element body, else reject loudly.
+        parts = schema["allOf"]
+        if len(parts) == 1:
+            merged = {**{k: v for k, v in schema.items() if k != "allOf"}, **parts[0]}
+            return _schema_ir_body(merged, defs, depth + 1, ws)
+        raise SchemaCompileError("multi-element allOf is not supported")
 
     t = schema.get("type")
     if isinstance(t, list):

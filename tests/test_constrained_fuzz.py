@@ -389,3 +389,22 @@ def test_dict_min_max_properties():
         assert _accepts(sch, ok, ws=False), ok
     for bad in ('{}', '{"a":1,"b":2,"c":3}'):
         assert not _accepts(sch, bad, ws=False), bad
+
+
+def test_single_element_allof_unwrapped():
+    """pydantic-v1-style {allOf: [{$ref}], description} wrappers compile to
+    the wrapped schema; multi-element allOf rejects loudly."""
+    import pytest as _pytest
+
+    from kllms_amd.engine.constrained import SchemaCompileError, schema_to_ir
+
+    defs = {"Inner": {"type": "object", "properties": {"x": {"type": "integer"}},
+                      "required": ["x"]}}
+    sch = {"type": "object",
+           "properties": {"inner": {"allOf": [{"$ref": "#/$defs/Inner"}],
+                                    "description": "d"}},
+           "required": ["inner"], "$defs": defs}
+    assert _accepts(sch, '{"inner":{"x":3}}', ws=False)
+    assert not _accepts(sch, '{"inner":{"x":"s"}}', ws=False)
+    with _pytest.raises(SchemaCompileError):
+        schema_to_ir({"allOf": [{"type": "integer"}, {"minimum": 3}]}, {})
